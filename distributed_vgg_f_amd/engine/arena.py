@@ -1,0 +1,143 @@
+"""Contiguous parameter/gradient/momentum arenas + fused SGD.
+
+The reference's optimizer state lived as one TF Variable per tensor on the
+parameter servers, updated by per-variable momentum-apply kernels. The
+MI355X-native design replaces that with four flat arenas sized for the whole
+model:
+
+    W32  fp32 master weights          (one tensor, device)
+    Wc   compute-dtype weights (bf16) (model parameters are views of this)
+    G    grad arena (bf16 default)    (param.grad are views; all-reduce
+                                       operates on contiguous slices = buckets)
+    M32  fp32 momentum
+
+ordered in REVERSE registration order (fc8 first ... conv1 last) so gradient
+buckets become ready in backward-completion order and all-reduce overlaps the
+remaining backward. One fused HIP kernel then
+updates all four arenas per step; on CPU the same semantics run via torch
+(Wc is W32 itself, fp32).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+_ALIGN = 128  # element alignment of every param slice (vector-width friendly)
+
+
+def _aligned(n: int) -> int:
+    return (n + _ALIGN - 1) // _ALIGN * _ALIGN
+
+
+@dataclass
+class ParamSlot:
+    name: str
+    offset: int
+    numel: int
+    shape: torch.Size
+    param: nn.Parameter
+
+
+@dataclass
+class Bucket:
+    start: int          # arena element offset
+    end: int
+    slots: List[ParamSlot]
+
+
+class ArenaOptimizer:
+    def __init__(self, model: nn.Module, device: torch.device,
+                 compute_dtype: torch.dtype = torch.bfloat16,
+                 grad_dtype: torch.dtype = torch.bfloat16,
+                 lr: float = 0.01, momentum: float = 0.9,
+                 weight_decay: float = 0.0, bucket_mb: int = 25):
+        self.device = device
+        if device.type != "cuda":
+            compute_dtype = torch.float32
+            grad_dtype = torch.float32
+        self.compute_dtype, self.grad_dtype = compute_dtype, grad_dtype
+        self.lr, self.momentum, self.weight_decay = lr, momentum, weight_decay
+
+        named = [(n, p) for n, p in model.named_parameters() if p.requires_grad]
+        named.reverse()  # fc8 ... conv1: bucket order == backward completion
+
+        total = sum(_aligned(p.numel()) for _, p in named)
+        self.numel = total
+        self.w32 = torch.zeros(total, dtype=torch.float32, device=device)
+        self.m32 = torch.zeros(total, dtype=torch.float32, device=device)
+        self.g = torch.zeros(total, dtype=grad_dtype, device=device)
+        if compute_dtype == torch.float32:
+            self.wc = self.w32
+        else:
+            self.wc = torch.zeros(total, dtype=compute_dtype, device=device)
+
+        self.slots: List[ParamSlot] = []
+        ofs = 0
+        for name, p in named:
+            n = p.numel()
+            self.w32[ofs:ofs + n].copy_(p.detach().float().reshape(-1).to(device))
+            view = self.wc[ofs:ofs + n].view(p.shape)
+            newp = nn.Parameter(view, requires_grad=True)
+            newp.grad = self.g[ofs:ofs + n].view(p.shape)
+            owner = model.get_submodule(name.rsplit(".", 1)[0]) if "." in name \
+                else model
+            owner._parameters[name.rsplit(".", 1)[-1]] = newp
+            self.slots.append(ParamSlot(name, ofs, n, p.shape, newp))
+            ofs += _aligned(n)
+        if compute_dtype != torch.float32:
+            self.wc.copy_(self.w32.to(compute_dtype))
+
+        # buckets: groups of whole params up to bucket_mb (grad-dtype MB)
+        self.buckets: List[Bucket] = []
+        limit = bucket_mb * (1 << 20) // self.g.element_size()
+        cur: List[ParamSlot] = []
+        cur_start = 0
+        cur_elems = 0
+        for s in self.slots:
+            if cur and cur_elems + _aligned(s.numel) > limit:
+                self.buckets.append(Bucket(cur_start, cur_start + cur_elems, cur))
+                cur, cur_start, cur_elems = [], s.offset, 0
+            if not cur:
+                cur_start = s.offset
+            cur.append(s)
+            cur_elems += _aligned(s.numel)
+        if cur:
+            self.buckets.append(Bucket(cur_start, cur_start + cur_elems, cur))
+
+    @property
+    def parameters(self) -> List[nn.Parameter]:
+        return [s.param for s in self.slots]
+
+    def zero_grad(self) -> None:
+        self.g.zero_()
+
+    def step(self, grad_scale: float = 1.0, lr: Optional[float] = None) -> None:
+        ops.sgd_momentum_step(self.w32, self.wc, self.g, self.m32,
+                              lr if lr is not None else self.lr,
+                              self.momentum, self.weight_decay, grad_scale)
+
+    # --- checkpoint interface (fp32 master + momentum) ---
+    def state_tensors(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        return self.w32, self.m32
+
+    def named_master(self):
+        for s in self.slots:
+            yield s.name, self.w32[s.offset:s.offset + s.numel].view(s.shape)
+
+    def load_master(self, name: str, value: torch.Tensor) -> None:
+        for s in self.slots:
+            if s.name == name:
+                self.w32[s.offset:s.offset + s.numel].copy_(
+                    value.reshape(-1).float().to(self.device))
+                return
+        raise KeyError(name)
+
+    def sync_compute_weights(self) -> None:
+        if self.wc.data_ptr() != self.w32.data_ptr():
+            self.wc.copy_(self.w32.to(self.compute_dtype))

@@ -1,0 +1,140 @@
+"""train()/eval() engine (SURVEY.md sections 3.4, 1.3).
+
+The reference's MonitoredTrainingSession step loop (sess.run(train_op) with
+checkpoint/logging hooks) becomes an explicit
+MI355X-native loop: static per-step schedule, zero host syncs inside the
+step (loss D2H only at log windows), bucketed all-reduce overlapped with
+backward, one fused SGD over the arenas.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Optional
+
+import torch
+
+from .. import ops
+from ..config import Config
+from ..data import SyntheticPipeline
+from ..models import build_model
+from ..parallel.dist import DistContext, GradReducer, broadcast_init, \
+    init_distributed
+from . import checkpoint as ckpt
+from .arena import ArenaOptimizer
+from .metrics import MetricsLogger, PhaseTimers
+
+
+def _device(ctx: DistContext) -> torch.device:
+    if torch.cuda.is_available():
+        return torch.device("cuda", ctx.local_rank)
+    return torch.device("cpu")
+
+
+class Trainer:
+    def __init__(self, cfg: Config, ctx: Optional[DistContext] = None,
+                 device: Optional[torch.device] = None):
+        self.cfg = cfg
+        self.ctx = ctx if ctx is not None else init_distributed(cfg.backend)
+        self.device = device if device is not None else _device(self.ctx)
+
+        torch.manual_seed(cfg.seed)          # same init on every rank; C4
+        self.model = build_model(cfg)        # broadcast still run for safety
+        self.model.train()
+
+        compute_dtype = torch.bfloat16 if (cfg.precision == "bf16" and
+                                           self.device.type == "cuda") \
+            else torch.float32
+        grad_dtype = torch.bfloat16 if (cfg.grad_dtype == "bf16" and
+                                        self.device.type == "cuda") \
+            else torch.float32
+        self.arena = ArenaOptimizer(self.model, self.device, compute_dtype,
+                                    grad_dtype, cfg.lr, cfg.momentum,
+                                    cfg.weight_decay, cfg.bucket_mb)
+        broadcast_init(self.arena, self.ctx)
+        self.reducer = GradReducer(self.arena, self.ctx, cfg.overlap_comm)
+        self.pipeline = SyntheticPipeline(cfg.batch_size, cfg.image_size,
+                                          cfg.channels, cfg.num_classes,
+                                          self.device, compute_dtype,
+                                          cfg.seed + self.ctx.rank)
+        self.metrics = MetricsLogger(cfg.metrics_path, self.ctx.rank)
+        self.timers = PhaseTimers(self.device, enabled=False)
+        self.step_i = 0
+        self.global_batch = cfg.batch_size * self.ctx.world_size
+
+        if cfg.resume and cfg.checkpoint_dir:
+            self.step_i = ckpt.restore_latest(cfg.checkpoint_dir, self.arena)
+            if self.ctx.is_chief and self.step_i:
+                self.metrics.log(event="resumed", step=self.step_i)
+
+    # ------------------------------------------------------------------ #
+    def train_step(self) -> torch.Tensor:
+        """One synchronous-SGD step over the global batch. Returns the
+        per-rank summed loss tensor (device, not synced)."""
+        cfg = self.cfg
+        if (cfg.fault_inject_step == self.step_i and
+                cfg.fault_inject_rank == self.ctx.rank):
+            raise RuntimeError(
+                f"fault-inject: rank {self.ctx.rank} dies at step {self.step_i}")
+        images, labels = self.pipeline.next()
+        self.reducer.prepare_step()
+        logits = self.model(images)
+        loss = ops.softmax_xent(logits, labels).sum()
+        loss.backward()
+        self.reducer.finish()
+        # loss summed per sample; SUM all-reduce over ranks => scale by 1/GB
+        self.arena.step(grad_scale=1.0 / self.global_batch)
+        self.arena.zero_grad()
+        self.step_i += 1
+        return loss
+
+    def train(self, steps: Optional[int] = None) -> None:
+        cfg = self.cfg
+        steps = steps if steps is not None else cfg.train_steps
+        window_t0 = time.time()
+        window_steps = 0
+        last_loss = None
+        for _ in range(steps):
+            last_loss = self.train_step()
+            window_steps += 1
+            if cfg.log_every and self.step_i % cfg.log_every == 0:
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                dt = time.time() - window_t0
+                ips = self.global_batch * window_steps / dt
+                self.metrics.log(step=self.step_i,
+                                 loss=float(last_loss) / cfg.batch_size,
+                                 images_per_sec=ips,
+                                 ms_per_step=1000.0 * dt / window_steps,
+                                 lr=cfg.lr)
+                window_t0 = time.time()
+                window_steps = 0
+            if (cfg.checkpoint_every and cfg.checkpoint_dir and
+                    self.step_i % cfg.checkpoint_every == 0 and
+                    self.ctx.is_chief):
+                ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i)
+        if cfg.checkpoint_dir and self.ctx.is_chief:
+            ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i)
+
+    # ------------------------------------------------------------------ #
+    @torch.no_grad()
+    def evaluate(self, steps: Optional[int] = None) -> dict:
+        """R7: accuracy over held-out (synthetic) batches from the latest
+        checkpoint state."""
+        cfg = self.cfg
+        steps = steps if steps is not None else cfg.eval_steps
+        self.model.eval()
+        correct = torch.zeros((), dtype=torch.long, device=self.device)
+        total = 0
+        for _ in range(steps):
+            images, labels = self.pipeline.next()
+            logits = self.model(images)
+            correct += (logits.float().argmax(1) == labels).sum()
+            total += labels.numel()
+        self.model.train()
+        acc = float(correct) / max(total, 1)
+        if self.ctx.is_chief:
+            self.metrics.log(event="eval", step=self.step_i, accuracy=acc,
+                             samples=total)
+        return {"accuracy": acc, "samples": total}
