@@ -1,0 +1,98 @@
+#!/usr/bin/env python3
+"""Headline benchmark (driver contract; BASELINE.json).
+
+Measures images/sec (whole node) for VGG-F 224x224, bs=256/GPU, bf16,
+synthetic data, random-init weights — fwd + bwd + all-reduce + fused SGD
+per step (nothing skipped in the timed region).
+
+Single GPU:     python bench.py --gpus 1 --steps 20 --warmup 5
+Multi (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--model", type=str, default="vggf")
+    p.add_argument("--bucket-mb", type=int, default=25)
+    p.add_argument("--overlap-comm", type=int, default=1)
+    args = p.parse_args()
+
+    from distributed_vgg_f_amd.config import Config
+    from distributed_vgg_f_amd.engine.trainer import Trainer
+    from distributed_vgg_f_amd.parallel.dist import init_distributed
+
+    use_gpu = torch.cuda.is_available()
+    cfg = Config(model=args.model, batch_size=args.batch_size,
+                 precision="bf16" if use_gpu else "fp32",
+                 log_every=0, bucket_mb=args.bucket_mb,
+                 overlap_comm=bool(args.overlap_comm))
+    ctx = init_distributed(cfg.backend)
+    world = ctx.world_size
+    if world > 1 and args.gpus != world:
+        print(f"[bench] note: WORLD_SIZE={world} overrides --gpus={args.gpus}",
+              file=sys.stderr)
+    n_gpus = world if world > 1 else 1
+
+    device = torch.device("cuda", ctx.local_rank) if use_gpu \
+        else torch.device("cpu")
+    t = Trainer(cfg, ctx=ctx, device=device)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        t.train_step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        t.train_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    et = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        torch.distributed.all_reduce(et, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(et[0])
+
+    global_batch = args.batch_size * n_gpus
+    ips = global_batch * args.steps / elapsed
+    ms = 1000.0 * elapsed / args.steps
+    if ctx.is_chief:
+        print(json.dumps({
+            "metric": "images/sec (whole node) VGG-F 224x224 bs=256/GPU",
+            "value": round(ips, 1),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model, "global_batch": global_batch,
+                       "image_size": 224, "seq_len": None,
+                       "parallelism": f"dp{n_gpus}"},
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,429 @@
+// Host launchers + auxiliary kernels for the MFMA GEMM family:
+// FC layers, implicit-GEMM convolutions (K1/K2/K3),
+// im2col for small-Cin first layers, transposes (K16 class), split-K
+// finalize. All activations NHWC bf16, conv weights HWIO bf16.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+#include "gemm_tile.h"
+
+namespace dvgg {
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+static inline const u16* bfp(const at::Tensor& t) {
+  return reinterpret_cast<const u16*>(t.data_ptr<at::BFloat16>());
+}
+static inline u16* bfp_mut(at::Tensor& t) {
+  return reinterpret_cast<u16*>(t.data_ptr<at::BFloat16>());
+}
+
+static void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+static inline int ceil8(int v) { return (v + 7) & ~7; }
+
+// ------------------------------------------------------------ transpose2d
+// [R][C] -> [C][Rp] (Rp >= R, zero-padded cols), bf16, LDS-tiled.
+__global__ void transpose2d_kernel(const u16* __restrict__ in,
+                                   u16* __restrict__ out, int R, int C,
+                                   int Rp) {
+  __shared__ u16 tile[32][33];
+  int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;  // 32 x 8
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int r = r0 + ty + i * 8, c = c0 + tx;
+    tile[ty + i * 8][tx] = (r < R && c < C) ? in[(i64)r * C + c] : u16(0);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int c = c0 + ty + i * 8, r = r0 + tx;
+    if (c < C && r < Rp) out[(i64)c * Rp + r] = tile[tx][ty + i * 8];
+  }
+}
+
+static at::Tensor transpose2d(const at::Tensor& in, int Rp = -1) {
+  int R = in.size(0), C = in.numel() / in.size(0);
+  if (Rp < 0) Rp = R;
+  auto out = at::zeros({C, Rp}, in.options());
+  dim3 grid(ceil_div(C, 32), ceil_div(R, 32));
+  hipLaunchKernelGGL(transpose2d_kernel, grid, dim3(256), 0, cur_stream(),
+                     bfp(in), bfp_mut(out), R, C, Rp);
+  return out;
+}
+
+// --------------------------------------------- NHWC -> C,N,Hp,Wp (padded)
+// For the wgrad gathers: writes x[n][ih][iw][c] to out[c][n][ih+ro][iw+co];
+// out pre-zeroed (halo + alignment pad).
+__global__ void nhwc2cnhw_kernel(const u16* __restrict__ in,
+                                 u16* __restrict__ out, int N, int H, int W,
+                                 int C, int Hp, int Wp, int ro, int co) {
+  __shared__ u16 tile[32][33];
+  int nh = blockIdx.x;            // n*H + ih
+  int w0 = blockIdx.y * 32, c0 = blockIdx.z * 32;
+  int n = nh / H, ih = nh % H;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int iw = w0 + ty + i * 8, c = c0 + tx;
+    tile[ty + i * 8][tx] = (iw < W && c < C)
+        ? in[(((i64)n * H + ih) * W + iw) * C + c] : u16(0);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int c = c0 + ty + i * 8, iw = w0 + tx;
+    if (c < C && iw < W)
+      out[(((i64)c * N + n) * Hp + (ih + ro)) * Wp + iw + co] =
+          tile[tx][ty + i * 8];
+  }
+}
+
+static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
+                                int co) {
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  auto out = at::zeros({C, N, Hp, Wp}, x.options());
+  dim3 grid(N * H, ceil_div(W, 32), ceil_div(C, 32));
+  hipLaunchKernelGGL(nhwc2cnhw_kernel, grid, dim3(256), 0, cur_stream(),
+                     bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
+  return out;
+}
+
+// ------------------------------------------------------------- im2col
+// For Cin % 8 != 0 first layers (conv1 11x11s4 Cin=3; ResNet stem 7x7s2):
+// A [Mp][Kp] (TRANS=false) or A^T [Kp][Mp] (TRANS=true), zero-padded.
+template <bool TRANS>
+__global__ void im2col_kernel(const u16* __restrict__ x, u16* __restrict__ out,
+                              int M, int Mp, int K, int Kp, int N, int H,
+                              int W, int C, int OH, int OW, int KH, int KW,
+                              int stride, int pad) {
+  i64 chunks = TRANS ? ((i64)Kp * (Mp / 8)) : ((i64)Mp * (Kp / 8));
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= chunks) return;
+  s16x8 v;
+  if (!TRANS) {
+    i64 m = id / (Kp / 8);
+    int k0 = int(id % (Kp / 8)) * 8;
+    int n = int(m / (OH * OW)), r = int(m % (OH * OW));
+    int oh = r / OW, ow = r % OW;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = k0 + j;
+      int f = k / C, c = k % C;
+      int kh = f / KW, kw = f % KW;
+      int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
+      bool ok = m < M && k < K && (unsigned)ih < (unsigned)H &&
+                (unsigned)iw < (unsigned)W;
+      v[j] = ok ? short(x[(((i64)n * H + ih) * W + iw) * C + c]) : short(0);
+    }
+    *reinterpret_cast<s16x8*>(out + m * Kp + k0) = v;
+  } else {
+    i64 k = id / (Mp / 8);
+    i64 m0 = (id % (Mp / 8)) * 8;
+    int f = int(k / C), c = int(k % C);
+    int kh = f / KW, kw = f % KW;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      i64 m = m0 + j;
+      int n = int(m / (OH * OW)), r = int(m % (OH * OW));
+      int oh = r / OW, ow = r % OW;
+      int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
+      bool ok = m < M && k < K && (unsigned)ih < (unsigned)H &&
+                (unsigned)iw < (unsigned)W;
+      v[j] = ok ? short(x[(((i64)n * H + ih) * W + iw) * C + c]) : short(0);
+    }
+    *reinterpret_cast<s16x8*>(out + k * Mp + m0) = v;
+  }
+}
+
+// ------------------------------------------------------- split-K finalize
+// out[r][c] = cast(c32[r][c] (+bias[c]) (relu)); out R rows (<= rows of c32)
+template <int EPI>
+__global__ void finalize_kernel(const float* __restrict__ c32,
+                                u16* __restrict__ out,
+                                const u16* __restrict__ bias, i64 total,
+                                int C) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= total) return;
+  float v = c32[id];
+  if (EPI == 1 || EPI == 2) v += bf2f(bias[id % C]);
+  if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
+  out[id] = f2bf(v);
+}
+
+static void run_finalize(const at::Tensor& c32, at::Tensor& out,
+                         const at::Tensor* bias, int epi, int R, int C) {
+  i64 total = (i64)R * C;
+  int blocks = int((total + 255) / 256);
+  const u16* bp = bias ? bfp(*bias) : nullptr;
+  auto s = cur_stream();
+  switch (epi) {
+    case 0: hipLaunchKernelGGL(finalize_kernel<0>, dim3(blocks), dim3(256), 0,
+                               s, c32.data_ptr<float>(), bfp_mut(out), bp,
+                               total, C); break;
+    case 1: hipLaunchKernelGGL(finalize_kernel<1>, dim3(blocks), dim3(256), 0,
+                               s, c32.data_ptr<float>(), bfp_mut(out), bp,
+                               total, C); break;
+    case 2: hipLaunchKernelGGL(finalize_kernel<2>, dim3(blocks), dim3(256), 0,
+                               s, c32.data_ptr<float>(), bfp_mut(out), bp,
+                               total, C); break;
+    default: hipLaunchKernelGGL(finalize_kernel<3>, dim3(blocks), dim3(256),
+                                0, s, c32.data_ptr<float>(), bfp_mut(out), bp,
+                                total, C); break;
+  }
+}
+
+// ------------------------------------------------------------ gemm driver
+template <class GA, class GB>
+static void launch_gemm(const GA& ga, const GB& gb, at::Tensor* outB,
+                        at::Tensor* c32, const at::Tensor* bias, int epi,
+                        int M, int N, int K, int splitk) {
+  int tiles = ceil_div(M, BM) * ceil_div(N, BN);
+  auto s = cur_stream();
+  const u16* bp = bias ? bfp(*bias) : nullptr;
+  if (splitk > 1) {
+    dim3 grid(tiles, splitk);
+    hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true>), grid, dim3(THREADS), 0,
+                       s, ga, gb, nullptr, c32->data_ptr<float>(), nullptr, M,
+                       N, K, 0, K);
+  } else {
+    dim3 grid(tiles);
+    switch (epi) {
+      case 0:
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, false>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, nullptr, M, N, K, 0, K);
+        break;
+      case 1:
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 1, false>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, bp, M, N, K, 0, K);
+        break;
+      case 2:
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 2, false>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, bp, M, N, K, 0, K);
+        break;
+      default:
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 3, false>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, nullptr, M, N, K, 0, K);
+        break;
+    }
+  }
+}
+
+static int pick_splitk(int tiles, int K, int target_blocks = 512) {
+  if (tiles >= target_blocks) return 1;
+  int kt = std::max(1, K / BK);
+  int want = std::min(kt, std::max(1, target_blocks / std::max(1, tiles)));
+  return want;
+}
+
+static int epi_of(bool has_bias, bool relu) {
+  return has_bias ? (relu ? 2 : 1) : (relu ? 3 : 0);
+}
+
+// ---------------------------------------------------------------- linear
+at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias, bool relu) {
+  check_bf16(x, "x"); check_bf16(w, "w");
+  int M = x.size(0), K = x.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(0) == K && K % 8 == 0 && N % 8 == 0, "bad fc shape");
+  auto y = at::empty({M, N}, x.options());
+  auto wT = transpose2d(w);  // [N][K]
+  PlainGather ga{bfp(x), M, K, K};
+  PlainGather gb{bfp(wT), N, K, K};
+  int tiles = ceil_div(M, BM) * ceil_div(N, BN);
+  int epi = epi_of(bias.has_value(), relu);
+  int sk = pick_splitk(tiles, K);
+  if (sk > 1) {
+    auto c32 = at::zeros({M, N}, x.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, N, K, sk);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N);
+  } else {
+    launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, N, K, 1);
+  }
+  return y;
+}
+
+at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
+  check_bf16(dy, "dy"); check_bf16(w, "w");
+  int M = dy.size(0), N = dy.size(1), K = w.size(0);
+  TORCH_CHECK(w.size(1) == N && N % 8 == 0, "bad fc dgrad shape");
+  auto dx = at::empty({M, K}, dy.options());
+  PlainGather ga{bfp(dy), M, N, N};
+  PlainGather gb{bfp(w), K, N, N};  // Bt[kin][n] = w — no transpose needed
+  int tiles = ceil_div(M, BM) * ceil_div(K, BN);
+  int sk = pick_splitk(tiles, N);
+  if (sk > 1) {
+    auto c32 = at::zeros({M, K}, dy.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, K, N, sk);
+    run_finalize(c32, dx, nullptr, 0, M, K);
+  } else {
+    launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, K, N, 1);
+  }
+  return dx;
+}
+
+at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
+  check_bf16(x, "x"); check_bf16(dy, "dy");
+  int M = x.size(0), K = x.size(1), N = dy.size(1);
+  TORCH_CHECK(dy.size(0) == M && M % 8 == 0, "bad fc wgrad shape");
+  auto xT = transpose2d(x);    // [K][M]
+  auto dyT = transpose2d(dy);  // [N][M]
+  auto dw = at::empty({K, N}, x.options());
+  PlainGather ga{bfp(xT), K, M, M};
+  PlainGather gb{bfp(dyT), N, M, M};
+  int tiles = ceil_div(K, BM) * ceil_div(N, BN);
+  int sk = pick_splitk(tiles, M);
+  if (sk > 1) {
+    auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, K, N, M, sk);
+    run_finalize(c32, dw, nullptr, 0, K, N);
+  } else {
+    launch_gemm(ga, gb, &dw, nullptr, nullptr, 0, K, N, M, 1);
+  }
+  return dw;
+}
+
+// ------------------------------------------------------------------ conv
+static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
+                           int& OH, int& OW) {
+  OH = (H + 2 * pad - KH) / stride + 1;
+  OW = (W + 2 * pad - KW) / stride + 1;
+}
+
+at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias, int64_t stride_,
+                      int64_t pad_, bool relu) {
+  check_bf16(x, "x"); check_bf16(w, "w");
+  int stride = int(stride_), pad = int(pad_);
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int KH = w.size(0), KW = w.size(1), Cout = w.size(3);
+  TORCH_CHECK(w.size(2) == Cin, "conv weight Cin mismatch");
+  int OH, OW;
+  conv_out_shape(H, W, KH, KW, stride, pad, OH, OW);
+  int M = N * OH * OW, Kfull = KH * KW * Cin;
+  auto y = at::empty({N, OH, OW, Cout}, x.options());
+  auto y2d = y.view({M, Cout});
+  int epi = epi_of(bias.has_value(), relu);
+
+  if (Cin % 8 != 0) {
+    // first-layer path: im2col (SURVEY.md section 7 hard part 2)
+    int Kp = ceil8(Kfull), Mp = ceil8(M);
+    auto A = at::empty({Mp, Kp}, x.options());
+    i64 chunks = (i64)Mp * (Kp / 8);
+    hipLaunchKernelGGL((im2col_kernel<false>),
+                       dim3(int((chunks + 255) / 256)), dim3(256), 0,
+                       cur_stream(), bfp(x), bfp_mut(A), M, Mp, Kfull, Kp, N,
+                       H, W, Cin, OH, OW, KH, KW, stride, pad);
+    auto wT = transpose2d(w.view({Kfull, Cout}), Kp);  // [Cout][Kp]
+    PlainGather ga{bfp(A), M, Kp, Kp};
+    PlainGather gb{bfp(wT), Cout, Kp, Kp};
+    launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
+                Kp, 1);
+    return y;
+  }
+  auto wT = transpose2d(w.view({Kfull, Cout}));  // [Cout][K]
+  ConvFwdGather ga{bfp(x), M, H, W, Cin, OH, OW, stride, pad, KW};
+  PlainGather gb{bfp(wT), Cout, Kfull, Kfull};
+  int tiles = ceil_div(M, BM) * ceil_div(Cout, BN);
+  int sk = pick_splitk(tiles, Kfull);
+  if (sk > 1) {
+    auto c32 = at::zeros({M, Cout}, x.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cout, Kfull, sk);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout);
+  } else {
+    launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
+                Kfull, 1);
+  }
+  return y;
+}
+
+at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
+                        int64_t W_, int64_t stride_, int64_t pad_) {
+  check_bf16(dy, "dy"); check_bf16(w, "w");
+  int H = int(H_), W = int(W_), stride = int(stride_), pad = int(pad_);
+  int N = dy.size(0), OH = dy.size(1), OW = dy.size(2), Cout = dy.size(3);
+  int KH = w.size(0), KW = w.size(1), Cin = w.size(2);
+  TORCH_CHECK(Cout % 8 == 0, "dgrad needs Cout % 8 == 0");
+  TORCH_CHECK(Cin % 8 == 0,
+              "dgrad for Cin<8 first layers is unused by design");
+  int M = N * H * W, K = KH * KW * Cout;
+  auto dx = at::empty({N, H, W, Cin}, dy.options());
+  // Bt[ci][(kh,kw,co)] = w[kh][kw][ci][co]: permute HWIO -> [Cin][KhKw*Cout]
+  auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
+      .view({Cin, KH * KW * Cout});
+  ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad, KW};
+  PlainGather gb{bfp(wD), Cin, K, K};
+  int tiles = ceil_div(M, BM) * ceil_div(Cin, BN);
+  int sk = pick_splitk(tiles, K);
+  if (sk > 1) {
+    auto c32 = at::zeros({M, Cin}, dy.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cin, K, sk);
+    run_finalize(c32, dx, nullptr, 0, M, Cin);
+  } else {
+    launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, Cin, K, 1);
+  }
+  return dx;
+}
+
+at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
+                        int64_t KW_, int64_t stride_, int64_t pad_) {
+  check_bf16(dy, "dy"); check_bf16(x, "x");
+  int KH = int(KH_), KW = int(KW_), stride = int(stride_), pad = int(pad_);
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int OH = dy.size(1), OW = dy.size(2), Cout = dy.size(3);
+  int Kfull = KH * KW * Cin;
+  auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
+  auto dw2d = dw.view({Kfull, Cout});
+
+  if (Cin % 8 != 0) {
+    // first-layer path: GEMM over im2col^T
+    int M = N * OH * OW, Kp = ceil8(Kfull), Mp = ceil8(M);
+    auto AT = at::empty({Kp, Mp}, x.options());
+    i64 chunks = (i64)Kp * (Mp / 8);
+    hipLaunchKernelGGL((im2col_kernel<true>),
+                       dim3(int((chunks + 255) / 256)), dim3(256), 0,
+                       cur_stream(), bfp(x), bfp_mut(AT), M, Mp, Kfull, Kp, N,
+                       H, W, Cin, OH, OW, KH, KW, stride, pad);
+    auto dyT = transpose2d(dy.view({M, Cout}), Mp);  // [Cout][Mp]
+    PlainGather ga{bfp(AT), Kfull, Mp, Mp};
+    PlainGather gb{bfp(dyT), Cout, Mp, Mp};
+    int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
+    int sk = pick_splitk(tiles, Mp, 1024);
+    auto c32 = at::zeros({Kp, Cout}, x.options().dtype(at::kFloat));
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Mp,
+                std::max(sk, 2));
+    run_finalize(c32.narrow(0, 0, Kfull).contiguous(), dw2d, nullptr, 0,
+                 Kfull, Cout);
+    return dw;
+  }
+
+  int OWp = ceil8(OW);
+  int Kred = N * OH * OWp;
+  int Hp = H + 2 * pad;
+  int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+  auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad);           // [Cin][N][Hp][Wp]
+  auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0);            // [Cout][N][OH][OWp]
+  WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride};
+  WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp};
+  int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  return dw;
+}
+
+}  // namespace dvgg

@@ -1,0 +1,72 @@
+// Python bindings for the CDNA4 kernel library (_hipops).
+// Compiled by hipcc together with the .hip TUs; see setup.py / build_hip.py.
+
+#include <torch/extension.h>
+
+namespace dvgg {
+
+// gemm_ops.hip
+at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&,
+                      const c10::optional<at::Tensor>&, bool);
+at::Tensor linear_dgrad(const at::Tensor&, const at::Tensor&);
+at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&);
+at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&,
+                      const c10::optional<at::Tensor>&, int64_t, int64_t,
+                      bool);
+at::Tensor conv2d_dgrad(const at::Tensor&, const at::Tensor&, int64_t,
+                        int64_t, int64_t, int64_t);
+at::Tensor conv2d_wgrad(const at::Tensor&, const at::Tensor&, int64_t,
+                        int64_t, int64_t, int64_t);
+
+// mem.hip
+at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
+at::Tensor add_relu_fwd(const at::Tensor&, const at::Tensor&);
+at::Tensor bias_grad(const at::Tensor&);
+void sgd_step(at::Tensor&, at::Tensor&, const at::Tensor&, at::Tensor&,
+              double, double, double, double);
+std::vector<at::Tensor> dropout_fwd(const at::Tensor&, double, int64_t,
+                                    int64_t);
+at::Tensor dropout_bwd(const at::Tensor&, const at::Tensor&, double);
+
+// lrnpool.hip
+std::vector<at::Tensor> lrn_fwd(const at::Tensor&, int64_t, double, double,
+                                double);
+at::Tensor lrn_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                   const at::Tensor&, int64_t, double, double);
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor&, int64_t, int64_t,
+                                    int64_t, int64_t);
+at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, int64_t, int64_t,
+                       int64_t, int64_t, int64_t, int64_t);
+
+// softbn.hip
+std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor&, const at::Tensor&);
+std::vector<at::Tensor> bn_fwd(const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, double, bool);
+std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&);
+
+}  // namespace dvgg
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X-native CDNA4 kernels for distributed_vgg_f_amd";
+  m.def("linear_fwd", &dvgg::linear_fwd);
+  m.def("linear_dgrad", &dvgg::linear_dgrad);
+  m.def("linear_wgrad", &dvgg::linear_wgrad);
+  m.def("conv2d_fwd", &dvgg::conv2d_fwd);
+  m.def("conv2d_dgrad", &dvgg::conv2d_dgrad);
+  m.def("conv2d_wgrad", &dvgg::conv2d_wgrad);
+  m.def("relu_bwd", &dvgg::relu_bwd);
+  m.def("add_relu_fwd", &dvgg::add_relu_fwd);
+  m.def("bias_grad", &dvgg::bias_grad);
+  m.def("sgd_step", &dvgg::sgd_step);
+  m.def("dropout_fwd", &dvgg::dropout_fwd);
+  m.def("dropout_bwd", &dvgg::dropout_bwd);
+  m.def("lrn_fwd", &dvgg::lrn_fwd);
+  m.def("lrn_bwd", &dvgg::lrn_bwd);
+  m.def("maxpool_fwd", &dvgg::maxpool_fwd);
+  m.def("maxpool_bwd", &dvgg::maxpool_bwd);
+  m.def("softmax_xent_fwd", &dvgg::softmax_xent_fwd);
+  m.def("bn_fwd", &dvgg::bn_fwd);
+  m.def("bn_bwd", &dvgg::bn_bwd);
+}

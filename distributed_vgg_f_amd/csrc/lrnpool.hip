@@ -1,0 +1,313 @@
+// LRN and 3x3/s2 maxpool (K6/K7), NHWC.
+//
+// LRN design notes (MI355X): channels are innermost (NHWC), so the
+// cross-channel window is a register sliding window per (n,h,w) position,
+// 8-wide vectorized. A naive powf costs 2-3 transcendentals per element
+// (~0.5 ms/step at bs=256 against a ~60 us bandwidth bound), so s^-beta and
+// s^-(1+beta) are evaluated with a plain-VALU exp2/log2 polynomial
+// (~1e-3 rel error, far below bf16 resolution) and BOTH are saved in fwd —
+// backward then needs no transcendentals at all:
+//   y  = x * sb,            sb  = s^-beta
+//   dx = dy*sb - (2*alpha*beta/n) * x * sum_win(dy * x * snb),
+//   snb = s^-(1+beta),      s = k + alpha/n * sum_win(x^2)
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace dvgg {
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+static inline const u16* bfp(const at::Tensor& t) {
+  return reinterpret_cast<const u16*>(t.data_ptr<at::BFloat16>());
+}
+static inline u16* bfp_mut(at::Tensor& t) {
+  return reinterpret_cast<u16*>(t.data_ptr<at::BFloat16>());
+}
+static void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 &&
+              t.is_contiguous(), name, " must be contiguous GPU bf16");
+}
+
+// ---- plain-VALU log2/exp2 (no v_log/v_exp: ~4x issue rate) --------------
+DEV float fast_log2(float s) {
+  union { float f; u32 u; } t;
+  t.f = s;
+  int e = int((t.u >> 23) & 255u) - 127;
+  t.u = (t.u & 0x007fffffu) | 0x3f800000u;  // m in [1,2)
+  float m = t.f;
+  if (m > 1.41421356f) { m *= 0.5f; e += 1; }
+  float f = m - 1.0f;  // in [-0.293, 0.414)
+  float p = f * (1.f - f * (0.5f - f * (0.33333333f - f * (0.25f - f * 0.2f))));
+  return float(e) + p * 1.44269504f;
+}
+
+DEV float fast_exp2(float y) {
+  float n = floorf(y);
+  float f = y - n;
+  float p = 1.f + f * (0.69314718f + f * (0.24022651f + f * (0.05550411f +
+            f * (0.00961813f + f * 0.00133336f))));
+  union { float f; u32 u; } t;
+  t.u = u32(int(n) + 127) << 23;
+  return p * t.f;
+}
+
+// ------------------------------------------------------------- LRN fwd
+// one thread per (n,h,w); C % 8 == 0. Saves sb, snb (bf16).
+__global__ void lrn_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
+                               u16* __restrict__ sb_out,
+                               u16* __restrict__ snb_out, i64 npos, int C,
+                               float an, float k, float beta) {
+  i64 pos = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (pos >= npos) return;
+  const u16* xr = x + pos * C;
+  u16* yr = y + pos * C;
+  u16* sbr = sb_out + pos * C;
+  u16* snbr = snb_out + pos * C;
+
+  float q[12];  // [0,1]: prev tail x^2, [2..9]: current chunk, [10,11]: next
+  q[0] = q[1] = 0.f;
+  float xc[8];
+  s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) { xc[j] = bf2f(u16(cur[j])); }
+  const int nchunks = C / 8;
+  for (int ch = 0; ch < nchunks; ++ch) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) q[2 + j] = xc[j] * xc[j];
+    float xn0 = 0.f, xn1 = 0.f;
+    s16x8 nxt{};
+    if (ch + 1 < nchunks) {
+      nxt = *reinterpret_cast<const s16x8*>(xr + (ch + 1) * 8);
+      xn0 = bf2f(u16(nxt[0]));
+      xn1 = bf2f(u16(nxt[1]));
+    }
+    q[10] = xn0 * xn0;
+    q[11] = xn1 * xn1;
+    s16x8 yv, sbv, snbv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float s = k + an * (q[j] + q[j + 1] + q[j + 2] + q[j + 3] + q[j + 4]);
+      float ls = fast_log2(s);
+      float sb = fast_exp2(-beta * ls);
+      float snb = fast_exp2(-(1.f + beta) * ls);
+      yv[j] = short(f2bf(xc[j] * sb));
+      sbv[j] = short(f2bf(sb));
+      snbv[j] = short(f2bf(snb));
+    }
+    *reinterpret_cast<s16x8*>(yr + ch * 8) = yv;
+    *reinterpret_cast<s16x8*>(sbr + ch * 8) = sbv;
+    *reinterpret_cast<s16x8*>(snbr + ch * 8) = snbv;
+    q[0] = q[8]; q[1] = q[9];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      xc[j] = (j < 2) ? ((j == 0) ? xn0 : xn1) : 0.f;
+    if (ch + 1 < nchunks) {
+#pragma unroll
+      for (int j = 2; j < 8; ++j) xc[j] = bf2f(u16(nxt[j]));
+    }
+  }
+}
+
+std::vector<at::Tensor> lrn_fwd(const at::Tensor& x, int64_t size,
+                                double alpha, double beta, double k) {
+  check_bf16(x, "x");
+  TORCH_CHECK(size == 5, "LRN kernel is specialized for size=5");
+  int C = x.size(-1);
+  TORCH_CHECK(C % 8 == 0, "LRN needs C % 8 == 0");
+  i64 npos = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto sb = at::empty_like(x);
+  auto snb = at::empty_like(x);
+  hipLaunchKernelGGL(lrn_fwd_kernel, dim3(int((npos + 255) / 256)), dim3(256),
+                     0, cur_stream(), bfp(x), bfp_mut(y), bfp_mut(sb),
+                     bfp_mut(snb), npos, C, float(alpha / size), float(k),
+                     float(beta));
+  return {y, sb, snb};
+}
+
+// ------------------------------------------------------------- LRN bwd
+__global__ void lrn_bwd_kernel(const u16* __restrict__ dy,
+                               const u16* __restrict__ x,
+                               const u16* __restrict__ sb,
+                               const u16* __restrict__ snb,
+                               u16* __restrict__ dx, i64 npos, int C,
+                               float coef) {  // coef = 2*alpha*beta/size
+  i64 pos = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (pos >= npos) return;
+  const u16* dyr = dy + pos * C;
+  const u16* xr = x + pos * C;
+  const u16* sbr = sb + pos * C;
+  const u16* snbr = snb + pos * C;
+  u16* dxr = dx + pos * C;
+
+  float t[12];  // t = dy * x * snb, sliding window like fwd
+  t[0] = t[1] = 0.f;
+  float xc[8], dc[8], sbc[8];
+  const int nchunks = C / 8;
+  // preload chunk 0
+  {
+    s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
+    s16x8 dv = *reinterpret_cast<const s16x8*>(dyr);
+    s16x8 sv = *reinterpret_cast<const s16x8*>(snbr);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      xc[j] = bf2f(u16(xv[j]));
+      dc[j] = bf2f(u16(dv[j]));
+      sbc[j] = bf2f(u16(sv[j]));  // snb values staged through sbc temp
+    }
+  }
+  for (int ch = 0; ch < nchunks; ++ch) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) t[2 + j] = dc[j] * xc[j] * sbc[j];
+    float tn0 = 0.f, tn1 = 0.f;
+    s16x8 xv{}, dv{}, sv{};
+    if (ch + 1 < nchunks) {
+      xv = *reinterpret_cast<const s16x8*>(xr + (ch + 1) * 8);
+      dv = *reinterpret_cast<const s16x8*>(dyr + (ch + 1) * 8);
+      sv = *reinterpret_cast<const s16x8*>(snbr + (ch + 1) * 8);
+      tn0 = bf2f(u16(dv[0])) * bf2f(u16(xv[0])) * bf2f(u16(sv[0]));
+      tn1 = bf2f(u16(dv[1])) * bf2f(u16(xv[1])) * bf2f(u16(sv[1]));
+    }
+    t[10] = tn0; t[11] = tn1;
+    s16x8 sbv = *reinterpret_cast<const s16x8*>(sbr + ch * 8);
+    s16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float wsum = t[j] + t[j + 1] + t[j + 2] + t[j + 3] + t[j + 4];
+      float v = dc[j] * bf2f(u16(sbv[j])) - coef * xc[j] * wsum;
+      out[j] = short(f2bf(v));
+    }
+    *reinterpret_cast<s16x8*>(dxr + ch * 8) = out;
+    t[0] = t[8]; t[1] = t[9];
+    if (ch + 1 < nchunks) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        xc[j] = bf2f(u16(xv[j]));
+        dc[j] = bf2f(u16(dv[j]));
+        sbc[j] = bf2f(u16(sv[j]));
+      }
+    }
+  }
+}
+
+at::Tensor lrn_bwd(const at::Tensor& dy, const at::Tensor& x,
+                   const at::Tensor& sb, const at::Tensor& snb, int64_t size,
+                   double alpha, double beta) {
+  check_bf16(dy, "dy");
+  int C = dy.size(-1);
+  i64 npos = dy.numel() / C;
+  auto dx = at::empty_like(dy);
+  hipLaunchKernelGGL(lrn_bwd_kernel, dim3(int((npos + 255) / 256)), dim3(256),
+                     0, cur_stream(), bfp(dy), bfp(x), bfp(sb), bfp(snb),
+                     bfp_mut(dx), npos, C,
+                     float(2.0 * alpha * beta / size));
+  return dx;
+}
+
+// ----------------------------------------------------------- maxpool fwd
+// thread per output element; saves u8 window-local argmax for backward.
+__global__ void maxpool_fwd_kernel(const u16* __restrict__ x,
+                                   u16* __restrict__ y,
+                                   unsigned char* __restrict__ idx, int N,
+                                   int H, int W, int C, int OH, int OW,
+                                   int kernel, int stride, int pl) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  i64 total = (i64)N * OH * OW * C;
+  if (id >= total) return;
+  int c = int(id % C);
+  i64 r = id / C;
+  int ow = int(r % OW); r /= OW;
+  int oh = int(r % OH); r /= OH;
+  int n = int(r);
+  int ih0 = oh * stride - pl, iw0 = ow * stride - pl;
+  float best = -3.4e38f;
+  int bi = 0;
+  for (int dh = 0; dh < kernel; ++dh) {
+    int ih = ih0 + dh;
+    if ((unsigned)ih >= (unsigned)H) continue;
+    for (int dw = 0; dw < kernel; ++dw) {
+      int iw = iw0 + dw;
+      if ((unsigned)iw >= (unsigned)W) continue;
+      float v = bf2f(x[(((i64)n * H + ih) * W + iw) * C + c]);
+      if (v > best) { best = v; bi = dh * kernel + dw; }
+    }
+  }
+  y[id] = f2bf(best);
+  idx[id] = (unsigned char)bi;
+}
+
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kernel,
+                                    int64_t stride, int64_t pl, int64_t pr) {
+  check_bf16(x, "x");
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int OH = (H + int(pl) + int(pr) - int(kernel)) / int(stride) + 1;
+  int OW = OH;  // square inputs throughout; generalize if needed
+  auto y = at::empty({N, OH, OW, C}, x.options());
+  auto idx = at::empty({N, OH, OW, C}, x.options().dtype(at::kByte));
+  i64 total = (i64)N * OH * OW * C;
+  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(int((total + 255) / 256)),
+                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(y),
+                     idx.data_ptr<unsigned char>(), N, H, W, C, OH, OW,
+                     int(kernel), int(stride), int(pl));
+  return {y, idx};
+}
+
+// ----------------------------------------------------------- maxpool bwd
+// gather form (no atomics): each dx element scans its covering windows
+// (3x3 s2 windows overlap).
+__global__ void maxpool_bwd_kernel(const u16* __restrict__ dy,
+                                   const unsigned char* __restrict__ idx,
+                                   u16* __restrict__ dx, int N, int H, int W,
+                                   int C, int OH, int OW, int kernel,
+                                   int stride, int pl) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  i64 total = (i64)N * H * W * C;
+  if (id >= total) return;
+  int c = int(id % C);
+  i64 r = id / C;
+  int iw = int(r % W); r /= W;
+  int ih = int(r % H); r /= H;
+  int n = int(r);
+  float acc = 0.f;
+  // windows (oh,ow) with oh*stride - pl <= ih < oh*stride - pl + kernel
+  int oh_hi = (ih + pl) / stride;
+  int oh_lo = (ih + pl - kernel) / stride + (((ih + pl - kernel) % stride) ? 1 : 0);
+  if (ih + pl - kernel < 0) oh_lo = 0;
+  int ow_hi = (iw + pl) / stride;
+  int ow_lo = (iw + pl - kernel) / stride + (((iw + pl - kernel) % stride) ? 1 : 0);
+  if (iw + pl - kernel < 0) ow_lo = 0;
+  for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+    if ((unsigned)oh >= (unsigned)OH) continue;
+    int dh = ih - (oh * stride - pl);
+    if ((unsigned)dh >= (unsigned)kernel) continue;
+    for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+      if ((unsigned)ow >= (unsigned)OW) continue;
+      int dw = iw - (ow * stride - pl);
+      if ((unsigned)dw >= (unsigned)kernel) continue;
+      i64 o = (((i64)n * OH + oh) * OW + ow) * C + c;
+      if (idx[o] == dh * kernel + dw) acc += bf2f(dy[o]);
+    }
+  }
+  dx[id] = f2bf(acc);
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx,
+                       int64_t H_, int64_t W_, int64_t kernel, int64_t stride,
+                       int64_t pl, int64_t pr) {
+  check_bf16(dy, "dy");
+  int N = dy.size(0), OH = dy.size(1), OW = dy.size(2), C = dy.size(3);
+  int H = int(H_), W = int(W_);
+  auto dx = at::empty({N, H, W, C}, dy.options());
+  i64 total = (i64)N * H * W * C;
+  hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(int((total + 255) / 256)),
+                     dim3(256), 0, cur_stream(), bfp(dy),
+                     idx.data_ptr<unsigned char>(), bfp_mut(dx), N, H, W, C,
+                     OH, OW, int(kernel), int(stride), int(pl));
+  return dx;
+}
+
+}  // namespace dvgg

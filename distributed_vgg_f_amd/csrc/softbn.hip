@@ -1,0 +1,247 @@
+// Fused softmax + cross-entropy and training-mode
+// BatchNorm fwd/bwd (K17, ResNet-50 config). NHWC; channels innermost.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace dvgg {
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+static inline const u16* bfp(const at::Tensor& t) {
+  return reinterpret_cast<const u16*>(t.data_ptr<at::BFloat16>());
+}
+static inline u16* bfp_mut(at::Tensor& t) {
+  return reinterpret_cast<u16*>(t.data_ptr<at::BFloat16>());
+}
+static void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 &&
+              t.is_contiguous(), name, " must be contiguous GPU bf16");
+}
+
+DEV float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, 64));
+  return __shfl(v, 0, 64);
+}
+DEV float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return __shfl(v, 0, 64);
+}
+
+// -------------------------------------------------- softmax + xent (K13)
+// One 256-thread block per row; computes per-row loss (fp32) AND
+// dlogits = softmax - onehot (bf16) in a single pass structure.
+__global__ void softmax_xent_kernel(const u16* __restrict__ logits,
+                                    const long* __restrict__ labels,
+                                    float* __restrict__ loss,
+                                    u16* __restrict__ dlogits, int M, int C) {
+  __shared__ float red[8];
+  int m = blockIdx.x;
+  if (m >= M) return;
+  const u16* row = logits + (i64)m * C;
+  u16* drow = dlogits + (i64)m * C;
+  int t = threadIdx.x, lane = t & 63, wave = t >> 6;
+
+  float mx = -3.4e38f;
+  for (int c = t; c < C; c += blockDim.x) mx = fmaxf(mx, bf2f(row[c]));
+  mx = wave_reduce_max(mx);
+  if (lane == 0) red[wave] = mx;
+  __syncthreads();
+  mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+
+  float sum = 0.f;
+  for (int c = t; c < C; c += blockDim.x) sum += __expf(bf2f(row[c]) - mx);
+  sum = wave_reduce_sum(sum);
+  if (lane == 0) red[4 + wave] = sum;
+  __syncthreads();
+  sum = red[4] + red[5] + red[6] + red[7];
+  float inv = 1.f / sum;
+
+  long lbl = labels[m];
+  for (int c = t; c < C; c += blockDim.x) {
+    float p = __expf(bf2f(row[c]) - mx) * inv;
+    drow[c] = f2bf(p - (c == lbl ? 1.f : 0.f));
+  }
+  if (t == 0) loss[m] = logf(sum) + mx - bf2f(row[lbl]);
+}
+
+std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor& logits,
+                                         const at::Tensor& labels) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(labels.scalar_type() == at::kLong && labels.is_cuda());
+  int M = logits.size(0), C = logits.size(1);
+  auto loss = at::empty({M}, logits.options().dtype(at::kFloat));
+  auto dlogits = at::empty_like(logits);
+  hipLaunchKernelGGL(softmax_xent_kernel, dim3(M), dim3(256), 0, cur_stream(),
+                     bfp(logits), labels.data_ptr<long>(),
+                     loss.data_ptr<float>(), bfp_mut(dlogits), M, C);
+  return {loss, dlogits};
+}
+
+// --------------------------------------------------------- BN fwd (K17)
+// Pass 1: per-channel sum / sumsq via block partials + fp32 atomics.
+// Pass 2: normalize (+affine, optional fused ReLU); saves mean/rstd fp32.
+__global__ void bn_reduce_kernel(const u16* __restrict__ x,
+                                 float* __restrict__ sum,
+                                 float* __restrict__ sumsq, i64 R, int C,
+                                 int rows_per_block) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  i64 r0 = (i64)blockIdx.y * rows_per_block;
+  i64 r1 = min(R, r0 + rows_per_block);
+  float s = 0.f, q = 0.f;
+  for (i64 r = r0; r < r1; ++r) {
+    float v = bf2f(x[r * C + c]);
+    s += v; q += v * v;
+  }
+  atomicAdd(&sum[c], s);
+  atomicAdd(&sumsq[c], q);
+}
+
+__global__ void bn_stats_kernel(const float* __restrict__ sum,
+                                const float* __restrict__ sumsq,
+                                float* __restrict__ mean,
+                                float* __restrict__ rstd, i64 R, int C,
+                                float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = sum[c] / float(R);
+  float v = sumsq[c] / float(R) - m * m;
+  mean[c] = m;
+  rstd[c] = rsqrtf(fmaxf(v, 0.f) + eps);
+}
+
+template <bool RELU>
+__global__ void bn_apply_kernel(const u16* __restrict__ x,
+                                const u16* __restrict__ gamma,
+                                const u16* __restrict__ beta,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ rstd,
+                                u16* __restrict__ y, i64 total, int C) {
+  i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int c = int(i % C);
+  float v = (bf2f(x[i]) - mean[c]) * rstd[c] * bf2f(gamma[c]) + bf2f(beta[c]);
+  if (RELU) v = v > 0.f ? v : 0.f;
+  y[i] = f2bf(v);
+}
+
+std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                               const at::Tensor& beta, double eps, bool relu) {
+  check_bf16(x, "x"); check_bf16(gamma, "gamma"); check_bf16(beta, "beta");
+  int C = x.size(-1);
+  i64 R = x.numel() / C;
+  auto f32 = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, f32), sumsq = at::zeros({C}, f32);
+  auto mean = at::empty({C}, f32), rstd = at::empty({C}, f32);
+  auto y = at::empty_like(x);
+  int rpb = 4096;
+  auto s = cur_stream();
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
+  hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
+                     sum.data_ptr<float>(), sumsq.data_ptr<float>(), R, C, rpb);
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(ceil_div(C, 256)), dim3(256), 0, s,
+                     sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), R, C,
+                     float(eps));
+  i64 total = x.numel();
+  if (relu)
+    hipLaunchKernelGGL(bn_apply_kernel<true>, dim3(int((total + 255) / 256)),
+                       dim3(256), 0, s, bfp(x), bfp(gamma), bfp(beta),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       bfp_mut(y), total, C);
+  else
+    hipLaunchKernelGGL(bn_apply_kernel<false>, dim3(int((total + 255) / 256)),
+                       dim3(256), 0, s, bfp(x), bfp(gamma), bfp(beta),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       bfp_mut(y), total, C);
+  return {y, mean, rstd};
+}
+
+// --------------------------------------------------------- BN bwd (K17)
+// reduce dy and dy*xhat per channel -> dbeta/dgamma; then
+// dx = rstd*gamma * (dy - mean(dy) - xhat * mean(dy*xhat))
+__global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
+                                     const u16* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     float* __restrict__ sdy,
+                                     float* __restrict__ sdyx, i64 R, int C,
+                                     int rows_per_block) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  i64 r0 = (i64)blockIdx.y * rows_per_block;
+  i64 r1 = min(R, r0 + rows_per_block);
+  float a = 0.f, b = 0.f, m = mean[c], rs = rstd[c];
+  for (i64 r = r0; r < r1; ++r) {
+    float d = bf2f(dy[r * C + c]);
+    float xh = (bf2f(x[r * C + c]) - m) * rs;
+    a += d; b += d * xh;
+  }
+  atomicAdd(&sdy[c], a);
+  atomicAdd(&sdyx[c], b);
+}
+
+__global__ void bn_bwd_apply_kernel(const u16* __restrict__ dy,
+                                    const u16* __restrict__ x,
+                                    const u16* __restrict__ gamma,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ rstd,
+                                    const float* __restrict__ sdy,
+                                    const float* __restrict__ sdyx,
+                                    u16* __restrict__ dx, i64 total, int C,
+                                    float invR) {
+  i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int c = int(i % C);
+  float xh = (bf2f(x[i]) - mean[c]) * rstd[c];
+  float v = bf2f(dy[i]) - sdy[c] * invR - xh * sdyx[c] * invR;
+  dx[i] = f2bf(v * rstd[c] * bf2f(gamma[c]));
+}
+
+__global__ void bn_param_grad_kernel(const float* __restrict__ sdy,
+                                     const float* __restrict__ sdyx,
+                                     u16* __restrict__ dgamma,
+                                     u16* __restrict__ dbeta, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  dgamma[c] = f2bf(sdyx[c]);
+  dbeta[c] = f2bf(sdy[c]);
+}
+
+std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
+                               const at::Tensor& gamma, const at::Tensor& mean,
+                               const at::Tensor& rstd) {
+  check_bf16(dy, "dy"); check_bf16(x, "x");
+  int C = dy.size(-1);
+  i64 R = dy.numel() / C;
+  auto f32 = dy.options().dtype(at::kFloat);
+  auto sdy = at::zeros({C}, f32), sdyx = at::zeros({C}, f32);
+  auto dx = at::empty_like(dy);
+  auto dgamma = at::empty({C}, dy.options());
+  auto dbeta = at::empty({C}, dy.options());
+  int rpb = 4096;
+  auto s = cur_stream();
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, rgrid, dim3(256), 0, s, bfp(dy),
+                     bfp(x), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(), R, C, rpb);
+  i64 total = dy.numel();
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(int((total + 255) / 256)),
+                     dim3(256), 0, s, bfp(dy), bfp(x), bfp(gamma),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                     bfp_mut(dx), total, C, float(1.0 / double(R)));
+  hipLaunchKernelGGL(bn_param_grad_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     0, s, sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                     bfp_mut(dgamma), bfp_mut(dbeta), C);
+  return {dx, dgamma, dbeta};
+}
+
+}  // namespace dvgg

@@ -41,7 +41,8 @@ class Trainer:
 
         torch.manual_seed(cfg.seed)          # same init on every rank; C4
         self.model = build_model(cfg)        # broadcast still run for safety
-        self.model.train()
+        self.model.to(self.device)           # buffers (BN stats); params get
+        self.model.train()                   # re-homed into arenas below
 
         compute_dtype = torch.bfloat16 if (cfg.precision == "bf16" and
                                            self.device.type == "cuda") \

@@ -85,20 +85,21 @@ class LinearFunction(torch.autograd.Function):
 
 
 class LRNFunction(torch.autograd.Function):
-    """K4/K5. Saves the fp32 scale tensor (k + alpha/n * window sum) for bwd."""
+    """K4/K5. Fwd saves sb = s^-beta and snb = s^-(1+beta) (bf16) so the
+    backward pass needs no transcendentals (see csrc/lrnpool.hip)."""
 
     @staticmethod
     def forward(ctx, x, size: int, alpha: float, beta: float, k: float):
-        y, scale = _ext().lrn_fwd(x, size, alpha, beta, k)
-        ctx.save_for_backward(x, y, scale)
-        ctx.params = (size, alpha, beta, k)
+        y, sb, snb = _ext().lrn_fwd(x, size, alpha, beta, k)
+        ctx.save_for_backward(x, sb, snb)
+        ctx.params = (size, alpha, beta)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, scale = ctx.saved_tensors
-        size, alpha, beta, k = ctx.params
-        dx = _ext().lrn_bwd(dy.contiguous(), x, y, scale, size, alpha, beta)
+        x, sb, snb = ctx.saved_tensors
+        size, alpha, beta = ctx.params
+        dx = _ext().lrn_bwd(dy.contiguous(), x, sb, snb, size, alpha, beta)
         return dx, None, None, None, None
 
 

@@ -1,0 +1,81 @@
+"""T-model on MI355X: whole-model fwd+bwd through the HIP kernel path vs the
+CPU fp32 reference; trainer loss decreases; native .so actually loaded."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from distributed_vgg_f_amd.config import Config
+from distributed_vgg_f_amd.engine.trainer import Trainer
+from distributed_vgg_f_amd.models import build_model
+
+DEV = torch.device("cuda", 0)
+
+
+def test_native_so_is_intree():
+    from distributed_vgg_f_amd.ops import native
+    path = native.ext().__file__
+    assert "distributed_vgg_f_amd" in path, path
+
+
+def test_vggf_fwd_bwd_matches_cpu():
+    cfg = Config(model="vggf", batch_size=4, seed=5)
+    torch.manual_seed(cfg.seed)
+    model = build_model(cfg)  # fp32 CPU init
+    x = torch.randn(4, 224, 224, 3) * 0.5
+    labels = torch.tensor([1, 2, 3, 4])
+
+    # CPU fp32 reference pass
+    logits_ref = model(x)
+    import torch.nn.functional as F
+    loss_ref = F.cross_entropy(logits_ref, labels, reduction="sum")
+    model.zero_grad()
+    loss_ref.backward()
+    ref_grads = {n: p.grad.clone() for n, p in model.named_parameters()}
+
+    # GPU bf16 pass through HIP kernels (same weights, bf16-cast)
+    gm = build_model(cfg)
+    gm.load_state_dict(model.state_dict())
+    gm = gm.to(DEV).bfloat16()
+    from distributed_vgg_f_amd import ops
+    logits = gm(x.bfloat16().to(DEV))
+    rel = float((logits.float().cpu() - logits_ref).abs().max() /
+                logits_ref.abs().max().clamp(min=1e-3))
+    assert rel < 0.12, f"logits rel err {rel}"
+    loss = ops.softmax_xent(logits, labels.to(DEV)).sum()
+    loss.backward()
+    assert abs(float(loss) - float(loss_ref)) / max(float(loss_ref), 1) < 0.1
+
+    for n, p in gm.named_parameters():
+        g = p.grad.float().cpu()
+        r = ref_grads[n]
+        scale = max(float(r.abs().max()), 1e-4)
+        err = float((g - r).abs().max())
+        # bf16 fwd + bf16 grads through 8 layers: loose band, catches layout
+        # bugs (those show up as ~100% error)
+        assert err <= 0.35 * scale, f"{n}: {err:.4g} vs {scale:.4g}"
+
+
+def test_trainer_gpu_loss_decreases():
+    cfg = Config(model="vggf", batch_size=16, train_steps=12, lr=0.02,
+                 log_every=0, precision="bf16")
+    t = Trainer(cfg, device=DEV)
+    losses = [float(t.train_step()) / cfg.batch_size for _ in range(12)]
+    assert losses[-1] < losses[0], losses
+    assert all(l == l for l in losses), losses  # no NaN
+
+
+def test_trainer_gpu_vgg16_resnet50_step():
+    for name in ("vgg16", "resnet50"):
+        cfg = Config(model=name, batch_size=4, train_steps=1, log_every=0)
+        t = Trainer(cfg, device=DEV)
+        loss = float(t.train_step()) / cfg.batch_size
+        assert loss == loss, name
+
+
+def test_eval_runs():
+    cfg = Config(model="vggf", batch_size=8, eval_steps=2, log_every=0)
+    t = Trainer(cfg, device=DEV)
+    res = t.evaluate(2)
+    assert 0.0 <= res["accuracy"] <= 1.0

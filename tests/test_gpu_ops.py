@@ -1,0 +1,243 @@
+"""T-unit on MI355X: every HIP kernel vs the CPU fp32 reference
+(SURVEY.md section 4.2). Inputs are bf16-rounded before the reference runs,
+so tolerances cover only accumulation-order + bf16 output rounding.
+
+Transpose-detection rule (cdna_hip_programming.md section 5.4 rule 16):
+all operands are random (asymmetric) tensors.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from distributed_vgg_f_amd.ops import native
+    ext = native.ext()
+from distributed_vgg_f_amd.ops import cpu_ref
+
+DEV = "cuda:0"
+
+
+def rnd(*shape, scale=1.0, seed=None):
+    if seed is not None:
+        torch.manual_seed(seed)
+    return (torch.randn(*shape) * scale).bfloat16()
+
+
+def check(got, ref, tol=0.05, name=""):
+    got = got.float().cpu()
+    ref = ref.float()
+    assert got.shape == ref.shape, f"{name}: {got.shape} vs {ref.shape}"
+    scale = max(float(ref.abs().max()), 1e-3)
+    err = float((got - ref).abs().max())
+    assert err <= tol * scale, f"{name}: max err {err:.4g} vs scale {scale:.4g}"
+
+
+# --------------------------------------------------------------- linear
+def test_linear_fwd_asymmetric():
+    torch.manual_seed(0)
+    x, w, b = rnd(256, 512), rnd(512, 384, scale=0.05), rnd(384)
+    y = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), True)
+    ref = cpu_ref.linear(x.float(), w.float(), b.float(), relu=True)
+    check(y, ref, 0.05, "linear_fwd")
+
+
+def test_linear_fwd_fc6_shape_splitk():
+    torch.manual_seed(1)
+    x, w, b = rnd(256, 9216, scale=0.1), rnd(9216, 4096, scale=0.02), rnd(4096)
+    y = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), False)
+    ref = cpu_ref.linear(x.float(), w.float(), b.float())
+    check(y, ref, 0.05, "fc6")
+
+
+def test_linear_fwd_fc8_nonmult128():
+    torch.manual_seed(2)
+    x, w, b = rnd(12, 4096, scale=0.1), rnd(4096, 1000, scale=0.02), rnd(1000)
+    y = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), False)
+    check(y, cpu_ref.linear(x.float(), w.float(), b.float()), 0.05, "fc8")
+
+
+def test_linear_dgrad_wgrad():
+    torch.manual_seed(3)
+    x, w = rnd(256, 1024, scale=0.1), rnd(1024, 512, scale=0.05)
+    dy = rnd(256, 512, scale=0.1)
+    dx = ext.linear_dgrad(dy.to(DEV), w.to(DEV))
+    dw = ext.linear_wgrad(x.to(DEV), dy.to(DEV))
+    check(dx, dy.float() @ w.float().t(), 0.05, "linear_dgrad")
+    check(dw, x.float().t() @ dy.float(), 0.05, "linear_wgrad")
+
+
+# ----------------------------------------------------------------- conv
+CONV_CASES = [
+    # (N, H, W, Cin, Cout, k, stride, pad, name)
+    (2, 27, 27, 64, 256, 5, 1, 2, "conv2"),
+    (2, 13, 13, 256, 256, 3, 1, 1, "conv3"),
+    (2, 224, 224, 3, 64, 11, 4, 0, "conv1-im2col"),
+    (2, 16, 16, 64, 128, 1, 1, 0, "conv1x1"),
+    (2, 16, 16, 64, 128, 3, 2, 1, "conv3x3s2"),
+    (2, 16, 16, 64, 128, 1, 2, 0, "conv1x1s2"),
+]
+
+
+@pytest.mark.parametrize("N,H,W,Ci,Co,k,s,p,name", CONV_CASES)
+def test_conv_fwd(N, H, W, Ci, Co, k, s, p, name):
+    torch.manual_seed(10)
+    x = rnd(N, H, W, Ci, scale=0.5)
+    w = rnd(k, k, Ci, Co, scale=1.0 / math.sqrt(k * k * Ci))
+    b = rnd(Co, scale=0.1)
+    y = ext.conv2d_fwd(x.to(DEV), w.to(DEV), b.to(DEV), s, p, True)
+    ref = cpu_ref.conv2d(x.float(), w.float(), b.float(), s, p, relu=True)
+    check(y, ref, 0.05, f"conv_fwd/{name}")
+
+
+@pytest.mark.parametrize("N,H,W,Ci,Co,k,s,p,name",
+                         [c for c in CONV_CASES if c[3] % 8 == 0])
+def test_conv_dgrad(N, H, W, Ci, Co, k, s, p, name):
+    torch.manual_seed(11)
+    w = rnd(k, k, Ci, Co, scale=1.0 / math.sqrt(k * k * Ci))
+    OH = (H + 2 * p - k) // s + 1
+    dy = rnd(N, OH, OH, Co, scale=0.5)
+    dx = ext.conv2d_dgrad(dy.to(DEV), w.to(DEV), H, W, s, p)
+    # reference via autograd on the CPU composition
+    x32 = torch.zeros(N, H, W, Ci, requires_grad=True)
+    y = cpu_ref.conv2d(x32, w.float(), None, s, p)
+    y.backward(dy.float())
+    check(dx, x32.grad, 0.05, f"conv_dgrad/{name}")
+
+
+@pytest.mark.parametrize("N,H,W,Ci,Co,k,s,p,name", CONV_CASES)
+def test_conv_wgrad(N, H, W, Ci, Co, k, s, p, name):
+    torch.manual_seed(12)
+    x = rnd(N, H, W, Ci, scale=0.5)
+    OH = (H + 2 * p - k) // s + 1
+    dy = rnd(N, OH, OH, Co, scale=0.5)
+    dw = ext.conv2d_wgrad(dy.to(DEV), x.to(DEV), k, k, s, p)
+    w32 = torch.zeros(k, k, Ci, Co, requires_grad=True)
+    y = cpu_ref.conv2d(x.float(), w32, None, s, p)
+    y.backward(dy.float())
+    check(dw, w32.grad, 0.06, f"conv_wgrad/{name}")
+
+
+# ------------------------------------------------------------- LRN/pool
+def test_lrn_fwd_bwd():
+    torch.manual_seed(20)
+    x = rnd(2, 9, 9, 64, scale=2.0)
+    y, sb, snb = ext.lrn_fwd(x.to(DEV), 5, 1e-4, 0.75, 2.0)
+    ref = cpu_ref.lrn(x.float(), 5, 1e-4, 0.75, 2.0)
+    check(y, ref, 0.03, "lrn_fwd")
+    dy = rnd(2, 9, 9, 64, seed=21)
+    dx = ext.lrn_bwd(dy.to(DEV), x.to(DEV), sb, snb, 5, 1e-4, 0.75)
+    x32 = x.float().requires_grad_(True)
+    cpu_ref.lrn(x32, 5, 1e-4, 0.75, 2.0).backward(dy.float())
+    check(dx, x32.grad, 0.05, "lrn_bwd")
+
+
+def test_lrn_large_alpha():
+    # make the normalizer actually bite (s >> k) to exercise the poly path
+    torch.manual_seed(22)
+    x = rnd(1, 5, 5, 128, scale=4.0)
+    y, _, _ = ext.lrn_fwd(x.to(DEV), 5, 0.5, 0.75, 1.0)
+    check(y, cpu_ref.lrn(x.float(), 5, 0.5, 0.75, 1.0), 0.04, "lrn_bite")
+
+
+def test_maxpool_fwd_bwd_asympad():
+    torch.manual_seed(23)
+    x = rnd(2, 54, 54, 64)
+    y, idx = ext.maxpool_fwd(x.to(DEV), 3, 2, 0, 1)
+    ref = cpu_ref.maxpool(x.float(), 3, 2, (0, 1))
+    check(y, ref, 0.01, "maxpool_fwd")
+    dy = rnd(2, 27, 27, 64, seed=24)
+    dx = ext.maxpool_bwd(dy.to(DEV), idx, 54, 54, 3, 2, 0, 1)
+    x32 = x.float().requires_grad_(True)
+    cpu_ref.maxpool(x32, 3, 2, (0, 1)).backward(dy.float())
+    check(dx, x32.grad, 0.01, "maxpool_bwd")
+
+
+def test_maxpool_nopad():
+    torch.manual_seed(25)
+    x = rnd(2, 27, 27, 32)
+    y, idx = ext.maxpool_fwd(x.to(DEV), 3, 2, 0, 0)
+    check(y, cpu_ref.maxpool(x.float(), 3, 2, (0, 0)), 0.01, "maxpool_np")
+
+
+# ------------------------------------------------------- softmax / misc
+def test_softmax_xent():
+    torch.manual_seed(30)
+    logits = rnd(64, 1000, scale=2.0)
+    labels = torch.randint(0, 1000, (64,))
+    loss, dlogits = ext.softmax_xent_fwd(logits.to(DEV), labels.to(DEV))
+    ref_loss, ref_d = cpu_ref.softmax_xent(logits.float(), labels)
+    check(loss, ref_loss, 0.01, "xent_loss")
+    check(dlogits, ref_d, 0.02, "xent_dlogits")
+
+
+def test_relu_bwd_add_relu_bias_grad():
+    torch.manual_seed(31)
+    y = rnd(1000), rnd(1000)
+    dy, yv = rnd(4, 250, seed=32), rnd(4, 250, seed=33)
+    got = ext.relu_bwd(dy.to(DEV), yv.to(DEV))
+    ref = dy.float() * (yv.float() > 0)
+    check(got, ref, 0.01, "relu_bwd")
+    a, b = rnd(4, 256, seed=34), rnd(4, 256, seed=35)
+    check(ext.add_relu_fwd(a.to(DEV), b.to(DEV)),
+          torch.relu(a.float() + b.float()), 0.02, "add_relu")
+    dyb = rnd(512, 64, seed=36)
+    check(ext.bias_grad(dyb.to(DEV)), dyb.float().sum(0), 0.02, "bias_grad")
+
+
+def test_dropout_stats_and_bwd():
+    torch.manual_seed(37)
+    x = torch.ones(100000).bfloat16().to(DEV)
+    y, mask = ext.dropout_fwd(x, 0.3, 1234, 7)
+    keep = float(mask.float().mean())
+    assert abs(keep - 0.7) < 0.02, keep
+    # kept elements scaled by 1/(1-p)
+    got = y.float()
+    assert torch.allclose(got[mask.bool()],
+                          torch.full_like(got[mask.bool()], 1.0 / 0.7),
+                          atol=0.01)
+    dy = torch.ones_like(x)
+    dx = ext.dropout_bwd(dy, mask, 0.3)
+    assert torch.allclose(dx.float()[mask.bool()],
+                          torch.full_like(got[mask.bool()], 1.0 / 0.7),
+                          atol=0.01)
+    assert float(dx.float()[~mask.bool()].abs().sum()) == 0.0
+
+
+def test_sgd_step_matches_cpu():
+    torch.manual_seed(38)
+    n = 4096
+    w32 = torch.randn(n)
+    g = torch.randn(n).bfloat16()
+    m32 = torch.randn(n)
+    w32_d = w32.clone().to(DEV)
+    wc_d = w32.bfloat16().to(DEV)
+    m32_d = m32.clone().to(DEV)
+    ext.sgd_step(w32_d, wc_d, g.to(DEV), m32_d, 0.05, 0.9, 1e-4, 0.25)
+    wr, mr = w32.clone(), m32.clone()
+    cpu_ref.sgd_momentum(wr, g.float(), mr, 0.05, 0.9, 1e-4, 0.25)
+    check(w32_d.bfloat16(), wr, 0.01, "sgd_w")
+    check(m32_d.bfloat16(), mr, 0.01, "sgd_m")
+    check(wc_d, wr.bfloat16().float(), 0.01, "sgd_wc")
+
+
+def test_bn_fwd_bwd():
+    torch.manual_seed(39)
+    x = rnd(8, 7, 7, 64, scale=2.0)
+    g, b = rnd(64, seed=40), rnd(64, seed=41)
+    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, False)
+    ref = cpu_ref.batchnorm(x.float(), g.float(), b.float())
+    check(y, ref, 0.05, "bn_fwd")
+    dy = rnd(8, 7, 7, 64, seed=42)
+    dx, dgamma, dbeta = ext.bn_bwd(dy.to(DEV), x.to(DEV), g.to(DEV), mean,
+                                   rstd)
+    x32 = x.float().requires_grad_(True)
+    g32 = g.float().requires_grad_(True)
+    b32 = b.float().requires_grad_(True)
+    cpu_ref.batchnorm(x32, g32, b32).backward(dy.float())
+    check(dx, x32.grad, 0.06, "bn_dx")
+    check(dgamma, g32.grad, 0.06, "bn_dgamma")
+    check(dbeta, b32.grad, 0.06, "bn_dbeta")
