@@ -88,10 +88,28 @@ __global__ void nhwc2cnhw_kernel(const u16* __restrict__ in,
   }
 }
 
+// Cached, zeroed, slack-padded buffers for the wgrad transposes: the
+// interior is fully rewritten every call and the zero halo is never
+// written, so the (expensive at bs=256) zero-fill happens once per shape.
+static at::Tensor cached_zero_buffer(const std::string& key, i64 numel,
+                                     const at::TensorOptions& opt) {
+  static std::unordered_map<std::string, at::Tensor> cache;
+  auto it = cache.find(key);
+  if (it == cache.end()) {
+    it = cache.emplace(key, at::zeros({numel + 8}, opt)).first;
+  }
+  return it->second;
+}
+
 static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
-                                int co) {
+                                int co, char role) {
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
-  auto out = at::zeros({C, N, Hp, Wp}, x.options());
+  i64 numel = (i64)C * N * Hp * Wp;
+  char key[96];
+  snprintf(key, sizeof key, "%c%d_%d_%d_%d_%d_%d_%d", role, x.get_device(),
+           C, N, Hp, Wp, ro, co);
+  auto out = cached_zero_buffer(key, numel, x.options())
+      .narrow(0, 0, numel).view({C, N, Hp, Wp});
   dim3 grid(N * H, ceil_div(W, 32), ceil_div(C, 32));
   hipLaunchKernelGGL(nhwc2cnhw_kernel, grid, dim3(256), 0, cur_stream(),
                      bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
@@ -278,20 +296,21 @@ at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
 at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
   check_bf16(x, "x"); check_bf16(dy, "dy");
   int M = x.size(0), K = x.size(1), N = dy.size(1);
-  TORCH_CHECK(dy.size(0) == M && M % 8 == 0, "bad fc wgrad shape");
-  auto xT = transpose2d(x);    // [K][M]
-  auto dyT = transpose2d(dy);  // [N][M]
+  TORCH_CHECK(dy.size(0) == M, "bad fc wgrad shape");
+  int Mp = ceil8(M);           // zero-padded reduction rows contribute 0
+  auto xT = transpose2d(x, Mp);    // [K][Mp]
+  auto dyT = transpose2d(dy, Mp);  // [N][Mp]
   auto dw = at::empty({K, N}, x.options());
-  PlainGather ga{bfp(xT), K, M, M};
-  PlainGather gb{bfp(dyT), N, M, M};
+  PlainGather ga{bfp(xT), K, Mp, Mp};
+  PlainGather gb{bfp(dyT), N, Mp, Mp};
   int tiles = ceil_div(K, BM) * ceil_div(N, BN);
-  int sk = pick_splitk(tiles, M);
+  int sk = pick_splitk(tiles, Mp);
   if (sk > 1) {
     auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
-    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, K, N, M, sk);
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, K, N, Mp, sk);
     run_finalize(c32, dw, nullptr, 0, K, N);
   } else {
-    launch_gemm(ga, gb, &dw, nullptr, nullptr, 0, K, N, M, 1);
+    launch_gemm(ga, gb, &dw, nullptr, nullptr, 0, K, N, Mp, 1);
   }
   return dw;
 }
@@ -388,34 +407,15 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
 
-  if (Cin % 8 != 0) {
-    // first-layer path: GEMM over im2col^T
-    int M = N * OH * OW, Kp = ceil8(Kfull), Mp = ceil8(M);
-    auto AT = at::empty({Kp, Mp}, x.options());
-    i64 chunks = (i64)Kp * (Mp / 8);
-    hipLaunchKernelGGL((im2col_kernel<true>),
-                       dim3(int((chunks + 255) / 256)), dim3(256), 0,
-                       cur_stream(), bfp(x), bfp_mut(AT), M, Mp, Kfull, Kp, N,
-                       H, W, Cin, OH, OW, KH, KW, stride, pad);
-    auto dyT = transpose2d(dy.view({M, Cout}), Mp);  // [Cout][Mp]
-    PlainGather ga{bfp(AT), Kfull, Mp, Mp};
-    PlainGather gb{bfp(dyT), Cout, Mp, Mp};
-    int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
-    int sk = pick_splitk(tiles, Mp, 1024);
-    auto c32 = at::zeros({Kp, Cout}, x.options().dtype(at::kFloat));
-    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Mp,
-                std::max(sk, 2));
-    run_finalize(c32.narrow(0, 0, Kfull).contiguous(), dw2d, nullptr, 0,
-                 Kfull, Cout);
-    return dw;
-  }
-
+  // One generic path for every conv (any Cin, any stride): transpose x and
+  // dy once into gather-friendly padded layouts, then split-K implicit GEMM
+  // over them (conv1's Cin=3 included: the WgradX gather is element-wise).
   int OWp = ceil8(OW);
   int Kred = N * OH * OWp;
   int Hp = H + 2 * pad;
   int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
-  auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad);           // [Cin][N][Hp][Wp]
-  auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0);            // [Cout][N][OH][OWp]
+  auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');           // [Cin][N][Hp][Wp]
+  auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');            // [Cout][N][OH][OWp]
   WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride};
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);

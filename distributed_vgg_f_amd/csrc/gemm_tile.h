@@ -112,6 +112,25 @@ struct WgradXGather {
     const u16* base = xT + (((i64)ci * N + n) * Hp + (oh * stride + kh)) * Wp +
                       (ow * stride + kw);
     s16x8 v;
+    if (stride == 1) {
+      // vector staging despite the odd +kw element shift: read the aligned
+      // dword window and funnel-shift (buffer has +8 elem slack for the
+      // trailing read; halo is pre-zeroed so no predicates).
+      uintptr_t ap = (uintptr_t)base;
+      const u32* q = reinterpret_cast<const u32*>(ap & ~(uintptr_t)3);
+      u32 a0 = q[0], a1 = q[1], a2 = q[2], a3 = q[3];
+      u32* o = reinterpret_cast<u32*>(&v);
+      if (ap & 2) {
+        u32 a4 = q[4];
+        o[0] = (a0 >> 16) | (a1 << 16);
+        o[1] = (a1 >> 16) | (a2 << 16);
+        o[2] = (a2 >> 16) | (a3 << 16);
+        o[3] = (a3 >> 16) | (a4 << 16);
+      } else {
+        o[0] = a0; o[1] = a1; o[2] = a2; o[3] = a3;
+      }
+      return v;
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) v[j] = short(base[(i64)j * stride]);
     return v;

@@ -264,35 +264,44 @@ __global__ void maxpool_bwd_kernel(const u16* __restrict__ dy,
                                    u16* __restrict__ dx, int N, int H, int W,
                                    int C, int OH, int OW, int kernel,
                                    int stride, int pl) {
+  // 8 channels per thread (C % 8 == 0 path), vectorized loads of dy / idx
   i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  i64 total = (i64)N * H * W * C;
+  i64 total = (i64)N * H * W * (C / 8);
   if (id >= total) return;
-  int c = int(id % C);
-  i64 r = id / C;
+  int c8 = int(id % (C / 8)) * 8;
+  i64 r = id / (C / 8);
   int iw = int(r % W); r /= W;
   int ih = int(r % H); r /= H;
   int n = int(r);
-  float acc = 0.f;
-  // windows (oh,ow) with oh*stride - pl <= ih < oh*stride - pl + kernel
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
   int oh_hi = (ih + pl) / stride;
-  int oh_lo = (ih + pl - kernel) / stride + (((ih + pl - kernel) % stride) ? 1 : 0);
-  if (ih + pl - kernel < 0) oh_lo = 0;
   int ow_hi = (iw + pl) / stride;
-  int ow_lo = (iw + pl - kernel) / stride + (((iw + pl - kernel) % stride) ? 1 : 0);
-  if (iw + pl - kernel < 0) ow_lo = 0;
-  for (int oh = oh_lo; oh <= oh_hi; ++oh) {
-    if ((unsigned)oh >= (unsigned)OH) continue;
+  for (int oh = oh_hi; oh >= 0 && oh * stride > ih + pl - kernel; --oh) {
+    if (oh >= OH) continue;
     int dh = ih - (oh * stride - pl);
-    if ((unsigned)dh >= (unsigned)kernel) continue;
-    for (int ow = ow_lo; ow <= ow_hi; ++ow) {
-      if ((unsigned)ow >= (unsigned)OW) continue;
+    for (int ow = ow_hi; ow >= 0 && ow * stride > iw + pl - kernel; --ow) {
+      if (ow >= OW) continue;
       int dw = iw - (ow * stride - pl);
-      if ((unsigned)dw >= (unsigned)kernel) continue;
-      i64 o = (((i64)n * OH + oh) * OW + ow) * C + c;
-      if (idx[o] == dh * kernel + dw) acc += bf2f(dy[o]);
+      i64 o = (((i64)n * OH + oh) * OW + ow) * C + c8;
+      unsigned char want = (unsigned char)(dh * kernel + dw);
+      // idx 8 bytes + dy 16 bytes, vector loads
+      const unsigned char* ip = idx + o;
+      u32 i0 = *reinterpret_cast<const u32*>(ip);
+      u32 i1 = *reinterpret_cast<const u32*>(ip + 4);
+      s16x8 d = *reinterpret_cast<const s16x8*>(dy + o);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned char iv = (unsigned char)(((j < 4 ? i0 : i1) >> ((j & 3) * 8)) & 0xff);
+        if (iv == want) acc[j] += bf2f(u16(d[j]));
+      }
     }
   }
-  dx[id] = f2bf(acc);
+  s16x8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = short(f2bf(acc[j]));
+  *reinterpret_cast<s16x8*>(dx + (((i64)n * H + ih) * W + iw) * C + c8) = out;
 }
 
 at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx,
@@ -301,8 +310,9 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx,
   check_bf16(dy, "dy");
   int N = dy.size(0), OH = dy.size(1), OW = dy.size(2), C = dy.size(3);
   int H = int(H_), W = int(W_);
+  TORCH_CHECK(C % 8 == 0, "maxpool bwd needs C % 8 == 0");
   auto dx = at::empty({N, H, W, C}, dy.options());
-  i64 total = (i64)N * H * W * C;
+  i64 total = (i64)N * H * W * (C / 8);
   hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(int((total + 255) / 256)),
                      dim3(256), 0, cur_stream(), bfp(dy),
                      idx.data_ptr<unsigned char>(), bfp_mut(dx), N, H, W, C,

@@ -120,7 +120,7 @@ at::Tensor bias_grad(const at::Tensor& dy) {
   int C = dy.size(-1);
   i64 R = dy.numel() / C;
   auto db32 = at::zeros({C}, dy.options().dtype(at::kFloat));
-  int rpb = 4096;
+  int rpb = 256;  // small row chunks: fill 256 CUs even at C=64
   dim3 grid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0, cur_stream(),
                      bfp(dy), db32.data_ptr<float>(), R, C, rpb);

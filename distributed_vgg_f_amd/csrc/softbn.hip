@@ -141,7 +141,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   auto sum = at::zeros({C}, f32), sumsq = at::zeros({C}, f32);
   auto mean = at::empty({C}, f32), rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
-  int rpb = 4096;
+  int rpb = 512;
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
@@ -226,7 +226,7 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dx = at::empty_like(dy);
   auto dgamma = at::empty({C}, dy.options());
   auto dbeta = at::empty({C}, dy.options());
-  int rpb = 4096;
+  int rpb = 512;
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, rgrid, dim3(256), 0, s, bfp(dy),

@@ -48,13 +48,14 @@ def test_vggf_fwd_bwd_matches_cpu():
     assert abs(float(loss) - float(loss_ref)) / max(float(loss_ref), 1) < 0.1
 
     for n, p in gm.named_parameters():
-        g = p.grad.float().cpu()
-        r = ref_grads[n]
-        scale = max(float(r.abs().max()), 1e-4)
-        err = float((g - r).abs().max())
-        # bf16 fwd + bf16 grads through 8 layers: loose band, catches layout
-        # bugs (those show up as ~100% error)
-        assert err <= 0.35 * scale, f"{n}: {err:.4g} vs {scale:.4g}"
+        g = p.grad.float().cpu().flatten()
+        r = ref_grads[n].flatten()
+        # bf16 chain noise inflates max-err; cosine + norm-ratio separates
+        # noise (cos ~0.999) from layout bugs (cos ~0 for a transposed grad)
+        cos = float(torch.nn.functional.cosine_similarity(g, r, dim=0))
+        ratio = float(g.norm() / r.norm().clamp(min=1e-8))
+        assert cos > 0.98, f"{n}: cosine {cos:.4f}"
+        assert 0.8 < ratio < 1.25, f"{n}: norm ratio {ratio:.3f}"
 
 
 def test_trainer_gpu_loss_decreases():
