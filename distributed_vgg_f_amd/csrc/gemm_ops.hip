@@ -116,53 +116,6 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
   return out;
 }
 
-// ------------------------------------------------------------- im2col
-// For Cin % 8 != 0 first layers (conv1 11x11s4 Cin=3; ResNet stem 7x7s2):
-// A [Mp][Kp] (TRANS=false) or A^T [Kp][Mp] (TRANS=true), zero-padded.
-template <bool TRANS>
-__global__ void im2col_kernel(const u16* __restrict__ x, u16* __restrict__ out,
-                              int M, int Mp, int K, int Kp, int N, int H,
-                              int W, int C, int OH, int OW, int KH, int KW,
-                              int stride, int pad) {
-  i64 chunks = TRANS ? ((i64)Kp * (Mp / 8)) : ((i64)Mp * (Kp / 8));
-  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (id >= chunks) return;
-  s16x8 v;
-  if (!TRANS) {
-    i64 m = id / (Kp / 8);
-    int k0 = int(id % (Kp / 8)) * 8;
-    int n = int(m / (OH * OW)), r = int(m % (OH * OW));
-    int oh = r / OW, ow = r % OW;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int k = k0 + j;
-      int f = k / C, c = k % C;
-      int kh = f / KW, kw = f % KW;
-      int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
-      bool ok = m < M && k < K && (unsigned)ih < (unsigned)H &&
-                (unsigned)iw < (unsigned)W;
-      v[j] = ok ? short(x[(((i64)n * H + ih) * W + iw) * C + c]) : short(0);
-    }
-    *reinterpret_cast<s16x8*>(out + m * Kp + k0) = v;
-  } else {
-    i64 k = id / (Mp / 8);
-    i64 m0 = (id % (Mp / 8)) * 8;
-    int f = int(k / C), c = int(k % C);
-    int kh = f / KW, kw = f % KW;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      i64 m = m0 + j;
-      int n = int(m / (OH * OW)), r = int(m % (OH * OW));
-      int oh = r / OW, ow = r % OW;
-      int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
-      bool ok = m < M && k < K && (unsigned)ih < (unsigned)H &&
-                (unsigned)iw < (unsigned)W;
-      v[j] = ok ? short(x[(((i64)n * H + ih) * W + iw) * C + c]) : short(0);
-    }
-    *reinterpret_cast<s16x8*>(out + k * Mp + m0) = v;
-  }
-}
-
 // ------------------------------------------------------- split-K finalize
 // out[r][c] = cast(c32[r][c] (+bias[c]) (relu)); out R rows (<= rows of c32)
 template <int EPI>
@@ -338,16 +291,12 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int epi = epi_of(bias.has_value(), relu);
 
   if (Cin % 8 != 0) {
-    // first-layer path: im2col (SURVEY.md section 7 hard part 2)
-    int Kp = ceil8(Kfull), Mp = ceil8(M);
-    auto A = at::empty({Mp, Kp}, x.options());
-    i64 chunks = (i64)Mp * (Kp / 8);
-    hipLaunchKernelGGL((im2col_kernel<false>),
-                       dim3(int((chunks + 255) / 256)), dim3(256), 0,
-                       cur_stream(), bfp(x), bfp_mut(A), M, Mp, Kfull, Kp, N,
-                       H, W, Cin, OH, OW, KH, KW, stride, pad);
-    auto wT = transpose2d(w.view({Kfull, Cout}), Kp);  // [Cout][Kp]
-    PlainGather ga{bfp(A), M, Kp, Kp};
+    // first-layer path (conv1 Cin=3, ResNet stem): scalar per-element
+    // gather straight from NHWC — no im2col materialization
+    int Kp = ceil8(Kfull);
+    auto wT = transpose2d(w.view({Kfull, Cout}), Kp);  // [Cout][Kp], 0-pad
+    ConvFwdSmallCGather ga{bfp(x), M, H, W, Cin, OH, OW, stride, pad, KW,
+                           Kfull};
     PlainGather gb{bfp(wT), Cout, Kp, Kp};
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
                 Kp, 1);
@@ -412,16 +361,22 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   // over them (conv1's Cin=3 included: the WgradX gather is element-wise).
   int OWp = ceil8(OW);
   int Kred = N * OH * OWp;
-  int Hp = H + 2 * pad;
-  int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
-  auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');           // [Cin][N][Hp][Wp]
-  auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');            // [Cout][N][OH][OWp]
-  WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride};
+  auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-  launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  if (Cin % 8 != 0) {
+    WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, KW, stride,
+                          pad};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  } else {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');  // [Cin][N][Hp][Wp]
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
   return dw;
 }

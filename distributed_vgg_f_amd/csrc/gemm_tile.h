@@ -137,6 +137,61 @@ struct WgradXGather {
   }
 };
 
+// Forward A for Cin % 8 != 0 first layers (conv1 Cin=3, ResNet stem):
+// per-element scalar gather straight from NHWC x — no im2col materialized.
+// k-space is Kp = ceil8(KH*KW*Cin); elements beyond Kfull are zero.
+struct ConvFwdSmallCGather {
+  const u16* x;
+  int M, H, W, Cin, OH, OW, stride, pad, KW_, Kfull;
+  DEV s16x8 load8(int m, int k) const {
+    s16x8 v{0, 0, 0, 0, 0, 0, 0, 0};
+    if (m >= M) return v;
+    int n = m / (OH * OW), r = m % (OH * OW);
+    int oh = r / OW, ow = r % OW;
+    int f = k / Cin, ci = k - f * Cin;
+    int kh = f / KW_, kw = f - kh * KW_;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int kk = k + j;
+      if (kk < Kfull) {
+        int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
+        if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
+          v[j] = short(x[(((i64)n * H + ih) * W + iw) * Cin + ci]);
+      }
+      if (++ci == Cin) {  // walk (kh,kw,ci) incrementally: no per-elem divs
+        ci = 0;
+        if (++kw == KW_) { kw = 0; ++kh; }
+      }
+    }
+    return v;
+  }
+};
+
+// Wgrad A for Cin % 8 != 0: scalar gather straight from NHWC x (padding via
+// per-element range predicates; padded-ow k's read garbage positions but the
+// B side (dyTpad) is zero there, and reads are range-clamped).
+struct WgradXDirectGather {
+  const u16* x;
+  int Mrows, Cin, N, H, W, OH, OWp, KW_, stride, pad;
+  DEV s16x8 load8(int m, int k) const {
+    s16x8 v{0, 0, 0, 0, 0, 0, 0, 0};
+    if (m >= Mrows) return v;
+    int f = m / Cin, ci = m - f * Cin;
+    int kh = f / KW_, kw = f - kh * KW_;
+    int n = k / (OH * OWp), r = k % (OH * OWp);
+    int oh = r / OWp, ow = r % OWp;
+    int ih = oh * stride - pad + kh;
+    if ((unsigned)ih >= (unsigned)H) return v;
+    const u16* row = x + (((i64)n * H + ih) * W) * Cin + ci;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int iw = (ow + j) * stride - pad + kw;
+      if ((unsigned)iw < (unsigned)W) v[j] = short(row[(i64)iw * Cin]);
+    }
+    return v;
+  }
+};
+
 // Wgrad B: virtual Bt[n'=co][k=(n,oh,ow-padded)] over dyTpad
 // [Cout][N][OH][OWp] (zero-filled pad, 16B rows) — clean vector loads.
 struct WgradDyGather {

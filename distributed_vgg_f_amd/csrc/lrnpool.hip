@@ -61,55 +61,50 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
                                u16* __restrict__ sb_out,
                                u16* __restrict__ snb_out, i64 npos, int C,
                                float an, float k, float beta) {
-  i64 pos = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (pos >= npos) return;
-  const u16* xr = x + pos * C;
-  u16* yr = y + pos * C;
-  u16* sbr = sb_out + pos * C;
-  u16* snbr = snb_out + pos * C;
-
-  float q[12];  // [0,1]: prev tail x^2, [2..9]: current chunk, [10,11]: next
-  q[0] = q[1] = 0.f;
-  float xc[8];
-  s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) { xc[j] = bf2f(u16(cur[j])); }
-  const int nchunks = C / 8;
-  for (int ch = 0; ch < nchunks; ++ch) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) q[2 + j] = xc[j] * xc[j];
-    float xn0 = 0.f, xn1 = 0.f;
-    s16x8 nxt{};
-    if (ch + 1 < nchunks) {
-      nxt = *reinterpret_cast<const s16x8*>(xr + (ch + 1) * 8);
-      xn0 = bf2f(u16(nxt[0]));
-      xn1 = bf2f(u16(nxt[1]));
-    }
-    q[10] = xn0 * xn0;
-    q[11] = xn1 * xn1;
-    s16x8 yv, sbv, snbv;
+  // one thread per (position, 8-channel chunk): fully parallel, neighbor
+  // chunks re-read through L1/L2
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  int nch = C / 8;
+  if (id >= npos * nch) return;
+  int ch = int(id % nch);
+  i64 pos = id / nch;
+  const u16* xr = x + pos * C + ch * 8;
+  float q[12];  // squares of channels ch*8-2 .. ch*8+9
+  {
+    s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float s = k + an * (q[j] + q[j + 1] + q[j + 2] + q[j + 3] + q[j + 4]);
-      float ls = fast_log2(s);
-      float sb = fast_exp2(-beta * ls);
-      float snb = fast_exp2(-(1.f + beta) * ls);
-      yv[j] = short(f2bf(xc[j] * sb));
-      sbv[j] = short(f2bf(sb));
-      snbv[j] = short(f2bf(snb));
+      float f = bf2f(u16(cur[j]));
+      q[2 + j] = f * f;
     }
-    *reinterpret_cast<s16x8*>(yr + ch * 8) = yv;
-    *reinterpret_cast<s16x8*>(sbr + ch * 8) = sbv;
-    *reinterpret_cast<s16x8*>(snbr + ch * 8) = snbv;
-    q[0] = q[8]; q[1] = q[9];
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      xc[j] = (j < 2) ? ((j == 0) ? xn0 : xn1) : 0.f;
-    if (ch + 1 < nchunks) {
-#pragma unroll
-      for (int j = 2; j < 8; ++j) xc[j] = bf2f(u16(nxt[j]));
+    q[0] = q[1] = 0.f;
+    if (ch > 0) {
+      float a = bf2f(xr[-2]), b = bf2f(xr[-1]);
+      q[0] = a * a; q[1] = b * b;
+    }
+    q[10] = q[11] = 0.f;
+    if (ch + 1 < nch) {
+      float a = bf2f(xr[8]), b = bf2f(xr[9]);
+      q[10] = a * a; q[11] = b * b;
     }
   }
+  s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
+  s16x8 yv, sbv, snbv;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float sum = q[j] + q[j + 1] + q[j + 2] + q[j + 3] + q[j + 4];
+    float sv = k + an * sum;
+    float ls = fast_log2(sv);
+    float sb = fast_exp2(-beta * ls);
+    float snb = fast_exp2(-(1.f + beta) * ls);
+    yv[j] = short(f2bf(bf2f(u16(cur[j])) * sb));
+    sbv[j] = short(f2bf(sb));
+    snbv[j] = short(f2bf(snb));
+  }
+  i64 o = pos * C + ch * 8;
+  *reinterpret_cast<s16x8*>(y + o) = yv;
+  *reinterpret_cast<s16x8*>(sb_out + o) = sbv;
+  *reinterpret_cast<s16x8*>(snb_out + o) = snbv;
 }
 
 std::vector<at::Tensor> lrn_fwd(const at::Tensor& x, int64_t size,
@@ -122,10 +117,11 @@ std::vector<at::Tensor> lrn_fwd(const at::Tensor& x, int64_t size,
   auto y = at::empty_like(x);
   auto sb = at::empty_like(x);
   auto snb = at::empty_like(x);
-  hipLaunchKernelGGL(lrn_fwd_kernel, dim3(int((npos + 255) / 256)), dim3(256),
-                     0, cur_stream(), bfp(x), bfp_mut(y), bfp_mut(sb),
-                     bfp_mut(snb), npos, C, float(alpha / size), float(k),
-                     float(beta));
+  i64 nthreads = npos * (C / 8);
+  hipLaunchKernelGGL(lrn_fwd_kernel, dim3(int((nthreads + 255) / 256)),
+                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(y),
+                     bfp_mut(sb), bfp_mut(snb), npos, C, float(alpha / size),
+                     float(k), float(beta));
   return {y, sb, snb};
 }
 
@@ -136,62 +132,46 @@ __global__ void lrn_bwd_kernel(const u16* __restrict__ dy,
                                const u16* __restrict__ snb,
                                u16* __restrict__ dx, i64 npos, int C,
                                float coef) {  // coef = 2*alpha*beta/size
-  i64 pos = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (pos >= npos) return;
-  const u16* dyr = dy + pos * C;
-  const u16* xr = x + pos * C;
-  const u16* sbr = sb + pos * C;
-  const u16* snbr = snb + pos * C;
-  u16* dxr = dx + pos * C;
-
-  float t[12];  // t = dy * x * snb, sliding window like fwd
-  t[0] = t[1] = 0.f;
-  float xc[8], dc[8], sbc[8];
-  const int nchunks = C / 8;
-  // preload chunk 0
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  int nch = C / 8;
+  if (id >= npos * nch) return;
+  int ch = int(id % nch);
+  i64 pos = id / nch;
+  i64 o = pos * C + ch * 8;
+  const u16* dyr = dy + o;
+  const u16* xr = x + o;
+  const u16* snbr = snb + o;
+  float t[12];  // t = dy * x * snb for channels ch*8-2 .. ch*8+9
   {
-    s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
     s16x8 dv = *reinterpret_cast<const s16x8*>(dyr);
+    s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
     s16x8 sv = *reinterpret_cast<const s16x8*>(snbr);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      xc[j] = bf2f(u16(xv[j]));
-      dc[j] = bf2f(u16(dv[j]));
-      sbc[j] = bf2f(u16(sv[j]));  // snb values staged through sbc temp
+    for (int j = 0; j < 8; ++j)
+      t[2 + j] = bf2f(u16(dv[j])) * bf2f(u16(xv[j])) * bf2f(u16(sv[j]));
+    t[0] = t[1] = 0.f;
+    if (ch > 0) {
+      t[0] = bf2f(dyr[-2]) * bf2f(xr[-2]) * bf2f(snbr[-2]);
+      t[1] = bf2f(dyr[-1]) * bf2f(xr[-1]) * bf2f(snbr[-1]);
+    }
+    t[10] = t[11] = 0.f;
+    if (ch + 1 < nch) {
+      t[10] = bf2f(dyr[8]) * bf2f(xr[8]) * bf2f(snbr[8]);
+      t[11] = bf2f(dyr[9]) * bf2f(xr[9]) * bf2f(snbr[9]);
     }
   }
-  for (int ch = 0; ch < nchunks; ++ch) {
+  s16x8 dv = *reinterpret_cast<const s16x8*>(dyr);
+  s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
+  s16x8 sbv = *reinterpret_cast<const s16x8*>(sb + o);
+  s16x8 out;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) t[2 + j] = dc[j] * xc[j] * sbc[j];
-    float tn0 = 0.f, tn1 = 0.f;
-    s16x8 xv{}, dv{}, sv{};
-    if (ch + 1 < nchunks) {
-      xv = *reinterpret_cast<const s16x8*>(xr + (ch + 1) * 8);
-      dv = *reinterpret_cast<const s16x8*>(dyr + (ch + 1) * 8);
-      sv = *reinterpret_cast<const s16x8*>(snbr + (ch + 1) * 8);
-      tn0 = bf2f(u16(dv[0])) * bf2f(u16(xv[0])) * bf2f(u16(sv[0]));
-      tn1 = bf2f(u16(dv[1])) * bf2f(u16(xv[1])) * bf2f(u16(sv[1]));
-    }
-    t[10] = tn0; t[11] = tn1;
-    s16x8 sbv = *reinterpret_cast<const s16x8*>(sbr + ch * 8);
-    s16x8 out;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float wsum = t[j] + t[j + 1] + t[j + 2] + t[j + 3] + t[j + 4];
-      float v = dc[j] * bf2f(u16(sbv[j])) - coef * xc[j] * wsum;
-      out[j] = short(f2bf(v));
-    }
-    *reinterpret_cast<s16x8*>(dxr + ch * 8) = out;
-    t[0] = t[8]; t[1] = t[9];
-    if (ch + 1 < nchunks) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        xc[j] = bf2f(u16(xv[j]));
-        dc[j] = bf2f(u16(dv[j]));
-        sbc[j] = bf2f(u16(sv[j]));
-      }
-    }
+  for (int j = 0; j < 8; ++j) {
+    float wsum = t[j] + t[j + 1] + t[j + 2] + t[j + 3] + t[j + 4];
+    float v = bf2f(u16(dv[j])) * bf2f(u16(sbv[j])) -
+              coef * bf2f(u16(xv[j])) * wsum;
+    out[j] = short(f2bf(v));
   }
+  *reinterpret_cast<s16x8*>(dx + o) = out;
 }
 
 at::Tensor lrn_bwd(const at::Tensor& dy, const at::Tensor& x,
@@ -201,9 +181,10 @@ at::Tensor lrn_bwd(const at::Tensor& dy, const at::Tensor& x,
   int C = dy.size(-1);
   i64 npos = dy.numel() / C;
   auto dx = at::empty_like(dy);
-  hipLaunchKernelGGL(lrn_bwd_kernel, dim3(int((npos + 255) / 256)), dim3(256),
-                     0, cur_stream(), bfp(dy), bfp(x), bfp(sb), bfp(snb),
-                     bfp_mut(dx), npos, C,
+  i64 nthreads = npos * (C / 8);
+  hipLaunchKernelGGL(lrn_bwd_kernel, dim3(int((nthreads + 255) / 256)),
+                     dim3(256), 0, cur_stream(), bfp(dy), bfp(x), bfp(sb),
+                     bfp(snb), bfp_mut(dx), npos, C,
                      float(2.0 * alpha * beta / size));
   return dx;
 }
@@ -215,29 +196,47 @@ __global__ void maxpool_fwd_kernel(const u16* __restrict__ x,
                                    unsigned char* __restrict__ idx, int N,
                                    int H, int W, int C, int OH, int OW,
                                    int kernel, int stride, int pl) {
+  // 8 channels per thread, vectorized window reads
   i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  i64 total = (i64)N * OH * OW * C;
+  i64 total = (i64)N * OH * OW * (C / 8);
   if (id >= total) return;
-  int c = int(id % C);
-  i64 r = id / C;
+  int c8 = int(id % (C / 8)) * 8;
+  i64 r = id / (C / 8);
   int ow = int(r % OW); r /= OW;
   int oh = int(r % OH); r /= OH;
   int n = int(r);
   int ih0 = oh * stride - pl, iw0 = ow * stride - pl;
-  float best = -3.4e38f;
-  int bi = 0;
+  float best[8];
+  int bi[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) { best[j] = -3.4e38f; bi[j] = 0; }
   for (int dh = 0; dh < kernel; ++dh) {
     int ih = ih0 + dh;
     if ((unsigned)ih >= (unsigned)H) continue;
     for (int dw = 0; dw < kernel; ++dw) {
       int iw = iw0 + dw;
       if ((unsigned)iw >= (unsigned)W) continue;
-      float v = bf2f(x[(((i64)n * H + ih) * W + iw) * C + c]);
-      if (v > best) { best = v; bi = dh * kernel + dw; }
+      s16x8 v = *reinterpret_cast<const s16x8*>(
+          x + (((i64)n * H + ih) * W + iw) * C + c8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(u16(v[j]));
+        if (f > best[j]) { best[j] = f; bi[j] = dh * kernel + dw; }
+      }
     }
   }
-  y[id] = f2bf(best);
-  idx[id] = (unsigned char)bi;
+  i64 o = (((i64)n * OH + oh) * OW + ow) * C + c8;
+  s16x8 out;
+  u32 i0 = 0, i1 = 0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    out[j] = short(f2bf(best[j]));
+    if (j < 4) i0 |= u32(bi[j]) << (j * 8);
+    else i1 |= u32(bi[j]) << ((j - 4) * 8);
+  }
+  *reinterpret_cast<s16x8*>(y + o) = out;
+  *reinterpret_cast<u32*>(idx + o) = i0;
+  *reinterpret_cast<u32*>(idx + o + 4) = i1;
 }
 
 std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kernel,
@@ -248,7 +247,8 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kernel,
   int OW = OH;  // square inputs throughout; generalize if needed
   auto y = at::empty({N, OH, OW, C}, x.options());
   auto idx = at::empty({N, OH, OW, C}, x.options().dtype(at::kByte));
-  i64 total = (i64)N * OH * OW * C;
+  TORCH_CHECK(C % 8 == 0, "maxpool needs C % 8 == 0");
+  i64 total = (i64)N * OH * OW * (C / 8);
   hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(int((total + 255) / 256)),
                      dim3(256), 0, cur_stream(), bfp(x), bfp_mut(y),
                      idx.data_ptr<unsigned char>(), N, H, W, C, OH, OW,

@@ -104,9 +104,17 @@ __global__ void bias_grad_kernel(const u16* __restrict__ dy,
   if (c >= C) return;
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
-  float acc = 0.f;
-  for (i64 r = r0; r < r1; ++r) acc += bf2f(dy[r * C + c]);
-  atomicAdd(&db32[c], acc);
+  // 4 independent accumulators: break the serial add chain (latency-bound)
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  i64 r = r0;
+  for (; r + 4 <= r1; r += 4) {
+    a0 += bf2f(dy[r * C + c]);
+    a1 += bf2f(dy[(r + 1) * C + c]);
+    a2 += bf2f(dy[(r + 2) * C + c]);
+    a3 += bf2f(dy[(r + 3) * C + c]);
+  }
+  for (; r < r1; ++r) a0 += bf2f(dy[r * C + c]);
+  atomicAdd(&db32[c], (a0 + a1) + (a2 + a3));
 }
 
 __global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
