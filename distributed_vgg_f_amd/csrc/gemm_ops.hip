@@ -154,43 +154,54 @@ static void run_finalize(const at::Tensor& c32, at::Tensor& out,
 }
 
 // ------------------------------------------------------------ gemm driver
-template <class GA, class GB>
-static void launch_gemm(const GA& ga, const GB& gb, at::Tensor* outB,
-                        at::Tensor* c32, const at::Tensor* bias, int epi,
-                        int M, int N, int K, int splitk) {
-  int tiles = ceil_div(M, BM) * ceil_div(N, BN);
+template <class GA, class GB, int NFRAG>
+static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
+                          at::Tensor* c32, const at::Tensor* bias, int epi,
+                          int M, int N, int K, int splitk) {
+  constexpr int BN_T = 32 * NFRAG;
+  int tiles = ceil_div(M, BM) * ceil_div(N, BN_T);
   auto s = cur_stream();
   const u16* bp = bias ? bfp(*bias) : nullptr;
   if (splitk > 1) {
     dim3 grid(tiles, splitk);
-    hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true>), grid, dim3(THREADS), 0,
-                       s, ga, gb, nullptr, c32->data_ptr<float>(), nullptr, M,
-                       N, K, 0, K);
+    hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, NFRAG>), grid,
+                       dim3(THREADS), 0, s, ga, gb, nullptr,
+                       c32->data_ptr<float>(), nullptr, M, N, K, 0, K);
   } else {
     dim3 grid(tiles);
     switch (epi) {
       case 0:
-        hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, false>), grid,
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, false, NFRAG>), grid,
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, nullptr, M, N, K, 0, K);
         break;
       case 1:
-        hipLaunchKernelGGL((gemm_kernel<GA, GB, 1, false>), grid,
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 1, false, NFRAG>), grid,
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, bp, M, N, K, 0, K);
         break;
       case 2:
-        hipLaunchKernelGGL((gemm_kernel<GA, GB, 2, false>), grid,
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 2, false, NFRAG>), grid,
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, bp, M, N, K, 0, K);
         break;
       default:
-        hipLaunchKernelGGL((gemm_kernel<GA, GB, 3, false>), grid,
+        hipLaunchKernelGGL((gemm_kernel<GA, GB, 3, false, NFRAG>), grid,
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, nullptr, M, N, K, 0, K);
         break;
     }
   }
+}
+
+template <class GA, class GB>
+static void launch_gemm(const GA& ga, const GB& gb, at::Tensor* outB,
+                        at::Tensor* c32, const at::Tensor* bias, int epi,
+                        int M, int N, int K, int splitk) {
+  if (N <= 64)
+    launch_gemm_t<GA, GB, 2>(ga, gb, outB, c32, bias, epi, M, N, K, splitk);
+  else
+    launch_gemm_t<GA, GB, 4>(ga, gb, outB, c32, bias, epi, M, N, K, splitk);
 }
 
 static int pick_splitk(int tiles, int K, int target_blocks = 512) {
@@ -362,19 +373,20 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   int OWp = ceil8(OW);
   int Kred = N * OH * OWp;
   auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
-  WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp};
+  WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, KW, stride,
-                          pad};
+                          pad, Kred};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   } else {
     int Hp = H + 2 * pad;
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');  // [Cin][N][Hp][Wp]
-    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride};
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride,
+                    Kred};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);

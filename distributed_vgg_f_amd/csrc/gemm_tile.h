@@ -3,28 +3,32 @@
 // "A"/"B" operands are gather functors, so conv fwd/dgrad/wgrad reuse one
 // tile pipeline with different address functions.
 //
-// Geometry: 128x128 block tile, BK=32 k-step, 256 threads = 4 waves in a
-// 2x2 grid, each wave computing a 64x64 sub-tile as 4x4 fragments of
-// v_mfma_f32_32x... (16x16x32_bf16), fp32 accumulation in AGPRs.
-// LDS tiles are [row][BK] row-major with +8 bf16 row pad (80 B row stride:
-// 16 consecutive rows land on distinct bank groups for ds_read_b128 —
-// cdna_hip_programming.md section 6 Guideline 4).
+// v2 geometry: 128 x (32*NFRAG) block tile (NFRAG=4 -> 128x128, NFRAG=2 ->
+// 128x64 for Cout=64 layers), BK=64 k-step, 256 threads = 4 waves in a 2x2
+// grid, each wave a 64 x (16*NFRAG) sub-tile of v_mfma_f32_16x16x32_bf16
+// fragments, fp32 accumulation. ONE barrier per k-tile: double-buffered LDS
+// where writing buf[p] at tile t is fenced from tile t-2's readers by tile
+// t-1's barrier; next tile's global loads issue right after the barrier so
+// HBM latency hides under the 32 MFMAs (T14 issue-early / write-late,
+// cdna_hip_programming.md Guideline 15).
 //
-// Staging is register-staged (global->VGPR->LDS) with the next tile's
-// global loads issued before the current tile's MFMAs (T14 issue-early /
-// write-late, cdna_hip_programming.md Guideline 15). Layout contracts:
-// every gather's k-chunks (8 consecutive virtual-k) must be either fully
-// valid 16B-aligned vector loads or zero; see the functors.
+// LDS rows are [row][BK] bf16 with +8 row pad (144 B stride: 16 consecutive
+// rows hit distinct bank groups for ds_read_b128 — Guideline 4).
+//
+// Gather functors expose prep(row) -> Row (hoists the row-side integer
+// divisions out of the K loop) and load8(Row, k) -> 8 bf16 (zero where out
+// of range; k is a multiple of 8).
 #pragma once
 
 #include "common.h"
 
 namespace dvgg {
 
-constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int BM = 128, BK = 64;
 constexpr int LDS_PAD = 8;                 // bf16 elements
-constexpr int LDS_STRIDE = BK + LDS_PAD;   // 40 elems = 80 B rows
+constexpr int LDS_STRIDE = BK + LDS_PAD;   // 72 elems = 144 B rows
 constexpr int THREADS = 256;
+constexpr int BN = 128;                    // default tile N (NFRAG=4)
 
 DEV bf16x8 as_bf16x8(s16x8 v) {
   union { s16x8 s; bf16x8 b; } u;
@@ -32,90 +36,127 @@ DEV bf16x8 as_bf16x8(s16x8 v) {
   return u.b;
 }
 
+DEV s16x8 zero8() { return s16x8{0, 0, 0, 0, 0, 0, 0, 0}; }
+
 // ---------------------------------------------------------------- gathers
-// Contract: load8(m, k) returns elements [m][k..k+7] of the virtual
-// matrix, ZERO where out of range. k is a multiple of 8.
 
 struct PlainGather {
   const u16* p;
   int rows, cols, ld;  // ld multiple of 8 (16B-aligned rows)
-  DEV s16x8 load8(int m, int k) const {
-    if (m < rows && k < cols)
-      return *reinterpret_cast<const s16x8*>(p + (i64)m * ld + k);
-    return s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  struct Row { const u16* rp; bool valid; };
+  DEV Row prep(int m) const { return {p + (i64)m * ld, m < rows}; }
+  DEV s16x8 load8(const Row& r, int k) const {
+    if (r.valid && k < cols)
+      return *reinterpret_cast<const s16x8*>(r.rp + k);
+    return zero8();
   }
 };
 
 // Implicit-GEMM forward A: virtual A[m=(n,oh,ow)][k=(kh,kw,ci)] = x NHWC.
-// Requires Cin % 8 == 0 (chunks stay inside one filter tap) — conv1-style
-// Cin=3 goes through im2col instead (SURVEY.md section 7 hard part 2).
+// Requires Cin % 8 == 0 (chunks stay inside one filter tap) — Cin<8 first
+// layers use ConvFwdSmallCGather instead.
 struct ConvFwdGather {
   const u16* x;
-  int M, H, W, Cin, OH, OW, stride, pad;
-  DEV s16x8 load8(int m, int k) const {
-    int f = k / Cin, c0 = k % Cin;      // Cin % 8 == 0 keeps chunk in-tap
-    int kw = 0, kh = 0;
-    // KW folded into f decomposition by caller packing: f = kh*KW + kw
-    kh = f / KW_;
-    kw = f % KW_;
+  int M, H, W, Cin, OH, OW, stride, pad, KW_;
+  struct Row { const u16* nrow; int ihb, iwb; bool valid; };
+  DEV Row prep(int m) const {
     int n = m / (OH * OW), r = m % (OH * OW);
     int oh = r / OW, ow = r % OW;
-    int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
-    if (m < M && (unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
-      return *reinterpret_cast<const s16x8*>(
-          x + (((i64)n * H + ih) * W + iw) * Cin + c0);
-    return s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    return {x + (i64)n * H * W * Cin, oh * stride - pad, ow * stride - pad,
+            m < M};
   }
-  int KW_;
+  DEV s16x8 load8(const Row& rs, int k) const {
+    int f = k / Cin, c0 = k % Cin;
+    int kh = f / KW_, kw = f % KW_;
+    int ih = rs.ihb + kh, iw = rs.iwb + kw;
+    if (rs.valid && (unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
+      return *reinterpret_cast<const s16x8*>(rs.nrow +
+                                             ((i64)ih * W + iw) * Cin + c0);
+    return zero8();
+  }
 };
 
 // Implicit-GEMM dgrad A: virtual A[m=(n,ih,iw)][k=(kh,kw,co)] =
-// dy[n][(ih+pad-kh)/s][(iw+pad-kw)/s][co] when divisible & in range.
-// Requires Cout % 8 == 0.
+// dy[n][(ih+pad-kh)/s][(iw+pad-kw)/s][co]. Requires Cout % 8 == 0.
 struct ConvDgradGather {
   const u16* dy;
   int M, H, W, Cout, OH, OW, stride, pad, KW_;
-  DEV s16x8 load8(int m, int k) const {
-    int f = k / Cout, c0 = k % Cout;
-    int kh = f / KW_, kw = f % KW_;
+  struct Row { const u16* nrow; int ihp, iwp; bool valid; };
+  DEV Row prep(int m) const {
     int n = m / (H * W), r = m % (H * W);
     int ih = r / W, iw = r % W;
-    int ohs = ih + pad - kh, ows = iw + pad - kw;
+    return {dy + (i64)n * OH * OW * Cout, ih + pad, iw + pad, m < M};
+  }
+  DEV s16x8 load8(const Row& rs, int k) const {
+    int f = k / Cout, c0 = k % Cout;
+    int kh = f / KW_, kw = f % KW_;
+    int ohs = rs.ihp - kh, ows = rs.iwp - kw;
     int oh = ohs / stride, ow = ows / stride;
-    bool ok = m < M && ohs >= 0 && ows >= 0 && oh * stride == ohs &&
+    bool ok = rs.valid && ohs >= 0 && ows >= 0 && oh * stride == ohs &&
               ow * stride == ows && oh < OH && ow < OW;
     if (ok)
-      return *reinterpret_cast<const s16x8*>(
-          dy + (((i64)n * OH + oh) * OW + ow) * Cout + c0);
-    return s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      return *reinterpret_cast<const s16x8*>(rs.nrow +
+                                             ((i64)oh * OW + ow) * Cout + c0);
+    return zero8();
   }
 };
 
-// Wgrad A (stride-1 convs): virtual A[m=(kh,kw,ci)][k=(n,oh,ow-padded)]
-// over xTpad [Cin][N][H+2p][Wp] (halo-padded, zero-filled, built by
-// nhwc2cnhw_pad). k-space uses OW rounded to OWp (mult of 8) so chunks
-// never cross rows; pad reads hit pre-zeroed halo -> no predicates, but
-// the +kw shift breaks 16B alignment so elements load scalar (v1; see
-// csrc/README note on the wgrad staging optimization).
+// Forward A for Cin % 8 != 0 first layers (conv1 Cin=3, ResNet stem):
+// per-element scalar gather straight from NHWC x — no im2col materialized.
+// k-space is Kp = ceil8(KH*KW*Cin); elements beyond Kfull are zero.
+struct ConvFwdSmallCGather {
+  const u16* x;
+  int M, H, W, Cin, OH, OW, stride, pad, KW_, Kfull;
+  struct Row { const u16* nrow; int ihb, iwb; bool valid; };
+  DEV Row prep(int m) const {
+    int n = m / (OH * OW), r = m % (OH * OW);
+    int oh = r / OW, ow = r % OW;
+    return {x + (i64)n * H * W * Cin, oh * stride - pad, ow * stride - pad,
+            m < M};
+  }
+  DEV s16x8 load8(const Row& rs, int k) const {
+    s16x8 v = zero8();
+    if (!rs.valid) return v;
+    int f = k / Cin, ci = k - f * Cin;
+    int kh = f / KW_, kw = f - kh * KW_;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int kk = k + j;
+      if (kk < Kfull) {
+        int ih = rs.ihb + kh, iw = rs.iwb + kw;
+        if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
+          v[j] = short(rs.nrow[((i64)ih * W + iw) * Cin + ci]);
+      }
+      if (++ci == Cin) {  // walk (kh,kw,ci) incrementally: no per-elem divs
+        ci = 0;
+        if (++kw == KW_) { kw = 0; ++kh; }
+      }
+    }
+    return v;
+  }
+};
+
+// Wgrad A (Cin % 8 == 0): virtual A[m=(kh,kw,ci)][k=(n,oh,ow-padded)] over
+// xTpad [Cin][N][H+2p][Wp] (halo-padded, zero-filled; +8 elem slack). The
+// k-space pads OW to OWp (mult of 8) so chunks never cross rows.
 struct WgradXGather {
   const u16* xT;
-  int Mrows, Cin, N, Hp, Wp, OH, OWp, KW_, stride;  // Hp=H+2p, Wp aligned
-  DEV s16x8 load8(int m, int k) const {
-    if (m >= Mrows) return s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-    int f = m / Cin;            // m = (kh*KW + kw) * Cin + ci
-    int ci = m % Cin;
-    int kh = f / KW_, kw = f % KW_;
+  int Mrows, Cin, N, Hp, Wp, OH, OWp, KW_, stride, Klim;
+  struct Row { const u16* cib; int kh, kw; bool valid; };
+  DEV Row prep(int m) const {
+    int f = m / Cin, ci = m - f * Cin;
+    return {xT + (i64)ci * N * Hp * Wp, f / KW_, f % KW_, m < Mrows};
+  }
+  DEV s16x8 load8(const Row& rs, int k) const {
+    if (!rs.valid || k >= Klim) return zero8();
     int n = k / (OH * OWp), r = k % (OH * OWp);
     int oh = r / OWp, ow = r % OWp;
-    // stored: x[ih][iw] at [ih+pad][iw+pad]; read ih=oh*s+kh-pad -> row
-    // (oh*s+kh), iw=(ow+j)*s+kw-pad -> col ((ow+j)*s+kw); halo is zero.
-    const u16* base = xT + (((i64)ci * N + n) * Hp + (oh * stride + kh)) * Wp +
-                      (ow * stride + kw);
+    const u16* base = rs.cib + ((i64)n * Hp + (oh * stride + rs.kh)) * Wp +
+                      (ow * stride + rs.kw);
     s16x8 v;
     if (stride == 1) {
-      // vector staging despite the odd +kw element shift: read the aligned
-      // dword window and funnel-shift (buffer has +8 elem slack for the
-      // trailing read; halo is pre-zeroed so no predicates).
+      // vector staging despite the odd +kw shift: aligned dword window +
+      // funnel shift (halo pre-zeroed, +8 elem slack for trailing read)
       uintptr_t ap = (uintptr_t)base;
       const u32* q = reinterpret_cast<const u32*>(ap & ~(uintptr_t)3);
       u32 a0 = q[0], a1 = q[1], a2 = q[2], a3 = q[3];
@@ -137,55 +178,27 @@ struct WgradXGather {
   }
 };
 
-// Forward A for Cin % 8 != 0 first layers (conv1 Cin=3, ResNet stem):
-// per-element scalar gather straight from NHWC x — no im2col materialized.
-// k-space is Kp = ceil8(KH*KW*Cin); elements beyond Kfull are zero.
-struct ConvFwdSmallCGather {
-  const u16* x;
-  int M, H, W, Cin, OH, OW, stride, pad, KW_, Kfull;
-  DEV s16x8 load8(int m, int k) const {
-    s16x8 v{0, 0, 0, 0, 0, 0, 0, 0};
-    if (m >= M) return v;
-    int n = m / (OH * OW), r = m % (OH * OW);
-    int oh = r / OW, ow = r % OW;
-    int f = k / Cin, ci = k - f * Cin;
-    int kh = f / KW_, kw = f - kh * KW_;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int kk = k + j;
-      if (kk < Kfull) {
-        int ih = oh * stride - pad + kh, iw = ow * stride - pad + kw;
-        if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
-          v[j] = short(x[(((i64)n * H + ih) * W + iw) * Cin + ci]);
-      }
-      if (++ci == Cin) {  // walk (kh,kw,ci) incrementally: no per-elem divs
-        ci = 0;
-        if (++kw == KW_) { kw = 0; ++kh; }
-      }
-    }
-    return v;
-  }
-};
-
 // Wgrad A for Cin % 8 != 0: scalar gather straight from NHWC x (padding via
-// per-element range predicates; padded-ow k's read garbage positions but the
-// B side (dyTpad) is zero there, and reads are range-clamped).
+// per-element range predicates; padded-ow k's are zeroed by the B side).
 struct WgradXDirectGather {
   const u16* x;
-  int Mrows, Cin, N, H, W, OH, OWp, KW_, stride, pad;
-  DEV s16x8 load8(int m, int k) const {
-    s16x8 v{0, 0, 0, 0, 0, 0, 0, 0};
-    if (m >= Mrows) return v;
+  int Mrows, Cin, N, H, W, OH, OWp, KW_, stride, pad, Klim;
+  struct Row { const u16* cib; int kh, kw; bool valid; };
+  DEV Row prep(int m) const {
     int f = m / Cin, ci = m - f * Cin;
-    int kh = f / KW_, kw = f - kh * KW_;
+    return {x + ci, f / KW_, f % KW_, m < Mrows};
+  }
+  DEV s16x8 load8(const Row& rs, int k) const {
+    s16x8 v = zero8();
+    if (!rs.valid || k >= Klim) return v;
     int n = k / (OH * OWp), r = k % (OH * OWp);
     int oh = r / OWp, ow = r % OWp;
-    int ih = oh * stride - pad + kh;
+    int ih = oh * stride - pad + rs.kh;
     if ((unsigned)ih >= (unsigned)H) return v;
-    const u16* row = x + (((i64)n * H + ih) * W) * Cin + ci;
+    const u16* row = rs.cib + (((i64)n * H + ih) * W) * Cin;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int iw = (ow + j) * stride - pad + kw;
+      int iw = (ow + j) * stride - pad + rs.kw;
       if ((unsigned)iw < (unsigned)W) v[j] = short(row[(i64)iw * Cin]);
     }
     return v;
@@ -196,33 +209,39 @@ struct WgradXDirectGather {
 // [Cout][N][OH][OWp] (zero-filled pad, 16B rows) — clean vector loads.
 struct WgradDyGather {
   const u16* dyT;
-  int Cout, N, OH, OWp;
-  DEV s16x8 load8(int m, int k) const {
-    if (m < Cout)
-      return *reinterpret_cast<const s16x8*>(dyT + (i64)m * N * OH * OWp + k);
-    return s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  int Cout, N, OH, OWp, Klim;
+  struct Row { const u16* rp; bool valid; };
+  DEV Row prep(int m) const {
+    return {dyT + (i64)m * N * OH * OWp, m < Cout};
+  }
+  DEV s16x8 load8(const Row& r, int k) const {
+    if (r.valid && k < Klim) return *reinterpret_cast<const s16x8*>(r.rp + k);
+    return zero8();
   }
 };
 
 // ------------------------------------------------------------------ core
 // EPI: 0 = plain, 1 = +bias, 2 = +bias+ReLU, 3 = ReLU only.
 // SPLITK: atomicAdd fp32 into c32 (epilogue applied by a finalize kernel).
+// NFRAG: B fragments per wave; tile N = 32*NFRAG (4 -> 128, 2 -> 64).
 
-template <class GA, class GB, int EPI, bool SPLITK>
+template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4>
 __global__ __launch_bounds__(THREADS) void gemm_kernel(
     GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
     const u16* __restrict__ bias, int M, int N, int K, int k_lo, int k_hi) {
-  __shared__ u16 As[BM * LDS_STRIDE];
-  __shared__ u16 Bs[BN * LDS_STRIDE];
+  constexpr int BN_T = 32 * NFRAG;
+  __shared__ u16 As[2][BM * LDS_STRIDE];
+  __shared__ u16 Bs[2][BN_T * LDS_STRIDE];
 
-  const int tiles_n = (N + BN - 1) / BN;
+  const int tiles_n = (N + BN_T - 1) / BN_T;
   const int tiles_m = (M + BM - 1) / BM;
   u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
   const int tm = wg / tiles_n, tn = wg % tiles_n;
-  const int m0 = tm * BM, n0 = tn * BN;
+  const int m0 = tm * BM, n0 = tn * BN_T;
   if (SPLITK) {
     int nchunk = gridDim.y;
-    int span = (K / BK + nchunk - 1) / nchunk * BK;
+    int kt = (K + BK - 1) / BK;
+    int span = (kt + nchunk - 1) / nchunk * BK;
     k_lo = blockIdx.y * span;
     k_hi = min(K, k_lo + span);
     if (k_lo >= k_hi) return;
@@ -230,60 +249,75 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
 
   const int t = threadIdx.x;
   const int lane = t & 63, wave = t >> 6;
-  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
 
-  // staging assignment: thread t stages row t>>1, k-chunks (t&1)*16 + {0,8}
-  const int srow = t >> 1, sk = (t & 1) * 16;
+  // staging assignment: A: 2 threads/row, 4 chunks each;
+  // B: 256/BN_T threads/row, BN_T/32 chunks each.
+  constexpr int CPT_B = BN_T / 32;            // chunks per thread (B)
+  constexpr int TPR_B = 256 / BN_T;           // threads per row (B)
+  const int arow = t >> 1, ak8 = (t & 1) * 32;
+  const int brow = t / TPR_B, bk8 = (t % TPR_B) * (CPT_B * 8);
 
-  f32x4 acc[4][4];
+  const auto rsa = ga.prep(m0 + arow);
+  const auto rsb = gb.prep(n0 + brow);
+
+  f32x4 acc[4][NFRAG];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NFRAG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nt = (k_hi - k_lo + BK - 1) / BK;
 
-  // prologue: gather tile 0
-  s16x8 ra0 = ga.load8(m0 + srow, k_lo + sk);
-  s16x8 ra1 = ga.load8(m0 + srow, k_lo + sk + 8);
-  s16x8 rb0 = gb.load8(n0 + srow, k_lo + sk);
-  s16x8 rb1 = gb.load8(n0 + srow, k_lo + sk + 8);
+  s16x8 ra[4], rb[CPT_B];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) ra[i] = ga.load8(rsa, k_lo + ak8 + i * 8);
+#pragma unroll
+  for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, k_lo + bk8 + i * 8);
 
+  int p = 0;
   for (int tile = 0; tile < nt; ++tile) {
-    __syncthreads();  // readers of previous tile done
-    *reinterpret_cast<s16x8*>(&As[srow * LDS_STRIDE + sk]) = ra0;
-    *reinterpret_cast<s16x8*>(&As[srow * LDS_STRIDE + sk + 8]) = ra1;
-    *reinterpret_cast<s16x8*>(&Bs[srow * LDS_STRIDE + sk]) = rb0;
-    *reinterpret_cast<s16x8*>(&Bs[srow * LDS_STRIDE + sk + 8]) = rb1;
-    __syncthreads();  // tile ready
-    if (tile + 1 < nt) {  // issue next tile's loads early (overlap MFMA)
-      int kb = k_lo + (tile + 1) * BK;
-      ra0 = ga.load8(m0 + srow, kb + sk);
-      ra1 = ga.load8(m0 + srow, kb + sk + 8);
-      rb0 = gb.load8(n0 + srow, kb + sk);
-      rb1 = gb.load8(n0 + srow, kb + sk + 8);
-    }
-    // fragment reads + 16 MFMA (one K-step of 32)
-    const int kfrag = (lane >> 4) * 8;
-    bf16x8 af[4], bf[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int row = wm + i * 16 + (lane & 15);
-      af[i] = as_bf16x8(
-          *reinterpret_cast<const s16x8*>(&As[row * LDS_STRIDE + kfrag]));
-    }
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = wn + j * 16 + (lane & 15);
-      bf[j] = as_bf16x8(
-          *reinterpret_cast<const s16x8*>(&Bs[col * LDS_STRIDE + kfrag]));
-    }
 #pragma unroll
     for (int i = 0; i < 4; ++i)
+      *reinterpret_cast<s16x8*>(&As[p][arow * LDS_STRIDE + ak8 + i * 8]) =
+          ra[i];
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf[j], acc[i][j], 0, 0, 0);
+    for (int i = 0; i < CPT_B; ++i)
+      *reinterpret_cast<s16x8*>(&Bs[p][brow * LDS_STRIDE + bk8 + i * 8]) =
+          rb[i];
+    // ONE barrier per tile: also fences tile-2's readers of buf[p^1]
+    __syncthreads();
+    if (tile + 1 < nt) {  // issue next tile's loads under the MFMAs
+      int kb = k_lo + (tile + 1) * BK;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) ra[i] = ga.load8(rsa, kb + ak8 + i * 8);
+#pragma unroll
+      for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, kb + bk8 + i * 8);
+    }
+    const int kfrag = (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[4], bf[NFRAG];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = wm + i * 16 + (lane & 15);
+        af[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            &As[p][row * LDS_STRIDE + ks * 32 + kfrag]));
+      }
+#pragma unroll
+      for (int j = 0; j < NFRAG; ++j) {
+        int col = wn + j * 16 + (lane & 15);
+        bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            &Bs[p][col * LDS_STRIDE + ks * 32 + kfrag]));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < NFRAG; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    p ^= 1;
   }
 
   // ---------------------------------------------------------- epilogue
@@ -291,7 +325,7 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   for (int i = 0; i < 4; ++i) {
     int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < NFRAG; ++j) {
       int col = n0 + wn + j * 16 + (lane & 15);
       if (col >= N) continue;
       float bv = (EPI == 1 || EPI == 2) ? bf2f(bias[col]) : 0.f;
