@@ -24,9 +24,9 @@
 
 namespace dvgg {
 
-constexpr int BM = 128, BK = 64;
+constexpr int BM = 128, BK = 32;
 constexpr int LDS_PAD = 8;                 // bf16 elements
-constexpr int LDS_STRIDE = BK + LDS_PAD;   // 72 elems = 144 B rows
+constexpr int LDS_STRIDE = BK + LDS_PAD;   // 40 elems = 80 B rows
 constexpr int THREADS = 256;
 constexpr int BN = 128;                    // default tile N (NFRAG=4)
 
@@ -251,11 +251,13 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   const int lane = t & 63, wave = t >> 6;
   const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
 
-  // staging assignment: A: 2 threads/row, 4 chunks each;
-  // B: 256/BN_T threads/row, BN_T/32 chunks each.
-  constexpr int CPT_B = BN_T / 32;            // chunks per thread (B)
-  constexpr int TPR_B = 256 / BN_T;           // threads per row (B)
-  const int arow = t >> 1, ak8 = (t & 1) * 32;
+  // staging assignment: chunk = 8 bf16; A: BM*BK/2048 chunks per thread,
+  // B: BN_T*BK/2048; threads tile rows first.
+  constexpr int CPT_A = BM * BK / 8 / THREADS;      // 2 at BK=32
+  constexpr int CPT_B = BN_T * BK / 8 / THREADS;    // 2 / 1
+  constexpr int TPR_A = (BK / 8) / CPT_A;           // threads per A row
+  constexpr int TPR_B = (BK / 8) / CPT_B;
+  const int arow = t / TPR_A, ak8 = (t % TPR_A) * (CPT_A * 8);
   const int brow = t / TPR_B, bk8 = (t % TPR_B) * (CPT_B * 8);
 
   const auto rsa = ga.prep(m0 + arow);
@@ -269,16 +271,16 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
 
   const int nt = (k_hi - k_lo + BK - 1) / BK;
 
-  s16x8 ra[4], rb[CPT_B];
+  s16x8 ra[CPT_A], rb[CPT_B];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) ra[i] = ga.load8(rsa, k_lo + ak8 + i * 8);
+  for (int i = 0; i < CPT_A; ++i) ra[i] = ga.load8(rsa, k_lo + ak8 + i * 8);
 #pragma unroll
   for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, k_lo + bk8 + i * 8);
 
   int p = 0;
   for (int tile = 0; tile < nt; ++tile) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < CPT_A; ++i)
       *reinterpret_cast<s16x8*>(&As[p][arow * LDS_STRIDE + ak8 + i * 8]) =
           ra[i];
 #pragma unroll
@@ -290,13 +292,13 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
     if (tile + 1 < nt) {  // issue next tile's loads under the MFMAs
       int kb = k_lo + (tile + 1) * BK;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) ra[i] = ga.load8(rsa, kb + ak8 + i * 8);
+      for (int i = 0; i < CPT_A; ++i) ra[i] = ga.load8(rsa, kb + ak8 + i * 8);
 #pragma unroll
       for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, kb + bk8 + i * 8);
     }
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[4], bf[NFRAG];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
