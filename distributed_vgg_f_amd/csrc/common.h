@@ -44,6 +44,25 @@ DEV f32x2 bf2x2(u32 packed) {
   return r;
 }
 
+// ---- magic division -----------------------------------------------------
+// Host-precomputed multiply-shift division (dividends < 2^28): replaces
+// ~20-cycle v_rcp-based u32 divides in the gather hot path with one 64-bit
+// multiply + shift.
+struct MDiv {
+  unsigned long long mul;
+  int sh;
+  unsigned int d;
+  MDiv() : mul(1), sh(0), d(1) {}
+  explicit MDiv(unsigned int d_) : d(d_) {
+    int L = 0;
+    while ((1u << L) < d_) ++L;
+    sh = 32 + L;
+    mul = ((1ull << sh) + d_ - 1) / d_;
+  }
+  DEV u32 div(u32 x) const { return u32(((unsigned long long)x * mul) >> sh); }
+  DEV u32 mod(u32 x, u32 q) const { return x - q * d; }
+};
+
 // ---- grid helpers -------------------------------------------------------
 // Bijective XCD-aware block remap (cdna_hip_programming.md section 5, T1):
 // consecutive remapped ids stay on one XCD so neighbor tiles share its L2.

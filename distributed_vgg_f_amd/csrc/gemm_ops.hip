@@ -116,6 +116,51 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
   return out;
 }
 
+__global__ void cast_f32_bf16_2(const float* __restrict__ in,
+                                u16* __restrict__ out, i64 n) {
+  i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = f2bf(in[i]);
+}
+
+// ------------------------- fused conv-backward preamble (K10+K11+K16)
+// One read of dy produces: ReLU-masked dy (NHWC, for dgrad), the padded
+// transpose dyTpad [C][N][OH][OWp] (for wgrad), and fp32 bias-grad partials.
+template <bool RELU, bool WRITE_DYM, bool DB>
+__global__ void bwd_preamble_kernel(const u16* __restrict__ dy,
+                                    const u16* __restrict__ y,
+                                    u16* __restrict__ dym,
+                                    u16* __restrict__ dyT,
+                                    float* __restrict__ db32, int N, int H,
+                                    int W, int C, int Wp) {
+  __shared__ u16 tile[32][33];
+  int nh = blockIdx.x;
+  int w0 = blockIdx.y * 32, c0 = blockIdx.z * 32;
+  int n = nh / H, ih = nh % H;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;
+  float dbacc = 0.f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int iw = w0 + ty + i * 8, c = c0 + tx;
+    u16 v = 0;
+    if (iw < W && c < C) {
+      i64 o = (((i64)n * H + ih) * W + iw) * C + c;
+      v = dy[o];
+      if (RELU && !(bf2f(y[o]) > 0.f)) v = 0;
+      if (WRITE_DYM) dym[o] = v;
+      if (DB) dbacc += bf2f(v);
+    }
+    tile[ty + i * 8][tx] = v;
+  }
+  if (DB) atomicAdd(&db32[c0 + tx], dbacc);
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int c = c0 + ty + i * 8, iw = w0 + tx;
+    if (c < C && iw < W)
+      dyT[(((i64)c * N + n) * H + ih) * Wp + iw] = tile[tx][ty + i * 8];
+  }
+}
+
 // ------------------------------------------------------- split-K finalize
 // out[r][c] = cast(c32[r][c] (+bias[c]) (relu)); out R rows (<= rows of c32)
 template <int EPI>
@@ -306,15 +351,18 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
     // gather straight from NHWC — no im2col materialization
     int Kp = ceil8(Kfull);
     auto wT = transpose2d(w.view({Kfull, Cout}), Kp);  // [Cout][Kp], 0-pad
-    ConvFwdSmallCGather ga{bfp(x), M, H, W, Cin, OH, OW, stride, pad, KW,
-                           Kfull};
+    ConvFwdSmallCGather ga{bfp(x), M, H, W, Cin, stride, pad, KW, Kfull,
+                           MDiv(u32(OH * OW)), MDiv(u32(OW)),
+                           MDiv(u32(Cin)), MDiv(u32(KW))};
     PlainGather gb{bfp(wT), Cout, Kp, Kp};
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
                 Kp, 1);
     return y;
   }
   auto wT = transpose2d(w.view({Kfull, Cout}));  // [Cout][K]
-  ConvFwdGather ga{bfp(x), M, H, W, Cin, OH, OW, stride, pad, KW};
+  ConvFwdGather ga{bfp(x), M, H, W, Cin, stride, pad,
+                   MDiv(u32(OH * OW)), MDiv(u32(OW)), MDiv(u32(Cin)),
+                   MDiv(u32(KW))};
   PlainGather gb{bfp(wT), Cout, Kfull, Kfull};
   int tiles = ceil_div(M, BM) * ceil_div(Cout, BN);
   int sk = pick_splitk(tiles, Kfull);
@@ -343,7 +391,10 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   // Bt[ci][(kh,kw,co)] = w[kh][kw][ci][co]: permute HWIO -> [Cin][KhKw*Cout]
   auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
       .view({Cin, KH * KW * Cout});
-  ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad, KW};
+  TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
+  ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad,
+                     MDiv(u32(H * W)), MDiv(u32(W)), MDiv(u32(Cout)),
+                     MDiv(u32(KW))};
   PlainGather gb{bfp(wD), Cin, K, K};
   int tiles = ceil_div(M, BM) * ceil_div(Cin, BN);
   int sk = pick_splitk(tiles, K);
@@ -378,19 +429,136 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
-    WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, KW, stride,
-                          pad, Kred};
+    WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
+                          pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
+                          MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   } else {
     int Hp = H + 2 * pad;
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');  // [Cin][N][Hp][Wp]
-    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, KW, stride,
-                    Kred};
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
   return dw;
+}
+
+// Fused conv backward (SURVEY.md C-path): preamble -> dgrad GEMM -> wgrad
+// GEMM. Returns {dx (or empty), dw, db (or empty)}.
+std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
+                                   const at::Tensor& x, const at::Tensor& w,
+                                   int64_t stride_, int64_t pad_, bool relu,
+                                   bool need_dx, bool need_db) {
+  check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
+  int stride = int(stride_), pad = int(pad_);
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int KH = w.size(0), KW = w.size(1), Cout = w.size(3);
+  int OH = dy.size(1), OW = dy.size(2);
+  int OWp = ceil8(OW);
+  int M = N * OH * OW, Kfull = KH * KW * Cin;
+  auto s = cur_stream();
+
+  // ---- preamble: masked dy + dyTpad + db32 in one pass over dy
+  i64 tnumel = (i64)Cout * N * OH * OWp;
+  char key[96];
+  snprintf(key, sizeof key, "b%d_%d_%d_%d_%d", dy.get_device(), Cout, N, OH,
+           OWp);
+  auto dyT = cached_zero_buffer(key, tnumel, dy.options())
+      .narrow(0, 0, tnumel).view({Cout, N, OH, OWp});
+  at::Tensor dym = (relu && need_dx) ? at::empty_like(dy) : dy;
+  auto db32 = need_db ? at::zeros({Cout}, dy.options().dtype(at::kFloat))
+                      : at::Tensor();
+  dim3 pgrid(N * OH, ceil_div(OW, 32), ceil_div(Cout, 32));
+  const u16* yp = relu ? bfp(y) : nullptr;
+  float* dbp = need_db ? db32.data_ptr<float>() : nullptr;
+  u16* dymp = (relu && need_dx) ? bfp_mut(dym) : nullptr;
+  u16* dyTp = bfp_mut(dyT);
+  if (relu && need_dx && need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu && need_dx)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu && need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else
+    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  // note: when relu && !need_dx, dgrad is skipped so dym (== dy) is unused;
+  // the masked values only exist in dyT. wgrad/db see masked dy either way.
+
+  // ---- dgrad
+  at::Tensor dx;
+  if (need_dx) {
+    TORCH_CHECK(Cout % 8 == 0 && Cin % 8 == 0, "dgrad channel alignment");
+    int K = KH * KW * Cout;
+    dx = at::empty({N, H, W, Cin}, dy.options());
+    auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
+        .view({Cin, KH * KW * Cout});
+    TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
+    int Mx = N * H * W;
+    ConvDgradGather ga{bfp(dym), Mx, H, W, Cout, OH, OW, stride, pad,
+                       MDiv(u32(H * W)), MDiv(u32(W)), MDiv(u32(Cout)),
+                       MDiv(u32(KW))};
+    PlainGather gb{bfp(wD), Cin, K, K};
+    int tiles = ceil_div(Mx, BM) * ceil_div(Cin, BN);
+    int sk = pick_splitk(tiles, K);
+    if (sk > 1) {
+      auto c32 = at::zeros({Mx, Cin}, dy.options().dtype(at::kFloat));
+      launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Mx, Cin, K, sk);
+      run_finalize(c32, dx, nullptr, 0, Mx, Cin);
+    } else {
+      launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, Mx, Cin, K, 1);
+    }
+  }
+
+  // ---- wgrad (reuses dyT from the preamble)
+  auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
+  auto dw2d = dw.view({Kfull, Cout});
+  int Kred = N * OH * OWp;
+  WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
+  int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  if (Cin % 8 != 0) {
+    WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
+                          pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
+                          MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  } else {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp))};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  }
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+
+  at::Tensor db;
+  if (need_db) {
+    db = at::empty({Cout}, dy.options());
+    hipLaunchKernelGGL(cast_f32_bf16_2, dim3(ceil_div(Cout, 256)), dim3(256),
+                       0, s, db32.data_ptr<float>(), bfp_mut(db), Cout);
+  }
+  return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
 }
 
 }  // namespace dvgg

@@ -54,21 +54,23 @@ struct PlainGather {
 
 // Implicit-GEMM forward A: virtual A[m=(n,oh,ow)][k=(kh,kw,ci)] = x NHWC.
 // Requires Cin % 8 == 0 (chunks stay inside one filter tap) — Cin<8 first
-// layers use ConvFwdSmallCGather instead.
+// layers use ConvFwdSmallCGather instead. All index divisions are
+// host-precomputed magic multiplies (MDiv).
 struct ConvFwdGather {
   const u16* x;
-  int M, H, W, Cin, OH, OW, stride, pad, KW_;
+  int M, H, W, Cin, stride, pad;
+  MDiv dOHOW, dOW, dCin, dKW;
   struct Row { const u16* nrow; int ihb, iwb; bool valid; };
   DEV Row prep(int m) const {
-    int n = m / (OH * OW), r = m % (OH * OW);
-    int oh = r / OW, ow = r % OW;
-    return {x + (i64)n * H * W * Cin, oh * stride - pad, ow * stride - pad,
-            m < M};
+    u32 n = dOHOW.div(m), r = dOHOW.mod(m, n);
+    u32 oh = dOW.div(r), ow = dOW.mod(r, oh);
+    return {x + (i64)n * H * W * Cin, int(oh) * stride - pad,
+            int(ow) * stride - pad, m < M};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
-    int f = k / Cin, c0 = k % Cin;
-    int kh = f / KW_, kw = f % KW_;
-    int ih = rs.ihb + kh, iw = rs.iwb + kw;
+    u32 f = dCin.div(k), c0 = dCin.mod(k, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    int ih = rs.ihb + int(kh), iw = rs.iwb + int(kw);
     if (rs.valid && (unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
       return *reinterpret_cast<const s16x8*>(rs.nrow +
                                              ((i64)ih * W + iw) * Cin + c0);
@@ -80,20 +82,28 @@ struct ConvFwdGather {
 // dy[n][(ih+pad-kh)/s][(iw+pad-kw)/s][co]. Requires Cout % 8 == 0.
 struct ConvDgradGather {
   const u16* dy;
-  int M, H, W, Cout, OH, OW, stride, pad, KW_;
+  int M, H, W, Cout, OH, OW, stride, pad;
+  MDiv dHW, dW_, dCout, dKW;
   struct Row { const u16* nrow; int ihp, iwp; bool valid; };
   DEV Row prep(int m) const {
-    int n = m / (H * W), r = m % (H * W);
-    int ih = r / W, iw = r % W;
-    return {dy + (i64)n * OH * OW * Cout, ih + pad, iw + pad, m < M};
+    u32 n = dHW.div(m), r = dHW.mod(m, n);
+    u32 ih = dW_.div(r), iw = dW_.mod(r, ih);
+    return {dy + (i64)n * OH * OW * Cout, int(ih) + pad, int(iw) + pad,
+            m < M};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
-    int f = k / Cout, c0 = k % Cout;
-    int kh = f / KW_, kw = f % KW_;
-    int ohs = rs.ihp - kh, ows = rs.iwp - kw;
-    int oh = ohs / stride, ow = ows / stride;
-    bool ok = rs.valid && ohs >= 0 && ows >= 0 && oh * stride == ohs &&
-              ow * stride == ows && oh < OH && ow < OW;
+    u32 f = dCout.div(k), c0 = dCout.mod(k, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    int ohs = rs.ihp - int(kh), ows = rs.iwp - int(kw);
+    int oh, ow;
+    bool ok = rs.valid && ohs >= 0 && ows >= 0;
+    if (stride == 1) {          // VGG path: no division at all
+      oh = ohs; ow = ows;
+    } else {
+      oh = ohs / 2; ow = ows / 2;  // zoo uses stride 2 otherwise
+      ok = ok && (ohs & 1) == 0 && (ows & 1) == 0;
+    }
+    ok = ok && oh < OH && ow < OW;
     if (ok)
       return *reinterpret_cast<const s16x8*>(rs.nrow +
                                              ((i64)oh * OW + ow) * Cout + c0);
@@ -106,30 +116,34 @@ struct ConvDgradGather {
 // k-space is Kp = ceil8(KH*KW*Cin); elements beyond Kfull are zero.
 struct ConvFwdSmallCGather {
   const u16* x;
-  int M, H, W, Cin, OH, OW, stride, pad, KW_, Kfull;
+  int M, H, W, Cin, stride, pad, KW_, Kfull;
+  MDiv dOHOW, dOW, dCin, dKW;
   struct Row { const u16* nrow; int ihb, iwb; bool valid; };
   DEV Row prep(int m) const {
-    int n = m / (OH * OW), r = m % (OH * OW);
-    int oh = r / OW, ow = r % OW;
-    return {x + (i64)n * H * W * Cin, oh * stride - pad, ow * stride - pad,
-            m < M};
+    u32 n = dOHOW.div(m), r = dOHOW.mod(m, n);
+    u32 oh = dOW.div(r), ow = dOW.mod(r, oh);
+    return {x + (i64)n * H * W * Cin, int(oh) * stride - pad,
+            int(ow) * stride - pad, m < M};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
     s16x8 v = zero8();
     if (!rs.valid) return v;
-    int f = k / Cin, ci = k - f * Cin;
-    int kh = f / KW_, kw = f - kh * KW_;
+    u32 f = dCin.div(k);
+    int ci = int(dCin.mod(k, f));
+    u32 kh = dKW.div(f);
+    int kw = int(dKW.mod(f, kh));
+    int khi = int(kh);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int kk = k + j;
       if (kk < Kfull) {
-        int ih = rs.ihb + kh, iw = rs.iwb + kw;
+        int ih = rs.ihb + khi, iw = rs.iwb + kw;
         if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
           v[j] = short(rs.nrow[((i64)ih * W + iw) * Cin + ci]);
       }
       if (++ci == Cin) {  // walk (kh,kw,ci) incrementally: no per-elem divs
         ci = 0;
-        if (++kw == KW_) { kw = 0; ++kh; }
+        if (++kw == KW_) { kw = 0; ++khi; }
       }
     }
     return v;
@@ -141,18 +155,20 @@ struct ConvFwdSmallCGather {
 // k-space pads OW to OWp (mult of 8) so chunks never cross rows.
 struct WgradXGather {
   const u16* xT;
-  int Mrows, Cin, N, Hp, Wp, OH, OWp, KW_, stride, Klim;
+  int Mrows, Cin, N, Hp, Wp, OH, OWp, stride, Klim;
+  MDiv dCin, dKW, dOHOWp, dOWp;
   struct Row { const u16* cib; int kh, kw; bool valid; };
   DEV Row prep(int m) const {
-    int f = m / Cin, ci = m - f * Cin;
-    return {xT + (i64)ci * N * Hp * Wp, f / KW_, f % KW_, m < Mrows};
+    u32 f = dCin.div(m), ci = dCin.mod(m, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    return {xT + (i64)ci * N * Hp * Wp, int(kh), int(kw), m < Mrows};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
     if (!rs.valid || k >= Klim) return zero8();
-    int n = k / (OH * OWp), r = k % (OH * OWp);
-    int oh = r / OWp, ow = r % OWp;
-    const u16* base = rs.cib + ((i64)n * Hp + (oh * stride + rs.kh)) * Wp +
-                      (ow * stride + rs.kw);
+    u32 n = dOHOWp.div(k), r = dOHOWp.mod(k, n);
+    u32 oh = dOWp.div(r), ow = dOWp.mod(r, oh);
+    const u16* base = rs.cib + ((i64)n * Hp + (int(oh) * stride + rs.kh)) * Wp +
+                      (int(ow) * stride + rs.kw);
     s16x8 v;
     if (stride == 1) {
       // vector staging despite the odd +kw shift: aligned dword window +
@@ -182,23 +198,25 @@ struct WgradXGather {
 // per-element range predicates; padded-ow k's are zeroed by the B side).
 struct WgradXDirectGather {
   const u16* x;
-  int Mrows, Cin, N, H, W, OH, OWp, KW_, stride, pad, Klim;
+  int Mrows, Cin, N, H, W, OH, OWp, stride, pad, Klim;
+  MDiv dCin, dKW, dOHOWp, dOWp;
   struct Row { const u16* cib; int kh, kw; bool valid; };
   DEV Row prep(int m) const {
-    int f = m / Cin, ci = m - f * Cin;
-    return {x + ci, f / KW_, f % KW_, m < Mrows};
+    u32 f = dCin.div(m), ci = dCin.mod(m, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    return {x + ci, int(kh), int(kw), m < Mrows};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
     s16x8 v = zero8();
     if (!rs.valid || k >= Klim) return v;
-    int n = k / (OH * OWp), r = k % (OH * OWp);
-    int oh = r / OWp, ow = r % OWp;
-    int ih = oh * stride - pad + rs.kh;
+    u32 n = dOHOWp.div(k), r = dOHOWp.mod(k, n);
+    u32 oh = dOWp.div(r), ow0 = dOWp.mod(r, oh);
+    int ih = int(oh) * stride - pad + rs.kh;
     if ((unsigned)ih >= (unsigned)H) return v;
     const u16* row = rs.cib + (((i64)n * H + ih) * W) * Cin;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int iw = (ow + j) * stride - pad + rs.kw;
+      int iw = (int(ow0) + j) * stride - pad + rs.kw;
       if ((unsigned)iw < (unsigned)W) v[j] = short(row[(i64)iw * Cin]);
     }
     return v;

@@ -17,6 +17,9 @@ at::Tensor conv2d_dgrad(const at::Tensor&, const at::Tensor&, int64_t,
                         int64_t, int64_t, int64_t);
 at::Tensor conv2d_wgrad(const at::Tensor&, const at::Tensor&, int64_t,
                         int64_t, int64_t, int64_t);
+std::vector<at::Tensor> conv2d_bwd(const at::Tensor&, const at::Tensor&,
+                                   const at::Tensor&, const at::Tensor&,
+                                   int64_t, int64_t, bool, bool, bool);
 
 // mem.hip
 at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
@@ -56,6 +59,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &dvgg::conv2d_fwd);
   m.def("conv2d_dgrad", &dvgg::conv2d_dgrad);
   m.def("conv2d_wgrad", &dvgg::conv2d_wgrad);
+  m.def("conv2d_bwd", &dvgg::conv2d_bwd);
   m.def("relu_bwd", &dvgg::relu_bwd);
   m.def("add_relu_fwd", &dvgg::add_relu_fwd);
   m.def("bias_grad", &dvgg::bias_grad);

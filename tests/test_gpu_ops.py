@@ -241,3 +241,23 @@ def test_bn_fwd_bwd():
     check(dx, x32.grad, 0.06, "bn_dx")
     check(dgamma, g32.grad, 0.06, "bn_dgamma")
     check(dbeta, b32.grad, 0.06, "bn_dbeta")
+
+
+def test_conv2d_bwd_fused_matches_parts():
+    torch.manual_seed(50)
+    N, H, W, Ci, Co, k, s, p = 2, 13, 13, 256, 256, 3, 1, 1
+    x = rnd(N, H, W, Ci, scale=0.5)
+    w = rnd(k, k, Ci, Co, scale=1.0 / math.sqrt(k * k * Ci))
+    b = rnd(Co, scale=0.1)
+    xg, wg, bg = x.to(DEV), w.to(DEV), b.to(DEV)
+    y = ext.conv2d_fwd(xg, wg, bg, s, p, True)
+    dy = rnd(N, H, W, Co, seed=51).to(DEV)
+    dx, dw, db = ext.conv2d_bwd(dy, y, xg, wg, s, p, True, True, True)
+    # reference: mask then separate ops
+    dym = ext.relu_bwd(dy, y)
+    dx2 = ext.conv2d_dgrad(dym, wg, H, W, s, p)
+    dw2 = ext.conv2d_wgrad(dym, xg, k, k, s, p)
+    db2 = ext.bias_grad(dym)
+    check(dx, dx2.float().cpu(), 0.01, "fused_dx")
+    check(dw, dw2.float().cpu(), 0.02, "fused_dw")
+    check(db, db2.float().cpu(), 0.02, "fused_db")
