@@ -42,19 +42,16 @@ class ConvFunction(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
-        dy = dy.contiguous()
-        if ctx.relu:
-            dy = _ext().relu_bwd(dy, y)          # dy * (y > 0)
-        dx = dw = db = None
-        if ctx.needs_input_grad[0]:
-            dx = _ext().conv2d_dgrad(dy, w, x.shape[1], x.shape[2],
-                                     ctx.stride, ctx.pad)
-        if ctx.needs_input_grad[1]:
-            dw = _ext().conv2d_wgrad(dy, x, w.shape[0], w.shape[1],
-                                     ctx.stride, ctx.pad)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = _ext().bias_grad(dy)
-        return dx, dw, db, None, None, None
+        need_dx = ctx.needs_input_grad[0]
+        need_db = ctx.has_bias and ctx.needs_input_grad[2]
+        # fused path: one pass over dy produces the ReLU-masked dy, the
+        # wgrad transpose and the bias grad, then the dgrad/wgrad GEMMs run
+        # on the shared intermediates (csrc conv2d_bwd: K10/K11 fused into
+        # the K2/K3 preamble)
+        dx, dw, db = _ext().conv2d_bwd(dy.contiguous(), y, x, w, ctx.stride,
+                                       ctx.pad, ctx.relu, need_dx, need_db)
+        return (dx if need_dx else None, dw, db if need_db else None,
+                None, None, None)
 
 
 class LinearFunction(torch.autograd.Function):
