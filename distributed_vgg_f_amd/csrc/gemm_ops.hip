@@ -122,6 +122,17 @@ __global__ void cast_f32_bf16_2(const float* __restrict__ in,
   if (i < n) out[i] = f2bf(in[i]);
 }
 
+// db[c] = bf16( sum over 256 slots of db32p[s][c] )
+__global__ void bias_slot_reduce_kernel(const float* __restrict__ db32p,
+                                        u16* __restrict__ db, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f;
+#pragma unroll 8
+  for (int s = 0; s < 256; ++s) a += db32p[(i64)s * C + c];
+  db[c] = f2bf(a);
+}
+
 // ------------------------- fused conv-backward preamble (K10+K11+K16)
 // One read of dy produces: ReLU-masked dy (NHWC, for dgrad), the padded
 // transpose dyTpad [C][N][OH][OWp] (for wgrad), and fp32 bias-grad partials.
@@ -469,7 +480,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   auto dyT = cached_zero_buffer(key, tnumel, dy.options())
       .narrow(0, 0, tnumel).view({Cout, N, OH, OWp});
   at::Tensor dym = (relu && need_dx) ? at::empty_like(dy) : dy;
-  auto db32 = need_db ? at::zeros({Cout}, dy.options().dtype(at::kFloat))
+  auto db32 = need_db ? at::zeros({256, Cout}, dy.options().dtype(at::kFloat))
                       : at::Tensor();
   dim3 pgrid(N * OH, ceil_div(OW, 32), ceil_div(Cout, 32));
   const u16* yp = relu ? bfp(y) : nullptr;
@@ -555,8 +566,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   at::Tensor db;
   if (need_db) {
     db = at::empty({Cout}, dy.options());
-    hipLaunchKernelGGL(cast_f32_bf16_2, dim3(ceil_div(Cout, 256)), dim3(256),
-                       0, s, db32.data_ptr<float>(), bfp_mut(db), Cout);
+    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                       dim3(256), 0, s, db32.data_ptr<float>(), bfp_mut(db),
+                       Cout);
   }
   return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
 }
