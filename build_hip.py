@@ -41,6 +41,9 @@ def build(verbose: bool = True) -> str:
     py_inc = sysconfig.get_paths()["include"]
     cflags = [
         f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
+        # fp32 atomicAdd must lower to global_atomic_add_f32, not a CAS
+        # loop (split-K accumulation + bias partials depend on it)
+        "-munsafe-fp-atomics",
         f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
         "-D__HIP_PLATFORM_AMD__=1", "-DUSE_ROCM=1",
         "-DTORCH_EXTENSION_NAME=_hipops",
