@@ -144,6 +144,7 @@ __global__ void bwd_preamble_kernel(const u16* __restrict__ dy,
                                     float* __restrict__ db32, int N, int H,
                                     int W, int C, int Wp) {
   __shared__ u16 tile[32][33];
+  __shared__ float dbred[8][32];
   int nh = blockIdx.x;
   int w0 = blockIdx.y * 32, c0 = blockIdx.z * 32;
   int n = nh / H, ih = nh % H;
@@ -162,8 +163,16 @@ __global__ void bwd_preamble_kernel(const u16* __restrict__ dy,
     }
     tile[ty + i * 8][tx] = v;
   }
-  if (DB) atomicAdd(&db32[c0 + tx], dbacc);
+  if (DB) dbred[ty][tx] = dbacc;
   __syncthreads();
+  // ONE atomic per channel per block, into 256-way slotted partials —
+  // per-thread atomics onto C floats serialized the whole kernel
+  if (DB && ty == 0 && c0 + tx < C) {
+    float a = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) a += dbred[j][tx];
+    atomicAdd(&db32[(blockIdx.x & 255) * C + c0 + tx], a);
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int c = c0 + ty + i * 8, iw = w0 + tx;
