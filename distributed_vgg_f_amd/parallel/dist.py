@@ -60,6 +60,8 @@ def init_distributed(backend: str = "auto",
     if backend == "auto":
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if backend == "nccl":
+        # tolerate more ranks than GPUs (T-dist-local smoke: 2 ranks, 1 GPU)
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(local_rank)
     if not dist.is_initialized():
         dist.init_process_group(backend=backend, rank=rank, world_size=world,
