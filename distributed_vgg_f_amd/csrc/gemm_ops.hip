@@ -11,6 +11,10 @@
 
 namespace dvgg {
 
+// defined in mem.hip
+at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
+at::Tensor bias_grad(const at::Tensor&);
+
 static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
@@ -362,6 +366,11 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int OH, OW;
   conv_out_shape(H, W, KH, KW, stride, pad, OH, OW);
   int M = N * OH * OW, Kfull = KH * KW * Cin;
+  if (KH == 1 && KW == 1 && stride == 1 && pad == 0 && Cin % 8 == 0) {
+    // 1x1/s1 conv IS the FC GEMM
+    auto y2 = linear_fwd(x.view({M, Cin}), w.view({Cin, Cout}), bias, relu);
+    return y2.view({N, OH, OW, Cout});
+  }
   auto y = at::empty({N, OH, OW, Cout}, x.options());
   auto y2d = y.view({M, Cout});
   int epi = epi_of(bias.has_value(), relu);
@@ -480,6 +489,20 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int OWp = ceil8(OW);
   int M = N * OH * OW, Kfull = KH * KW * Cin;
   auto s = cur_stream();
+
+  if (KH == 1 && KW == 1 && stride == 1 && pad == 0 && Cin % 8 == 0) {
+    // 1x1/s1 conv backward == FC backward on flattened spatial dims:
+    // no transposed dy staging, no implicit-GEMM gathers
+    auto dy2 = dy.view({M, Cout});
+    if (relu) dy2 = relu_bwd(dy2, y.view({M, Cout}));
+    at::Tensor dx1;
+    if (need_dx)
+      dx1 = linear_dgrad(dy2, w.view({Cin, Cout})).view({N, H, W, Cin});
+    auto dw1 = linear_wgrad(x.view({M, Cin}), dy2).view({1, 1, Cin, Cout});
+    at::Tensor db1;
+    if (need_db) db1 = bias_grad(dy2);
+    return {dx1, dw1, db1};
+  }
 
   // ---- preamble: masked dy + dyTpad + db32 in one pass over dy
   i64 tnumel = (i64)Cout * N * OH * OWp;

@@ -87,6 +87,8 @@ std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor& logits,
 // --------------------------------------------------------- BN fwd (K17)
 // Pass 1: per-channel sum / sumsq via block partials + fp32 atomics.
 // Pass 2: normalize (+affine, optional fused ReLU); saves mean/rstd fp32.
+// partials land in 256 slots per channel (slot = blockIdx.y & 255):
+// direct atomics onto C floats from thousands of blocks serialize.
 __global__ void bn_reduce_kernel(const u16* __restrict__ x,
                                  float* __restrict__ sum,
                                  float* __restrict__ sumsq, i64 R, int C,
@@ -95,13 +97,16 @@ __global__ void bn_reduce_kernel(const u16* __restrict__ x,
   if (c >= C) return;
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
-  float s = 0.f, q = 0.f;
-  for (i64 r = r0; r < r1; ++r) {
-    float v = bf2f(x[r * C + c]);
-    s += v; q += v * v;
+  float s0 = 0.f, s1 = 0.f, q0 = 0.f, q1 = 0.f;
+  i64 r = r0;
+  for (; r + 2 <= r1; r += 2) {
+    float v0 = bf2f(x[r * C + c]), v1 = bf2f(x[(r + 1) * C + c]);
+    s0 += v0; q0 += v0 * v0; s1 += v1; q1 += v1 * v1;
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], q);
+  for (; r < r1; ++r) { float v = bf2f(x[r * C + c]); s0 += v; q0 += v * v; }
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sum[(i64)slot * C + c], s0 + s1);
+  atomicAdd(&sumsq[(i64)slot * C + c], q0 + q1);
 }
 
 __global__ void bn_stats_kernel(const float* __restrict__ sum,
@@ -111,8 +116,14 @@ __global__ void bn_stats_kernel(const float* __restrict__ sum,
                                 float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = sum[c] / float(R);
-  float v = sumsq[c] / float(R) - m * m;
+  float sa = 0.f, qa = 0.f;
+#pragma unroll 8
+  for (int t = 0; t < 256; ++t) {
+    sa += sum[(i64)t * C + c];
+    qa += sumsq[(i64)t * C + c];
+  }
+  float m = sa / float(R);
+  float v = qa / float(R) - m * m;
   mean[c] = m;
   rstd[c] = rsqrtf(fmaxf(v, 0.f) + eps);
 }
@@ -138,7 +149,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   int C = x.size(-1);
   i64 R = x.numel() / C;
   auto f32 = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({C}, f32), sumsq = at::zeros({C}, f32);
+  auto sum = at::zeros({256, C}, f32), sumsq = at::zeros({256, C}, f32);
   auto mean = at::empty({C}, f32), rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
   int rpb = 512;
@@ -184,8 +195,19 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
     float xh = (bf2f(x[r * C + c]) - m) * rs;
     a += d; b += d * xh;
   }
-  atomicAdd(&sdy[c], a);
-  atomicAdd(&sdyx[c], b);
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sdy[(i64)slot * C + c], a);
+  atomicAdd(&sdyx[(i64)slot * C + c], b);
+}
+
+__global__ void bn_slot_reduce_kernel(const float* __restrict__ part,
+                                      float* __restrict__ out, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f;
+#pragma unroll 8
+  for (int t = 0; t < 256; ++t) a += part[(i64)t * C + c];
+  out[c] = a;
 }
 
 __global__ void bn_bwd_apply_kernel(const u16* __restrict__ dy,
@@ -222,7 +244,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   int C = dy.size(-1);
   i64 R = dy.numel() / C;
   auto f32 = dy.options().dtype(at::kFloat);
-  auto sdy = at::zeros({C}, f32), sdyx = at::zeros({C}, f32);
+  auto sdyp = at::zeros({256, C}, f32), sdyxp = at::zeros({256, C}, f32);
+  auto sdy = at::empty({C}, f32), sdyx = at::empty({C}, f32);
   auto dx = at::empty_like(dy);
   auto dgamma = at::empty({C}, dy.options());
   auto dbeta = at::empty({C}, dy.options());
@@ -231,7 +254,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, rgrid, dim3(256), 0, s, bfp(dy),
                      bfp(x), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(), R, C, rpb);
+                     sdyp.data_ptr<float>(), sdyxp.data_ptr<float>(), R, C,
+                     rpb);
+  hipLaunchKernelGGL(bn_slot_reduce_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     0, s, sdyp.data_ptr<float>(), sdy.data_ptr<float>(), C);
+  hipLaunchKernelGGL(bn_slot_reduce_kernel, dim3(ceil_div(C, 256)), dim3(256),
+                     0, s, sdyxp.data_ptr<float>(), sdyx.data_ptr<float>(), C);
   i64 total = dy.numel();
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(int((total + 255) / 256)),
                      dim3(256), 0, s, bfp(dy), bfp(x), bfp(gamma),
