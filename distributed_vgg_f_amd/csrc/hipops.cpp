@@ -47,7 +47,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, double, bool);
 std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
-                               const at::Tensor&);
+                               const at::Tensor&, const at::Tensor&, bool);
 
 }  // namespace dvgg
 

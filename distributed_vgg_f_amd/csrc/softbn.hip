@@ -134,13 +134,25 @@ __global__ void bn_apply_kernel(const u16* __restrict__ x,
                                 const u16* __restrict__ beta,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ rstd,
-                                u16* __restrict__ y, i64 total, int C) {
-  i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total) return;
-  int c = int(i % C);
-  float v = (bf2f(x[i]) - mean[c]) * rstd[c] * bf2f(gamma[c]) + bf2f(beta[c]);
-  if (RELU) v = v > 0.f ? v : 0.f;
-  y[i] = f2bf(v);
+                                u16* __restrict__ y, i64 total8, int C) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= total8) return;
+  int nch = C / 8;
+  int c8 = int(id % nch) * 8;
+  i64 o = (id / nch) * C + c8;
+  s16x8 xv = *reinterpret_cast<const s16x8*>(x + o);
+  s16x8 gv = *reinterpret_cast<const s16x8*>(gamma + c8);
+  s16x8 bv = *reinterpret_cast<const s16x8*>(beta + c8);
+  s16x8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c8 + j;
+    float v = (bf2f(u16(xv[j])) - mean[c]) * rstd[c] * bf2f(u16(gv[j])) +
+              bf2f(u16(bv[j]));
+    if (RELU) v = v > 0.f ? v : 0.f;
+    out[j] = short(f2bf(v));
+  }
+  *reinterpret_cast<s16x8*>(y + o) = out;
 }
 
 std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
@@ -161,25 +173,29 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), R, C,
                      float(eps));
-  i64 total = x.numel();
+  TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
+  i64 total8 = x.numel() / 8;
   if (relu)
-    hipLaunchKernelGGL(bn_apply_kernel<true>, dim3(int((total + 255) / 256)),
+    hipLaunchKernelGGL(bn_apply_kernel<true>, dim3(int((total8 + 255) / 256)),
                        dim3(256), 0, s, bfp(x), bfp(gamma), bfp(beta),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       bfp_mut(y), total, C);
+                       bfp_mut(y), total8, C);
   else
-    hipLaunchKernelGGL(bn_apply_kernel<false>, dim3(int((total + 255) / 256)),
-                       dim3(256), 0, s, bfp(x), bfp(gamma), bfp(beta),
+    hipLaunchKernelGGL(bn_apply_kernel<false>,
+                       dim3(int((total8 + 255) / 256)), dim3(256), 0, s,
+                       bfp(x), bfp(gamma), bfp(beta),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       bfp_mut(y), total, C);
+                       bfp_mut(y), total8, C);
   return {y, mean, rstd};
 }
 
 // --------------------------------------------------------- BN bwd (K17)
 // reduce dy and dy*xhat per channel -> dbeta/dgamma; then
 // dx = rstd*gamma * (dy - mean(dy) - xhat * mean(dy*xhat))
+template <bool RELU>
 __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
                                      const u16* __restrict__ x,
+                                     const u16* __restrict__ yr,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      float* __restrict__ sdy,
@@ -189,15 +205,28 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
   if (c >= C) return;
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
-  float a = 0.f, b = 0.f, m = mean[c], rs = rstd[c];
-  for (i64 r = r0; r < r1; ++r) {
+  float a0 = 0.f, a1 = 0.f, b0 = 0.f, b1 = 0.f;
+  float m = mean[c], rs = rstd[c];
+  i64 r = r0;
+  for (; r + 2 <= r1; r += 2) {
+    float d0 = bf2f(dy[r * C + c]), d1 = bf2f(dy[(r + 1) * C + c]);
+    if (RELU) {
+      d0 = bf2f(yr[r * C + c]) > 0.f ? d0 : 0.f;
+      d1 = bf2f(yr[(r + 1) * C + c]) > 0.f ? d1 : 0.f;
+    }
+    float h0 = (bf2f(x[r * C + c]) - m) * rs;
+    float h1 = (bf2f(x[(r + 1) * C + c]) - m) * rs;
+    a0 += d0; b0 += d0 * h0; a1 += d1; b1 += d1 * h1;
+  }
+  for (; r < r1; ++r) {
     float d = bf2f(dy[r * C + c]);
-    float xh = (bf2f(x[r * C + c]) - m) * rs;
-    a += d; b += d * xh;
+    if (RELU) d = bf2f(yr[r * C + c]) > 0.f ? d : 0.f;
+    float h = (bf2f(x[r * C + c]) - m) * rs;
+    a0 += d; b0 += d * h;
   }
   int slot = blockIdx.y & 255;
-  atomicAdd(&sdy[(i64)slot * C + c], a);
-  atomicAdd(&sdyx[(i64)slot * C + c], b);
+  atomicAdd(&sdy[(i64)slot * C + c], a0 + a1);
+  atomicAdd(&sdyx[(i64)slot * C + c], b0 + b1);
 }
 
 __global__ void bn_slot_reduce_kernel(const float* __restrict__ part,
@@ -210,21 +239,38 @@ __global__ void bn_slot_reduce_kernel(const float* __restrict__ part,
   out[c] = a;
 }
 
+template <bool RELU>
 __global__ void bn_bwd_apply_kernel(const u16* __restrict__ dy,
                                     const u16* __restrict__ x,
+                                    const u16* __restrict__ yr,
                                     const u16* __restrict__ gamma,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
                                     const float* __restrict__ sdy,
                                     const float* __restrict__ sdyx,
-                                    u16* __restrict__ dx, i64 total, int C,
+                                    u16* __restrict__ dx, i64 total8, int C,
                                     float invR) {
-  i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total) return;
-  int c = int(i % C);
-  float xh = (bf2f(x[i]) - mean[c]) * rstd[c];
-  float v = bf2f(dy[i]) - sdy[c] * invR - xh * sdyx[c] * invR;
-  dx[i] = f2bf(v * rstd[c] * bf2f(gamma[c]));
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= total8) return;
+  int nch = C / 8;
+  int c8 = int(id % nch) * 8;
+  i64 o = (id / nch) * C + c8;
+  s16x8 dv = *reinterpret_cast<const s16x8*>(dy + o);
+  s16x8 xv = *reinterpret_cast<const s16x8*>(x + o);
+  s16x8 gv = *reinterpret_cast<const s16x8*>(gamma + c8);
+  s16x8 yv{};
+  if (RELU) yv = *reinterpret_cast<const s16x8*>(yr + o);
+  s16x8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c8 + j;
+    float d = bf2f(u16(dv[j]));
+    if (RELU) d = bf2f(u16(yv[j])) > 0.f ? d : 0.f;
+    float xh = (bf2f(u16(xv[j])) - mean[c]) * rstd[c];
+    float v = d - sdy[c] * invR - xh * sdyx[c] * invR;
+    out[j] = short(f2bf(v * rstd[c] * bf2f(u16(gv[j]))));
+  }
+  *reinterpret_cast<s16x8*>(dx + o) = out;
 }
 
 __global__ void bn_param_grad_kernel(const float* __restrict__ sdy,
@@ -239,9 +285,11 @@ __global__ void bn_param_grad_kernel(const float* __restrict__ sdy,
 
 std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
                                const at::Tensor& gamma, const at::Tensor& mean,
-                               const at::Tensor& rstd) {
+                               const at::Tensor& rstd, const at::Tensor& yr,
+                               bool relu) {
   check_bf16(dy, "dy"); check_bf16(x, "x");
   int C = dy.size(-1);
+  TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = dy.numel() / C;
   auto f32 = dy.options().dtype(at::kFloat);
   auto sdyp = at::zeros({256, C}, f32), sdyxp = at::zeros({256, C}, f32);
@@ -252,20 +300,36 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   int rpb = 512;
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, rgrid, dim3(256), 0, s, bfp(dy),
-                     bfp(x), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     sdyp.data_ptr<float>(), sdyxp.data_ptr<float>(), R, C,
-                     rpb);
+  const u16* yp = relu ? bfp(yr) : nullptr;
+  if (relu)
+    hipLaunchKernelGGL(bn_bwd_reduce_kernel<true>, rgrid, dim3(256), 0, s,
+                       bfp(dy), bfp(x), yp, mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), sdyp.data_ptr<float>(),
+                       sdyxp.data_ptr<float>(), R, C, rpb);
+  else
+    hipLaunchKernelGGL(bn_bwd_reduce_kernel<false>, rgrid, dim3(256), 0, s,
+                       bfp(dy), bfp(x), yp, mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), sdyp.data_ptr<float>(),
+                       sdyxp.data_ptr<float>(), R, C, rpb);
   hipLaunchKernelGGL(bn_slot_reduce_kernel, dim3(ceil_div(C, 256)), dim3(256),
                      0, s, sdyp.data_ptr<float>(), sdy.data_ptr<float>(), C);
   hipLaunchKernelGGL(bn_slot_reduce_kernel, dim3(ceil_div(C, 256)), dim3(256),
                      0, s, sdyxp.data_ptr<float>(), sdyx.data_ptr<float>(), C);
-  i64 total = dy.numel();
-  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(int((total + 255) / 256)),
-                     dim3(256), 0, s, bfp(dy), bfp(x), bfp(gamma),
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
-                     bfp_mut(dx), total, C, float(1.0 / double(R)));
+  i64 total8 = dy.numel() / 8;
+  if (relu)
+    hipLaunchKernelGGL(bn_bwd_apply_kernel<true>,
+                       dim3(int((total8 + 255) / 256)), dim3(256), 0, s,
+                       bfp(dy), bfp(x), yp, bfp(gamma),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                       bfp_mut(dx), total8, C, float(1.0 / double(R)));
+  else
+    hipLaunchKernelGGL(bn_bwd_apply_kernel<false>,
+                       dim3(int((total8 + 255) / 256)), dim3(256), 0, s,
+                       bfp(dy), bfp(x), yp, bfp(gamma),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
+                       bfp_mut(dx), total8, C, float(1.0 / double(R)));
   hipLaunchKernelGGL(bn_param_grad_kernel, dim3(ceil_div(C, 256)), dim3(256),
                      0, s, sdy.data_ptr<float>(), sdyx.data_ptr<float>(),
                      bfp_mut(dgamma), bfp_mut(dbeta), C);

@@ -167,10 +167,9 @@ class BatchNormFunction(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, gamma, mean, rstd, y = ctx.saved_tensors
-        dy = dy.contiguous()
-        if ctx.relu:
-            dy = _ext().relu_bwd(dy, y)
-        dx, dgamma, dbeta = _ext().bn_bwd(dy, x, gamma, mean, rstd)
+        # ReLU mask fused into the BN backward kernels (no separate pass)
+        dx, dgamma, dbeta = _ext().bn_bwd(dy.contiguous(), x, gamma, mean,
+                                          rstd, y, ctx.relu)
         return dx, dgamma, dbeta, None, None
 
 

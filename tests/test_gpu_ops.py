@@ -233,7 +233,7 @@ def test_bn_fwd_bwd():
     check(y, ref, 0.05, "bn_fwd")
     dy = rnd(8, 7, 7, 64, seed=42)
     dx, dgamma, dbeta = ext.bn_bwd(dy.to(DEV), x.to(DEV), g.to(DEV), mean,
-                                   rstd)
+                                   rstd, y, False)
     x32 = x.float().requires_grad_(True)
     g32 = g.float().requires_grad_(True)
     b32 = b.float().requires_grad_(True)
@@ -261,3 +261,20 @@ def test_conv2d_bwd_fused_matches_parts():
     check(dx, dx2.float().cpu(), 0.01, "fused_dx")
     check(dw, dw2.float().cpu(), 0.02, "fused_dw")
     check(db, db2.float().cpu(), 0.02, "fused_db")
+
+
+def test_bn_bwd_fused_relu():
+    torch.manual_seed(43)
+    x = rnd(4, 5, 5, 32, scale=2.0)
+    g, b = rnd(32, seed=44), rnd(32, seed=45)
+    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, True)
+    dy = rnd(4, 5, 5, 32, seed=46)
+    dx, dgamma, dbeta = ext.bn_bwd(dy.to(DEV), x.to(DEV), g.to(DEV), mean,
+                                   rstd, y, True)
+    x32 = x.float().requires_grad_(True)
+    g32 = g.float().requires_grad_(True)
+    b32 = b.float().requires_grad_(True)
+    cpu_ref.batchnorm(x32, g32, b32, relu=True).backward(dy.float())
+    check(dx, x32.grad, 0.06, "bnrelu_dx")
+    check(dgamma, g32.grad, 0.06, "bnrelu_dgamma")
+    check(dbeta, b32.grad, 0.06, "bnrelu_dbeta")
