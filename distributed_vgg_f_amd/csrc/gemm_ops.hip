@@ -273,6 +273,38 @@ static void launch_gemm(const GA& ga, const GB& gb, at::Tensor* outB,
     launch_gemm_t<GA, GB, 4>(ga, gb, outB, c32, bias, epi, M, N, K, splitk);
 }
 
+// TN Plain x Plain launcher (fc / 1x1-conv wgrad): operands in
+// [reduction][col] layout, transposed during LDS staging.
+template <int NFRAG>
+static void launch_gemm_tn_t(const PlainGather& ga, const PlainGather& gb,
+                             at::Tensor* outB, at::Tensor* c32, int M, int N,
+                             int K, int splitk) {
+  constexpr int BN_T = 32 * NFRAG;
+  int tiles = ceil_div(M, BM) * ceil_div(N, BN_T);
+  auto s = cur_stream();
+  if (splitk > 1) {
+    dim3 grid(tiles, splitk);
+    hipLaunchKernelGGL((gemm_kernel<PlainGather, PlainGather, 0, true, NFRAG,
+                                    true, true>), grid, dim3(THREADS), 0, s,
+                       ga, gb, nullptr, c32->data_ptr<float>(), nullptr, M, N,
+                       K, 0, K);
+  } else {
+    hipLaunchKernelGGL((gemm_kernel<PlainGather, PlainGather, 0, false, NFRAG,
+                                    true, true>), dim3(tiles), dim3(THREADS),
+                       0, s, ga, gb, bfp_mut(*outB), nullptr, nullptr, M, N,
+                       K, 0, K);
+  }
+}
+
+static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
+                           at::Tensor* outB, at::Tensor* c32, int M, int N,
+                           int K, int splitk) {
+  if (N <= 64)
+    launch_gemm_tn_t<2>(ga, gb, outB, c32, M, N, K, splitk);
+  else
+    launch_gemm_tn_t<4>(ga, gb, outB, c32, M, N, K, splitk);
+}
+
 static int pick_splitk(int tiles, int K, int target_blocks = 512) {
   if (tiles >= target_blocks) return 1;
   int kt = std::max(1, K / BK);
@@ -330,20 +362,21 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
   check_bf16(x, "x"); check_bf16(dy, "dy");
   int M = x.size(0), K = x.size(1), N = dy.size(1);
   TORCH_CHECK(dy.size(0) == M, "bad fc wgrad shape");
-  int Mp = ceil8(M);           // zero-padded reduction rows contribute 0
-  auto xT = transpose2d(x, Mp);    // [K][Mp]
-  auto dyT = transpose2d(dy, Mp);  // [N][Mp]
+  // TN-staged GEMM: x [M][K] and dy [M][N] consumed directly (reduction =
+  // their outer dim, transposed in LDS during staging) — no global
+  // transpose pass, no reduction padding.
   auto dw = at::empty({K, N}, x.options());
-  PlainGather ga{bfp(xT), K, Mp, Mp};
-  PlainGather gb{bfp(dyT), N, Mp, Mp};
+  int Mred = ((M + BK - 1) / BK) * BK;   // k-loop granularity (rows pred'd)
+  PlainGather ga{bfp(x), M, K, K};       // rows = reduction
+  PlainGather gb{bfp(dy), M, N, N};
   int tiles = ceil_div(K, BM) * ceil_div(N, BN);
-  int sk = pick_splitk(tiles, Mp);
+  int sk = pick_splitk(tiles, Mred);
   if (sk > 1) {
     auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
-    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, K, N, Mp, sk);
+    launch_gemm_tn(ga, gb, nullptr, &c32, K, N, Mred, sk);
     run_finalize(c32, dw, nullptr, 0, K, N);
   } else {
-    launch_gemm(ga, gb, &dw, nullptr, nullptr, 0, K, N, Mp, 1);
+    launch_gemm_tn(ga, gb, &dw, nullptr, K, N, Mred, 1);
   }
   return dw;
 }

@@ -175,17 +175,13 @@ struct WgradXGather {
       // funnel shift (halo pre-zeroed, +8 elem slack for trailing read)
       uintptr_t ap = (uintptr_t)base;
       const u32* q = reinterpret_cast<const u32*>(ap & ~(uintptr_t)3);
-      u32 a0 = q[0], a1 = q[1], a2 = q[2], a3 = q[3];
+      u32 a0 = q[0], a1 = q[1], a2 = q[2], a3 = q[3], a4 = q[4];
+      u32 sh = (ap & 2) ? 16u : 0u;   // branchless: v_alignbit selects
       u32* o = reinterpret_cast<u32*>(&v);
-      if (ap & 2) {
-        u32 a4 = q[4];
-        o[0] = (a0 >> 16) | (a1 << 16);
-        o[1] = (a1 >> 16) | (a2 << 16);
-        o[2] = (a2 >> 16) | (a3 << 16);
-        o[3] = (a3 >> 16) | (a4 << 16);
-      } else {
-        o[0] = a0; o[1] = a1; o[2] = a2; o[3] = a3;
-      }
+      o[0] = __builtin_amdgcn_alignbit(a1, a0, sh);
+      o[1] = __builtin_amdgcn_alignbit(a2, a1, sh);
+      o[2] = __builtin_amdgcn_alignbit(a3, a2, sh);
+      o[3] = __builtin_amdgcn_alignbit(a4, a3, sh);
       return v;
     }
 #pragma unroll
@@ -243,7 +239,12 @@ struct WgradDyGather {
 // SPLITK: atomicAdd fp32 into c32 (epilogue applied by a finalize kernel).
 // NFRAG: B fragments per wave; tile N = 32*NFRAG (4 -> 128, 2 -> 64).
 
-template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4>
+// TNA/TNB: operand supplied in [reduction][col] layout (e.g. fc/1x1-conv
+// wgrad, where both x and dy have the reduction as their OUTER dim); the
+// transpose happens during LDS staging (scalar ds_writes) instead of a
+// separate global transpose pass, and the reduction dim needs no padding.
+template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4,
+          bool TNA = false, bool TNB = false>
 __global__ __launch_bounds__(THREADS) void gemm_kernel(
     GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
     const u16* __restrict__ bias, int M, int N, int K, int k_lo, int k_hi) {
@@ -270,16 +271,20 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
 
   // staging assignment: chunk = 8 bf16; A: BM*BK/2048 chunks per thread,
-  // B: BN_T*BK/2048; threads tile rows first.
+  // B: BN_T*BK/2048; threads tile rows first (k-contiguous operands) or
+  // k-rows first (TN operands: chunk = 8 columns of one reduction row).
   constexpr int CPT_A = BM * BK / 8 / THREADS;      // 2 at BK=32
   constexpr int CPT_B = BN_T * BK / 8 / THREADS;    // 2 / 1
   constexpr int TPR_A = (BK / 8) / CPT_A;           // threads per A row
   constexpr int TPR_B = (BK / 8) / CPT_B;
   const int arow = t / TPR_A, ak8 = (t % TPR_A) * (CPT_A * 8);
   const int brow = t / TPR_B, bk8 = (t % TPR_B) * (CPT_B * 8);
+  // TN assignments: chunk id -> (local k, col8)
+  const int atk = (t * CPT_A) / (BM / 8), atc = ((t * CPT_A) % (BM / 8)) * 8;
+  const int btk = (t * CPT_B) / (BN_T / 8), btc = ((t * CPT_B) % (BN_T / 8)) * 8;
 
-  const auto rsa = ga.prep(m0 + arow);
-  const auto rsb = gb.prep(n0 + brow);
+  const auto rsa = ga.prep(TNA ? 0 : m0 + arow);
+  const auto rsb = gb.prep(TNB ? 0 : n0 + brow);
 
   f32x4 acc[4][NFRAG];
 #pragma unroll
@@ -291,28 +296,50 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
 
   s16x8 ra[CPT_A], rb[CPT_B];
 #pragma unroll
-  for (int i = 0; i < CPT_A; ++i) ra[i] = ga.load8(rsa, k_lo + ak8 + i * 8);
+  for (int i = 0; i < CPT_A; ++i)
+    ra[i] = TNA ? ga.load8(ga.prep(k_lo + atk), m0 + atc + i * 8)
+                : ga.load8(rsa, k_lo + ak8 + i * 8);
 #pragma unroll
-  for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, k_lo + bk8 + i * 8);
+  for (int i = 0; i < CPT_B; ++i)
+    rb[i] = TNB ? gb.load8(gb.prep(k_lo + btk), n0 + btc + i * 8)
+                : gb.load8(rsb, k_lo + bk8 + i * 8);
 
   int p = 0;
   for (int tile = 0; tile < nt; ++tile) {
 #pragma unroll
-    for (int i = 0; i < CPT_A; ++i)
-      *reinterpret_cast<s16x8*>(&As[p][arow * LDS_STRIDE + ak8 + i * 8]) =
-          ra[i];
+    for (int i = 0; i < CPT_A; ++i) {
+      if (TNA) {
 #pragma unroll
-    for (int i = 0; i < CPT_B; ++i)
-      *reinterpret_cast<s16x8*>(&Bs[p][brow * LDS_STRIDE + bk8 + i * 8]) =
-          rb[i];
+        for (int j = 0; j < 8; ++j)
+          As[p][(atc + i * 8 + j) * LDS_STRIDE + atk] = u16(ra[i][j]);
+      } else {
+        *reinterpret_cast<s16x8*>(&As[p][arow * LDS_STRIDE + ak8 + i * 8]) =
+            ra[i];
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < CPT_B; ++i) {
+      if (TNB) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          Bs[p][(btc + i * 8 + j) * LDS_STRIDE + btk] = u16(rb[i][j]);
+      } else {
+        *reinterpret_cast<s16x8*>(&Bs[p][brow * LDS_STRIDE + bk8 + i * 8]) =
+            rb[i];
+      }
+    }
     // ONE barrier per tile: also fences tile-2's readers of buf[p^1]
     __syncthreads();
     if (tile + 1 < nt) {  // issue next tile's loads under the MFMAs
       int kb = k_lo + (tile + 1) * BK;
 #pragma unroll
-      for (int i = 0; i < CPT_A; ++i) ra[i] = ga.load8(rsa, kb + ak8 + i * 8);
+      for (int i = 0; i < CPT_A; ++i)
+        ra[i] = TNA ? ga.load8(ga.prep(kb + atk), m0 + atc + i * 8)
+                    : ga.load8(rsa, kb + ak8 + i * 8);
 #pragma unroll
-      for (int i = 0; i < CPT_B; ++i) rb[i] = gb.load8(rsb, kb + bk8 + i * 8);
+      for (int i = 0; i < CPT_B; ++i)
+        rb[i] = TNB ? gb.load8(gb.prep(kb + btk), n0 + btc + i * 8)
+                    : gb.load8(rsb, kb + bk8 + i * 8);
     }
     const int kfrag = (lane >> 4) * 8;
 #pragma unroll
