@@ -186,39 +186,55 @@ __global__ void bwd_preamble_kernel(const u16* __restrict__ dy,
 }
 
 // ------------------------------------------------------- split-K finalize
-// out[r][c] = cast(c32[r][c] (+bias[c]) (relu)); out R rows (<= rows of c32)
+// out[id] = cast(sum over slices of c32[z][id] (+bias[c]) (relu));
+// c32 is [slices][R][C], 8 fp32 per thread per slice.
 template <int EPI>
 __global__ void finalize_kernel(const float* __restrict__ c32,
                                 u16* __restrict__ out,
                                 const u16* __restrict__ bias, i64 total,
-                                int C) {
-  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (id >= total) return;
-  float v = c32[id];
-  if (EPI == 1 || EPI == 2) v += bf2f(bias[id % C]);
-  if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
-  out[id] = f2bf(v);
+                                int C, int slices, i64 slice_stride) {
+  i64 id8 = ((i64)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (id8 >= total) return;
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  for (int z = 0; z < slices; ++z) {
+    const float* p = c32 + (i64)z * slice_stride + id8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (id8 + j < total) acc[j] += p[j];
+  }
+  for (int j = 0; j < 8; ++j) {
+    i64 id = id8 + j;
+    if (id >= total) break;
+    float v = acc[j];
+    if (EPI == 1 || EPI == 2) v += bf2f(bias[id % C]);
+    if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
+    out[id] = f2bf(v);
+  }
 }
 
 static void run_finalize(const at::Tensor& c32, at::Tensor& out,
-                         const at::Tensor* bias, int epi, int R, int C) {
+                         const at::Tensor* bias, int epi, int R, int C,
+                         int slices = 1) {
   i64 total = (i64)R * C;
-  int blocks = int((total + 255) / 256);
+  i64 sstride = (i64)R * C;
+  int blocks = int((total / 8 + 256) / 256);
   const u16* bp = bias ? bfp(*bias) : nullptr;
   auto s = cur_stream();
   switch (epi) {
     case 0: hipLaunchKernelGGL(finalize_kernel<0>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C); break;
+                               total, C, slices, sstride); break;
     case 1: hipLaunchKernelGGL(finalize_kernel<1>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C); break;
+                               total, C, slices, sstride); break;
     case 2: hipLaunchKernelGGL(finalize_kernel<2>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C); break;
+                               total, C, slices, sstride); break;
     default: hipLaunchKernelGGL(finalize_kernel<3>, dim3(blocks), dim3(256),
                                 0, s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                                total, C); break;
+                                total, C, slices, sstride); break;
   }
 }
 
@@ -330,9 +346,9 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   int epi = epi_of(bias.has_value(), relu);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::zeros({M, N}, x.options().dtype(at::kFloat));
+    auto c32 = at::empty({sk, M, N}, x.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, N, K, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N, sk);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, N, K, 1);
   }
@@ -349,9 +365,9 @@ at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
   int tiles = ceil_div(M, BM) * ceil_div(K, BN);
   int sk = pick_splitk(tiles, N);
   if (sk > 1) {
-    auto c32 = at::zeros({M, K}, dy.options().dtype(at::kFloat));
+    auto c32 = at::empty({sk, M, K}, dy.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, K, N, sk);
-    run_finalize(c32, dx, nullptr, 0, M, K);
+    run_finalize(c32, dx, nullptr, 0, M, K, sk);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, K, N, 1);
   }
@@ -372,9 +388,9 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
   int tiles = ceil_div(K, BM) * ceil_div(N, BN);
   int sk = pick_splitk(tiles, Mred);
   if (sk > 1) {
-    auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
+    auto c32 = at::empty({sk, K, N}, x.options().dtype(at::kFloat));
     launch_gemm_tn(ga, gb, nullptr, &c32, K, N, Mred, sk);
-    run_finalize(c32, dw, nullptr, 0, K, N);
+    run_finalize(c32, dw, nullptr, 0, K, N, sk);
   } else {
     launch_gemm_tn(ga, gb, &dw, nullptr, K, N, Mred, 1);
   }
@@ -429,9 +445,9 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int tiles = ceil_div(M, BM) * ceil_div(Cout, BN);
   int sk = pick_splitk(tiles, Kfull);
   if (sk > 1) {
-    auto c32 = at::zeros({M, Cout}, x.options().dtype(at::kFloat));
+    auto c32 = at::empty({sk, M, Cout}, x.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cout, Kfull, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout, sk);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
                 Kfull, 1);
@@ -461,9 +477,9 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   int tiles = ceil_div(M, BM) * ceil_div(Cin, BN);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::zeros({M, Cin}, dy.options().dtype(at::kFloat));
+    auto c32 = at::empty({sk, M, Cin}, dy.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cin, K, sk);
-    run_finalize(c32, dx, nullptr, 0, M, Cin);
+    run_finalize(c32, dx, nullptr, 0, M, Cin, sk);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, Cin, K, 1);
   }
@@ -489,7 +505,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = at::empty({sk, Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
@@ -504,7 +520,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
                     MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, sk);
   return dw;
 }
 
@@ -596,9 +612,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     int tiles = ceil_div(Mx, BM) * ceil_div(Cin, BN);
     int sk = pick_splitk(tiles, K);
     if (sk > 1) {
-      auto c32 = at::zeros({Mx, Cin}, dy.options().dtype(at::kFloat));
+      auto c32 = at::empty({sk, Mx, Cin}, dy.options().dtype(at::kFloat));
       launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Mx, Cin, K, sk);
-      run_finalize(c32, dx, nullptr, 0, Mx, Cin);
+      run_finalize(c32, dx, nullptr, 0, Mx, Cin, sk);
     } else {
       launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, Mx, Cin, K, 1);
     }
@@ -611,7 +627,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = at::empty({sk, Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
@@ -626,7 +642,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                     MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, sk);
 
   at::Tensor db;
   if (need_db) {

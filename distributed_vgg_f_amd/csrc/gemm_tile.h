@@ -382,7 +382,9 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
         if (row >= M) continue;
         float v = acc[i][j][r];
         if (SPLITK) {
-          atomicAdd(&c32[(i64)row * N + col], v);
+          // per-slice fp32 slab (plain stores); a finalize kernel sums the
+          // slices — split-K atomics serialized at this volume
+          c32[(i64)blockIdx.y * M * N + (i64)row * N + col] = v;
         } else {
           v += bv;
           if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
