@@ -325,7 +325,9 @@ static int pick_splitk(int tiles, int K, int target_blocks = 512) {
   if (tiles >= target_blocks) return 1;
   int kt = std::max(1, K / BK);
   int want = std::min(kt, std::max(1, target_blocks / std::max(1, tiles)));
-  return want;
+  // collapse to the number of NON-EMPTY slices at this span
+  int span = (kt + want - 1) / want;
+  return (kt + span - 1) / span;
 }
 
 static int epi_of(bool has_bias, bool relu) {

@@ -263,7 +263,10 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
     int span = (kt + nchunk - 1) / nchunk * BK;
     k_lo = blockIdx.y * span;
     k_hi = min(K, k_lo + span);
-    if (k_lo >= k_hi) return;
+    // NO early return on an empty span: the slab epilogue must still write
+    // zeros (slabs are not pre-zeroed); nt <= 0 skips the k-loop. The host
+    // avoids empty slices, this is the safety net.
+    if (k_lo > k_hi) k_hi = k_lo;
   }
 
   const int t = threadIdx.x;
@@ -295,6 +298,7 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   const int nt = (k_hi - k_lo + BK - 1) / BK;
 
   s16x8 ra[CPT_A], rb[CPT_B];
+  if (nt > 0) {
 #pragma unroll
   for (int i = 0; i < CPT_A; ++i)
     ra[i] = TNA ? ga.load8(ga.prep(k_lo + atk), m0 + atc + i * 8)
@@ -303,6 +307,7 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   for (int i = 0; i < CPT_B; ++i)
     rb[i] = TNB ? gb.load8(gb.prep(k_lo + btk), n0 + btc + i * 8)
                 : gb.load8(rsb, k_lo + bk8 + i * 8);
+  }
 
   int p = 0;
   for (int tile = 0; tile < nt; ++tile) {
