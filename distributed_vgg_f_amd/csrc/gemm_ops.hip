@@ -348,9 +348,9 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   int epi = epi_of(bias.has_value(), relu);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::empty({sk, M, N}, x.options().dtype(at::kFloat));
+    auto c32 = at::zeros({M, N}, x.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, N, K, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N, sk);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, N, K, 1);
   }
@@ -367,9 +367,9 @@ at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
   int tiles = ceil_div(M, BM) * ceil_div(K, BN);
   int sk = pick_splitk(tiles, N);
   if (sk > 1) {
-    auto c32 = at::empty({sk, M, K}, dy.options().dtype(at::kFloat));
+    auto c32 = at::zeros({M, K}, dy.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, K, N, sk);
-    run_finalize(c32, dx, nullptr, 0, M, K, sk);
+    run_finalize(c32, dx, nullptr, 0, M, K);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, K, N, 1);
   }
@@ -390,9 +390,9 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
   int tiles = ceil_div(K, BM) * ceil_div(N, BN);
   int sk = pick_splitk(tiles, Mred);
   if (sk > 1) {
-    auto c32 = at::empty({sk, K, N}, x.options().dtype(at::kFloat));
+    auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
     launch_gemm_tn(ga, gb, nullptr, &c32, K, N, Mred, sk);
-    run_finalize(c32, dw, nullptr, 0, K, N, sk);
+    run_finalize(c32, dw, nullptr, 0, K, N);
   } else {
     launch_gemm_tn(ga, gb, &dw, nullptr, K, N, Mred, 1);
   }
@@ -447,9 +447,9 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int tiles = ceil_div(M, BM) * ceil_div(Cout, BN);
   int sk = pick_splitk(tiles, Kfull);
   if (sk > 1) {
-    auto c32 = at::empty({sk, M, Cout}, x.options().dtype(at::kFloat));
+    auto c32 = at::zeros({M, Cout}, x.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cout, Kfull, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout, sk);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
                 Kfull, 1);
@@ -479,9 +479,9 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   int tiles = ceil_div(M, BM) * ceil_div(Cin, BN);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::empty({sk, M, Cin}, dy.options().dtype(at::kFloat));
+    auto c32 = at::zeros({M, Cin}, dy.options().dtype(at::kFloat));
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cin, K, sk);
-    run_finalize(c32, dx, nullptr, 0, M, Cin, sk);
+    run_finalize(c32, dx, nullptr, 0, M, Cin);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, Cin, K, 1);
   }
@@ -507,7 +507,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::empty({sk, Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
@@ -522,7 +522,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
                     MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, sk);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
   return dw;
 }
 
@@ -614,9 +614,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     int tiles = ceil_div(Mx, BM) * ceil_div(Cin, BN);
     int sk = pick_splitk(tiles, K);
     if (sk > 1) {
-      auto c32 = at::empty({sk, Mx, Cin}, dy.options().dtype(at::kFloat));
+      auto c32 = at::zeros({Mx, Cin}, dy.options().dtype(at::kFloat));
       launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Mx, Cin, K, sk);
-      run_finalize(c32, dx, nullptr, 0, Mx, Cin, sk);
+      run_finalize(c32, dx, nullptr, 0, Mx, Cin);
     } else {
       launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, Mx, Cin, K, 1);
     }
@@ -629,7 +629,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::empty({sk, Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
@@ -644,7 +644,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                     MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, sk);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
 
   at::Tensor db;
   if (need_db) {

@@ -263,10 +263,7 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
     int span = (kt + nchunk - 1) / nchunk * BK;
     k_lo = blockIdx.y * span;
     k_hi = min(K, k_lo + span);
-    // NO early return on an empty span: the slab epilogue must still write
-    // zeros (slabs are not pre-zeroed); nt <= 0 skips the k-loop. The host
-    // avoids empty slices, this is the safety net.
-    if (k_lo > k_hi) k_hi = k_lo;
+    if (k_lo >= k_hi) return;  // empty slice: c32 is pre-zeroed
   }
 
   const int t = threadIdx.x;
@@ -387,9 +384,10 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
         if (row >= M) continue;
         float v = acc[i][j][r];
         if (SPLITK) {
-          // per-slice fp32 slab (plain stores); a finalize kernel sums the
-          // slices — split-K atomics serialized at this volume
-          c32[(i64)blockIdx.y * M * N + (i64)row * N + col] = v;
+          // fp32 atomics measure FASTER than per-slice slabs + summing
+          // finalize on gfx950 (native global_atomic_add_f32; slabs cost
+          // an extra write+read of sk x tile bytes) — A/B'd both ways
+          atomicAdd(&c32[(i64)row * N + col], v);
         } else {
           v += bv;
           if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
