@@ -167,23 +167,52 @@ struct WgradXGather {
     if (!rs.valid || k >= Klim) return zero8();
     u32 n = dOHOWp.div(k), r = dOHOWp.mod(k, n);
     u32 oh = dOWp.div(r), ow = dOWp.mod(r, oh);
-    const u16* base = rs.cib + ((i64)n * Hp + (int(oh) * stride + rs.kh)) * Wp +
-                      (int(ow) * stride + rs.kw);
     s16x8 v;
-    if (stride == 1) {
-      // vector staging despite the odd +kw shift: aligned dword window +
-      // funnel shift (halo pre-zeroed, +8 elem slack for trailing read)
-      uintptr_t ap = (uintptr_t)base;
-      const u32* q = reinterpret_cast<const u32*>(ap & ~(uintptr_t)3);
-      u32 a0 = q[0], a1 = q[1], a2 = q[2], a3 = q[3], a4 = q[4];
-      u32 sh = (ap & 2) ? 16u : 0u;   // branchless: v_alignbit selects
+    if (stride == 1 && KW_ <= 8) {
+      // The row base (cib + ...*Wp) is 16B-aligned and ow is a multiple of
+      // 8, so the only misalignment is +kw elements (kw < 8). Read the
+      // aligned 32B window with TWO b128 loads (the kernel is VMEM-issue
+      // bound; 5 dword loads cost 2.5x the TA slots) and extract 8 elems
+      // with alignbit. Halo is pre-zeroed; buffer has +8 elem slack.
+      const u16* A = rs.cib + ((i64)n * Hp + (oh + rs.kh)) * Wp + ow;
+      s16x8 lo = *reinterpret_cast<const s16x8*>(A);
+      s16x8 hi = *reinterpret_cast<const s16x8*>(A + 8);
+      u32 w[8];
+      *reinterpret_cast<s16x8*>(&w[0]) = lo;
+      *reinterpret_cast<s16x8*>(&w[4]) = hi;
       u32* o = reinterpret_cast<u32*>(&v);
-      o[0] = __builtin_amdgcn_alignbit(a1, a0, sh);
-      o[1] = __builtin_amdgcn_alignbit(a2, a1, sh);
-      o[2] = __builtin_amdgcn_alignbit(a3, a2, sh);
-      o[3] = __builtin_amdgcn_alignbit(a4, a3, sh);
+      int kw = rs.kw;
+      u32 r = u32(kw & 1) * 16u;
+      switch (kw >> 1) {  // row-constant: at most 3 targets per wave
+        case 0:
+          o[0] = __builtin_amdgcn_alignbit(w[1], w[0], r);
+          o[1] = __builtin_amdgcn_alignbit(w[2], w[1], r);
+          o[2] = __builtin_amdgcn_alignbit(w[3], w[2], r);
+          o[3] = __builtin_amdgcn_alignbit(w[4], w[3], r);
+          break;
+        case 1:
+          o[0] = __builtin_amdgcn_alignbit(w[2], w[1], r);
+          o[1] = __builtin_amdgcn_alignbit(w[3], w[2], r);
+          o[2] = __builtin_amdgcn_alignbit(w[4], w[3], r);
+          o[3] = __builtin_amdgcn_alignbit(w[5], w[4], r);
+          break;
+        case 2:
+          o[0] = __builtin_amdgcn_alignbit(w[3], w[2], r);
+          o[1] = __builtin_amdgcn_alignbit(w[4], w[3], r);
+          o[2] = __builtin_amdgcn_alignbit(w[5], w[4], r);
+          o[3] = __builtin_amdgcn_alignbit(w[6], w[5], r);
+          break;
+        default:
+          o[0] = __builtin_amdgcn_alignbit(w[4], w[3], r);
+          o[1] = __builtin_amdgcn_alignbit(w[5], w[4], r);
+          o[2] = __builtin_amdgcn_alignbit(w[6], w[5], r);
+          o[3] = __builtin_amdgcn_alignbit(w[7], w[6], r);
+          break;
+      }
       return v;
     }
+    const u16* base = rs.cib + ((i64)n * Hp + (int(oh) * stride + rs.kh)) * Wp +
+                      (int(ow) * stride + rs.kw);
 #pragma unroll
     for (int j = 0; j < 8; ++j) v[j] = short(base[(i64)j * stride]);
     return v;
