@@ -168,7 +168,7 @@ struct WgradXGather {
     u32 n = dOHOWp.div(k), r = dOHOWp.mod(k, n);
     u32 oh = dOWp.div(r), ow = dOWp.mod(r, oh);
     s16x8 v;
-    if (stride == 1 && KW_ <= 8) {
+    if (stride == 1 && dKW.d <= 8) {
       // The row base (cib + ...*Wp) is 16B-aligned and ow is a multiple of
       // 8, so the only misalignment is +kw elements (kw < 8). Read the
       // aligned 32B window with TWO b128 loads (the kernel is VMEM-issue
