@@ -120,40 +120,6 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
   return out;
 }
 
-// zero-halo NHWC pad copy for the per-tap wgrad: xp[n][ih+p][iw+p][c] =
-// x[n][ih][iw][c]; halo stays zero (cached zeroed buffer).
-__global__ void pad_nhwc_kernel(const u16* __restrict__ x,
-                                u16* __restrict__ xp, int N, int H, int W,
-                                int C, int Hp, int Wp, int p) {
-  i64 row_elems = (i64)W * C;
-  i64 id8 = ((i64)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  i64 total = (i64)N * H * row_elems;
-  if (id8 >= total) return;
-  i64 nh = id8 / row_elems;
-  i64 off = id8 % row_elems;
-  int n = int(nh / H), ih = int(nh % H);
-  const u16* src = x + nh * row_elems + off;
-  u16* dst = xp + (((i64)n * Hp + ih + p) * Wp + p) * C + off;
-  *reinterpret_cast<s16x8*>(dst) = *reinterpret_cast<const s16x8*>(src);
-}
-
-static at::Tensor pad_nhwc(const at::Tensor& x, int p) {
-  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
-  if (p == 0) return x;
-  int Hp = H + 2 * p, Wp = W + 2 * p;
-  i64 numel = (i64)N * Hp * Wp * C;
-  char key[96];
-  snprintf(key, sizeof key, "p%d_%d_%d_%d_%d", x.get_device(), N, Hp, Wp, C);
-  auto xp = cached_zero_buffer(key, numel, x.options())
-      .narrow(0, 0, numel).view({N, Hp, Wp, C});
-  i64 total = (i64)N * H * W * C;
-  TORCH_CHECK((W * C) % 8 == 0, "pad_nhwc needs W*C %% 8 == 0");
-  hipLaunchKernelGGL(pad_nhwc_kernel, dim3(int((total / 8 + 255) / 256)),
-                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(xp), N, H, W,
-                     C, Hp, Wp, p);
-  return xp;
-}
-
 __global__ void cast_f32_bf16_2(const float* __restrict__ in,
                                 u16* __restrict__ out, i64 n) {
   i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
@@ -346,18 +312,6 @@ static void launch_gemm_tn_t(const PlainGather& ga, const PlainGather& gb,
   }
 }
 
-template <int NFRAG>
-static void launch_wgrad_tap_t(const WgradTapGather& ga, const PlainGather& gb,
-                               float* c32, int M, int N, int K, int splitk) {
-  constexpr int BN_T = 32 * NFRAG;
-  int tiles = ceil_div(M, BM) * ceil_div(N, BN_T);
-  dim3 grid(tiles, splitk);
-  hipLaunchKernelGGL((gemm_kernel<WgradTapGather, PlainGather, 0, true, NFRAG,
-                                  true, true>), grid, dim3(THREADS), 0,
-                     cur_stream(), ga, gb, nullptr, c32, nullptr, M, N, K, 0,
-                     K);
-}
-
 static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
                            at::Tensor* outB, at::Tensor* c32, int M, int N,
                            int K, int splitk) {
@@ -443,37 +397,6 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
     launch_gemm_tn(ga, gb, &dw, nullptr, K, N, Mred, 1);
   }
   return dw;
-}
-
-// Per-tap wgrad: KH*KW TN GEMMs over padded NHWC x, accumulating into one
-// fp32 c32 [KH*KW*Cin][Cout] (atomics; finalize casts to dw).
-static void wgrad_by_taps(const at::Tensor& x, const at::Tensor& dym, int KH,
-                          int KW, int stride, int pad, at::Tensor& dw2d,
-                          int Cout) {
-  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
-  int OH = dym.size(1), OW = dym.size(2);
-  int Mred = N * OH * OW;
-  int Kred = ((Mred + BK - 1) / BK) * BK;
-  auto xp = pad_nhwc(x, pad);
-  int Hp = H + 2 * pad, Wp = W + 2 * pad;
-  int Kfull = KH * KW * Cin;
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-  PlainGather gb{bfp(dym), Mred, Cout, Cout};
-  int tiles = ceil_div(Cin, BM) * ceil_div(Cout, Cout <= 64 ? 64 : 128);
-  int sk = std::max(2, pick_splitk(tiles, Kred, 1024 / (KH * KW) + 64));
-  MDiv dOHOW{u32(OH * OW)}, dOW{u32(OW)};
-  for (int kh = 0; kh < KH; ++kh)
-    for (int kw = 0; kw < KW; ++kw) {
-      WgradTapGather ga{bfp(xp), Mred, Hp, Wp, Cin, stride, kh, kw,
-                        dOHOW, dOW};
-      float* cp = c32.data_ptr<float>() +
-                  (i64)(kh * KW + kw) * Cin * Cout;
-      if (Cout <= 64)
-        launch_wgrad_tap_t<2>(ga, gb, cp, Cin, Cout, Kred, sk);
-      else
-        launch_wgrad_tap_t<4>(ga, gb, cp, Cin, Cout, Kred, sk);
-    }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
 }
 
 // ------------------------------------------------------------------ conv
@@ -575,15 +498,9 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
 
-  if (Cin % 8 == 0) {
-    // per-tap TN GEMMs over zero-padded NHWC x: all vector loads, no
-    // transposed staging buffers, reduction space exactly N*OH*OW
-    auto dyc = dy.contiguous();
-    wgrad_by_taps(x, dyc, KH, KW, stride, pad, dw2d, Cout);
-    return dw;
-  }
-  // Cin < 8 (conv1/stem): element-wise gather straight from NHWC x against
-  // the transposed, ow-padded dy
+  // One generic path for every conv (any Cin, any stride): transpose x and
+  // dy once into gather-friendly padded layouts, then split-K implicit GEMM
+  // over them (conv1's Cin=3 included: the WgradX gather is element-wise).
   int OWp = ceil8(OW);
   int Kred = N * OH * OWp;
   auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
@@ -591,10 +508,20 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-  WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
-                        pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
-                        MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
-  launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  if (Cin % 8 != 0) {
+    WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
+                          pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
+                          MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  } else {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');  // [Cin][N][Hp][Wp]
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp))};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
   return dw;
 }
@@ -610,11 +537,13 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
   int KH = w.size(0), KW = w.size(1), Cout = w.size(3);
   int OH = dy.size(1), OW = dy.size(2);
+  int OWp = ceil8(OW);
   int M = N * OH * OW, Kfull = KH * KW * Cin;
   auto s = cur_stream();
 
   if (KH == 1 && KW == 1 && stride == 1 && pad == 0 && Cin % 8 == 0) {
-    // 1x1/s1 conv backward == FC backward on flattened spatial dims
+    // 1x1/s1 conv backward == FC backward on flattened spatial dims:
+    // no transposed dy staging, no implicit-GEMM gathers
     auto dy2 = dy.view({M, Cout});
     if (relu) dy2 = relu_bwd(dy2, y.view({M, Cout}));
     at::Tensor dx1;
@@ -626,20 +555,57 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     return {dx1, dw1, db1};
   }
 
-  // masked dy once (K10); bias grad from it (K11)
-  at::Tensor dym = relu ? relu_bwd(dy, y) : dy;
-  at::Tensor db;
-  if (need_db) db = bias_grad(dym.view({M, Cout}));
+  // ---- preamble: masked dy + dyTpad + db32 in one pass over dy
+  i64 tnumel = (i64)Cout * N * OH * OWp;
+  char key[96];
+  snprintf(key, sizeof key, "b%d_%d_%d_%d_%d", dy.get_device(), Cout, N, OH,
+           OWp);
+  auto dyT = cached_zero_buffer(key, tnumel, dy.options())
+      .narrow(0, 0, tnumel).view({Cout, N, OH, OWp});
+  at::Tensor dym = (relu && need_dx) ? at::empty_like(dy) : dy;
+  auto db32 = need_db ? at::zeros({256, Cout}, dy.options().dtype(at::kFloat))
+                      : at::Tensor();
+  dim3 pgrid(N * OH, ceil_div(OW, 32), ceil_div(Cout, 32));
+  const u16* yp = relu ? bfp(y) : nullptr;
+  float* dbp = need_db ? db32.data_ptr<float>() : nullptr;
+  u16* dymp = (relu && need_dx) ? bfp_mut(dym) : nullptr;
+  u16* dyTp = bfp_mut(dyT);
+  if (relu && need_dx && need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu && need_dx)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu && need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (relu)
+    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else if (need_db)
+    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, true>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  else
+    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, false>), pgrid,
+                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
+                       OW, Cout, OWp);
+  // note: when relu && !need_dx, dgrad is skipped so dym (== dy) is unused;
+  // the masked values only exist in dyT. wgrad/db see masked dy either way.
 
-  // ---- dgrad (K2)
+  // ---- dgrad
   at::Tensor dx;
   if (need_dx) {
     TORCH_CHECK(Cout % 8 == 0 && Cin % 8 == 0, "dgrad channel alignment");
-    TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
     int K = KH * KW * Cout;
     dx = at::empty({N, H, W, Cin}, dy.options());
     auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
         .view({Cin, KH * KW * Cout});
+    TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
     int Mx = N * H * W;
     ConvDgradGather ga{bfp(dym), Mx, H, W, Cout, OH, OW, stride, pad,
                        MDiv(u32(H * W)), MDiv(u32(W)), MDiv(u32(Cout)),
@@ -656,26 +622,36 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     }
   }
 
-  // ---- wgrad (K3)
+  // ---- wgrad (reuses dyT from the preamble)
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
-  if (Cin % 8 == 0) {
-    auto dymc = dym.contiguous();
-    wgrad_by_taps(x, dymc, KH, KW, stride, pad, dw2d, Cout);
-  } else {
-    // conv1/stem: element-wise gather from NHWC x vs transposed padded dy
-    int OWp = ceil8(OW);
-    int Kred = N * OH * OWp;
-    auto dyT = nhwc2cnhw_pad(dym, OH, OWp, 0, 0, 'd');
-    WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
-    int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
-    int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-    auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  int Kred = N * OH * OWp;
+  WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
+  int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
                           MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
-    run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  } else {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
+    WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp))};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  }
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+
+  at::Tensor db;
+  if (need_db) {
+    db = at::empty({Cout}, dy.options());
+    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                       dim3(256), 0, s, db32.data_ptr<float>(), bfp_mut(db),
+                       Cout);
   }
   return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
 }

@@ -248,30 +248,6 @@ struct WgradXDirectGather {
   }
 };
 
-// Per-tap wgrad A (TN mode): for one filter tap (kh,kw),
-// dw[kh][kw][ci][co] = sum over m=(n,oh,ow) of
-//   xpad[n][oh*s+kh][ow*s+kw][ci] * dy[m][co].
-// Used as a TN operand: prep(m) resolves the reduction row's pointer into
-// the zero-padded NHWC x; load8 reads 8 consecutive ci (16B vectors).
-struct WgradTapGather {
-  const u16* xp;           // [N][Hp][Wp][C], zero halo
-  int Mred, Hp, Wp, C, stride, kh, kw;
-  MDiv dOHOW, dOW;
-  struct Row { const u16* rp; bool valid; };
-  DEV Row prep(int m) const {
-    u32 n = dOHOW.div(m), r = dOHOW.mod(m, n);
-    u32 oh = dOW.div(r), ow = dOW.mod(r, oh);
-    const u16* rp = xp + (((i64)n * Hp + (int(oh) * stride + kh)) * Wp +
-                          (int(ow) * stride + kw)) * C;
-    return {rp, m < Mred};
-  }
-  DEV s16x8 load8(const Row& r, int ci) const {
-    if (r.valid && ci < C)
-      return *reinterpret_cast<const s16x8*>(r.rp + ci);
-    return zero8();
-  }
-};
-
 // Wgrad B: virtual Bt[n'=co][k=(n,oh,ow-padded)] over dyTpad
 // [Cout][N][OH][OWp] (zero-filled pad, 16B rows) — clean vector loads.
 struct WgradDyGather {
