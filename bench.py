@@ -75,8 +75,11 @@ def main():
     ips = global_batch * args.steps / elapsed
     ms = 1000.0 * elapsed / args.steps
     if ctx.is_chief:
+        model_label = {"vggf": "VGG-F", "vgg16": "VGG-16",
+                       "resnet50": "ResNet-50"}.get(args.model, args.model)
         print(json.dumps({
-            "metric": "images/sec (whole node) VGG-F 224x224 bs=256/GPU",
+            "metric": f"images/sec (whole node) {model_label} 224x224 "
+                      f"bs={args.batch_size}/GPU",
             "value": round(ips, 1),
             "unit": "images/sec",
             "n_gpus": n_gpus,

@@ -164,7 +164,9 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   auto sum = at::zeros({256, C}, f32), sumsq = at::zeros({256, C}, f32);
   auto mean = at::empty({C}, f32), rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
-  int rpb = 512;
+  // target ~2048 blocks regardless of C (late ResNet layers have small R,
+  // huge C; early ones the opposite)
+  int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
@@ -297,7 +299,7 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dx = at::empty_like(dy);
   auto dgamma = at::empty({C}, dy.options());
   auto dbeta = at::empty({C}, dy.options());
-  int rpb = 512;
+  int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   const u16* yp = relu ? bfp(yr) : nullptr;
