@@ -506,7 +506,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
-  int sk = std::max(2, pick_splitk(tiles, Kred, 384));
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
@@ -628,7 +628,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int Kred = N * OH * OWp;
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
-  int sk = std::max(2, pick_splitk(tiles, Kred, 384));
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
