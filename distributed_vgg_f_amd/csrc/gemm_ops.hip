@@ -239,6 +239,10 @@ static void run_finalize(const at::Tensor& c32, at::Tensor& out,
 }
 
 // ------------------------------------------------------------ gemm driver
+static const u16* zero_page(const at::TensorOptions& opt) {
+  return bfp(cached_zero_buffer("zpage", 64, opt));
+}
+
 template <class GA, class GB, int NFRAG>
 static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
                           at::Tensor* c32, const at::Tensor* bias, int epi,
@@ -247,6 +251,41 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
   int tiles = ceil_div(M, BM) * ceil_div(N, BN_T);
   auto s = cur_stream();
   const u16* bp = bias ? bfp(*bias) : nullptr;
+  constexpr bool GLDS = GldsOk<GA>::value && GldsOk<GB>::value;
+  if constexpr (GLDS) {
+    const u16* zp = zero_page(outB ? outB->options() : c32->options()
+                                  .dtype(at::kBFloat16));
+    if (splitk > 1) {
+      dim3 grid(tiles, splitk);
+      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 0, true, NFRAG>), grid,
+                         dim3(THREADS), 0, s, ga, gb, nullptr,
+                         c32->data_ptr<float>(), nullptr, zp, M, N, K, 0, K);
+      return;
+    }
+    dim3 grid(tiles);
+    switch (epi) {
+      case 0:
+        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 0, false, NFRAG>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, nullptr, zp, M, N, K, 0, K);
+        return;
+      case 1:
+        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 1, false, NFRAG>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, bp, zp, M, N, K, 0, K);
+        return;
+      case 2:
+        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 2, false, NFRAG>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, bp, zp, M, N, K, 0, K);
+        return;
+      default:
+        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 3, false, NFRAG>), grid,
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, nullptr, zp, M, N, K, 0, K);
+        return;
+    }
+  }
   if (splitk > 1) {
     dim3 grid(tiles, splitk);
     hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, NFRAG>), grid,

@@ -50,7 +50,15 @@ struct PlainGather {
       return *reinterpret_cast<const s16x8*>(r.rp + k);
     return zero8();
   }
+  DEV const u16* ptr8(const Row& r, int k, const u16* zpage) const {
+    return (r.valid && k < cols) ? r.rp + k : zpage;
+  }
 };
+
+// which gathers can stage via global_load_lds (single 16B load or a zero
+// page per chunk; funnel/scalar gathers cannot)
+template <class G> struct GldsOk { static constexpr bool value = false; };
+template <> struct GldsOk<PlainGather> { static constexpr bool value = true; };
 
 // Implicit-GEMM forward A: virtual A[m=(n,oh,ow)][k=(kh,kw,ci)] = x NHWC.
 // Requires Cin % 8 == 0 (chunks stay inside one filter tap) — Cin<8 first
@@ -76,7 +84,16 @@ struct ConvFwdGather {
                                              ((i64)ih * W + iw) * Cin + c0);
     return zero8();
   }
+  DEV const u16* ptr8(const Row& rs, int k, const u16* zpage) const {
+    u32 f = dCin.div(k), c0 = dCin.mod(k, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    int ih = rs.ihb + int(kh), iw = rs.iwb + int(kw);
+    if (rs.valid && (unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
+      return rs.nrow + ((i64)ih * W + iw) * Cin + c0;
+    return zpage;
+  }
 };
+template <> struct GldsOk<ConvFwdGather> { static constexpr bool value = true; };
 
 // Implicit-GEMM dgrad A: virtual A[m=(n,ih,iw)][k=(kh,kw,co)] =
 // dy[n][(ih+pad-kh)/s][(iw+pad-kw)/s][co]. Requires Cout % 8 == 0.
@@ -109,7 +126,20 @@ struct ConvDgradGather {
                                              ((i64)oh * OW + ow) * Cout + c0);
     return zero8();
   }
+  DEV const u16* ptr8(const Row& rs, int k, const u16* zpage) const {
+    u32 f = dCout.div(k), c0 = dCout.mod(k, f);
+    u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
+    int ohs = rs.ihp - int(kh), ows = rs.iwp - int(kw);
+    int oh, ow;
+    bool ok = rs.valid && ohs >= 0 && ows >= 0;
+    if (stride == 1) { oh = ohs; ow = ows; }
+    else { oh = ohs / 2; ow = ows / 2;
+           ok = ok && (ohs & 1) == 0 && (ows & 1) == 0; }
+    ok = ok && oh < OH && ow < OW;
+    return ok ? rs.nrow + ((i64)oh * OW + ow) * Cout + c0 : zpage;
+  }
 };
+template <> struct GldsOk<ConvDgradGather> { static constexpr bool value = true; };
 
 // Forward A for Cin % 8 != 0 first layers (conv1 Cin=3, ResNet stem):
 // per-element scalar gather straight from NHWC x — no im2col materialized.
@@ -416,6 +446,135 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
           // fp32 atomics measure FASTER than per-slice slabs + summing
           // finalize on gfx950 (native global_atomic_add_f32; slabs cost
           // an extra write+read of sk x tile bytes) — A/B'd both ways
+          atomicAdd(&c32[(i64)row * N + col], v);
+        } else {
+          v += bv;
+          if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
+          C[(i64)row * N + col] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// ------------------------------------------------- glds-staged variant
+// For operands whose chunks are single 16B loads (ConvFwd/ConvDgrad/Plain):
+// stage via __builtin_amdgcn_global_load_lds with PER-LANE source addresses
+// (predicated-off chunks point at a 16B zero page). No staging VGPRs, no
+// ds_writes; LDS is lane-linear [row][BK] (64 B rows; the residual 4-way
+// ds_read_b128 conflict is the known cost of linear LDS — ladder step 3).
+// ONE __shared__ arena (a second __shared__ object makes hipcc drain the
+// glds queue before every ds_read — cdna_hip_programming.md section 5
+// ".s-level traps" (a)); __syncthreads() itself emits the vmcnt(0) that
+// completes the in-flight tile.
+template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4>
+__global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
+    GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
+    const u16* __restrict__ bias, const u16* __restrict__ zpage, int M, int N,
+    int K, int k_lo, int k_hi) {
+  constexpr int BN_T = 32 * NFRAG;
+  constexpr int AE = BM * BK;            // A elems per buffer
+  constexpr int BE = BN_T * BK;
+  __shared__ u16 smem[2][AE + BE];
+
+  const int tiles_n = (N + BN_T - 1) / BN_T;
+  const int tiles_m = (M + BM - 1) / BM;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int tm = wg / tiles_n, tn = wg % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN_T;
+  if (SPLITK) {
+    int nchunk = gridDim.y;
+    int kt = (K + BK - 1) / BK;
+    int span = (kt + nchunk - 1) / nchunk * BK;
+    k_lo = blockIdx.y * span;
+    k_hi = min(K, k_lo + span);
+    if (k_lo >= k_hi) return;
+  }
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
+
+  // per-lane glds slots: instruction g of wave w lands 1 KiB at
+  // (w*GI + g)*512 elems; lane covers elems +lane*8
+  constexpr int AGI = AE * 2 / 1024 / 4;   // glds instrs per wave (A)
+  constexpr int BGI = BE * 2 / 1024 / 4;
+  typename GA::Row rsa[AGI];
+  int ak8[AGI];
+  u16* alds[AGI];
+#pragma unroll
+  for (int g = 0; g < AGI; ++g) {
+    int off = (wave * AGI + g) * 512 + lane * 8;
+    rsa[g] = ga.prep(m0 + (off >> 5));
+    ak8[g] = off & 31;
+    alds[g] = &smem[0][(wave * AGI + g) * 512];
+  }
+  typename GB::Row rsb[BGI];
+  int bk8[BGI];
+  u16* blds[BGI];
+#pragma unroll
+  for (int g = 0; g < BGI; ++g) {
+    int off = (wave * BGI + g) * 512 + lane * 8;
+    rsb[g] = gb.prep(n0 + (off >> 5));
+    bk8[g] = off & 31;
+    blds[g] = &smem[0][AE + (wave * BGI + g) * 512];
+  }
+  constexpr int BUF = AE + BE;
+
+#define STAGE_GLDS(pbuf, kb)                                                   do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < AGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + ak8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (alds[g] + (pbuf) * BUF), 16, 0, 0);                               _Pragma("unroll")                                                            for (int g = 0; g < BGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb.ptr8(rsb[g], (kb) + bk8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (blds[g] + (pbuf) * BUF), 16, 0, 0);                             } while (0)
+
+  f32x4 acc[4][NFRAG];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = (k_hi - k_lo + BK - 1) / BK;
+  STAGE_GLDS(0, k_lo);
+  __syncthreads();          // vmcnt(0) inside: tile 0 landed
+
+  int p = 0;
+  for (int tile = 0; tile < nt; ++tile) {
+    if (tile + 1 < nt) STAGE_GLDS(p ^ 1, k_lo + (tile + 1) * BK);
+    const int kfrag = (lane >> 4) * 8;
+    bf16x8 af[4], bf[NFRAG];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wm + i * 16 + (lane & 15);
+      af[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          &smem[p][row * BK + kfrag]));
+    }
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = wn + j * 16 + (lane & 15);
+      bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          &smem[p][AE + col * BK + kfrag]));
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < NFRAG; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf[j], acc[i][j], 0, 0, 0);
+    __syncthreads();        // drains next tile's glds; frees buf p
+    p ^= 1;
+  }
+#undef STAGE_GLDS
+
+  // epilogue (same as gemm_kernel)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = n0 + wn + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = (EPI == 1 || EPI == 2) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row_base + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r];
+        if (SPLITK) {
           atomicAdd(&c32[(i64)row * N + col], v);
         } else {
           v += bv;
