@@ -438,6 +438,30 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
   return dw;
 }
 
+// Channel pad for Cin < 8 first layers: x [N,H,W,C] -> [N,H,W,8] with zero
+// fill, so conv1/stem take the vector-gather + glds GEMM paths instead of
+// scalar gathers (weights padded likewise; padded channels produce zero
+// grads/outputs).
+__global__ void pad_channels_kernel(const u16* __restrict__ x,
+                                    u16* __restrict__ out, i64 npos, int C) {
+  i64 pos = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (pos >= npos) return;
+  const u16* src = x + pos * C;
+  s16x8 v{0, 0, 0, 0, 0, 0, 0, 0};
+  for (int c = 0; c < C; ++c) v[c] = short(src[c]);
+  *reinterpret_cast<s16x8*>(out + pos * 8) = v;
+}
+
+static at::Tensor pad_channels8(const at::Tensor& x) {
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  auto out = at::empty({N, H, W, 8}, x.options());
+  i64 npos = (i64)N * H * W;
+  hipLaunchKernelGGL(pad_channels_kernel, dim3(int((npos + 255) / 256)),
+                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(out), npos,
+                     C);
+  return out;
+}
+
 // ------------------------------------------------------------------ conv
 static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
                            int& OH, int& OW) {
@@ -465,19 +489,15 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   auto y2d = y.view({M, Cout});
   int epi = epi_of(bias.has_value(), relu);
 
-  if (Cin % 8 != 0) {
-    // first-layer path (conv1 Cin=3, ResNet stem): scalar per-element
-    // gather straight from NHWC — no im2col materialization
-    int Kp = ceil8(Kfull);
-    auto wT = transpose2d(w.view({Kfull, Cout}), Kp);  // [Cout][Kp], 0-pad
-    ConvFwdSmallCGather ga{bfp(x), M, H, W, Cin, stride, pad, KW, Kfull,
-                           MDiv(u32(OH * OW)), MDiv(u32(OW)),
-                           MDiv(u32(Cin)), MDiv(u32(KW))};
-    PlainGather gb{bfp(wT), Cout, Kp, Kp};
-    launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
-                Kp, 1);
-    return y;
+  if (Cin < 8) {
+    // first layers (conv1 Cin=3, ResNet stem): pad channels to 8 and take
+    // the vector-gather glds path; zero channels contribute nothing
+    auto x8 = pad_channels8(x);
+    auto w8 = at::zeros({KH, KW, 8, Cout}, w.options());
+    w8.narrow(2, 0, Cin).copy_(w);
+    return conv2d_fwd(x8, w8, bias, stride_, pad_, relu);
   }
+  TORCH_CHECK(Cin % 8 == 0, "conv fwd needs Cin % 8 == 0 (or Cin < 8)");
   auto wT = transpose2d(w.view({Kfull, Cout}));  // [Cout][K]
   ConvFwdGather ga{bfp(x), M, H, W, Cin, stride, pad,
                    MDiv(u32(OH * OW)), MDiv(u32(OW)), MDiv(u32(Cin)),
@@ -537,9 +557,16 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
 
-  // One generic path for every conv (any Cin, any stride): transpose x and
-  // dy once into gather-friendly padded layouts, then split-K implicit GEMM
-  // over them (conv1's Cin=3 included: the WgradX gather is element-wise).
+  if (Cin < 8) {
+    // conv1/stem: pad channels to 8, recurse, slice the grad back
+    auto x8 = pad_channels8(x);
+    auto dw8 = conv2d_wgrad(dy, x8, KH_, KW_, stride_, pad_);
+    dw.copy_(dw8.narrow(2, 0, Cin));
+    return dw;
+  }
+
+  // Generic path: transpose x and dy once into gather-friendly padded
+  // layouts, then split-K implicit GEMM over them.
   int OWp = ceil8(OW);
   int Kred = N * OH * OWp;
   auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
@@ -592,6 +619,20 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor db1;
     if (need_db) db1 = bias_grad(dy2);
     return {dx1, dw1, db1};
+  }
+
+  if (Cin < 8) {
+    // conv1/stem backward: pad channels to 8, recurse, slice dw back
+    // (dx is never needed for a first layer)
+    TORCH_CHECK(!need_dx, "dx is not supported for Cin<8 first layers");
+    auto x8 = pad_channels8(x);
+    auto w8 = at::zeros({KH, KW, 8, Cout}, w.options());
+    w8.narrow(2, 0, Cin).copy_(w);
+    auto outs = conv2d_bwd(dy, y, x8, w8, stride_, pad_, relu, false,
+                           need_db);
+    auto dw1 = at::empty({KH, KW, Cin, Cout}, w.options());
+    dw1.copy_(outs[1].narrow(2, 0, Cin));
+    return {at::Tensor(), dw1, outs[2]};
   }
 
   // ---- preamble: masked dy + dyTpad + db32 in one pass over dy
