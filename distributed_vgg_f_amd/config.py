@@ -60,6 +60,7 @@ class Config:
     resume: bool = False
     log_every: int = 10
     metrics_path: Optional[str] = None  # JSONL metrics
+    phase_timers: bool = False          # hipEvent per-phase ms (fwd/bwd/opt)
 
     # debug / fault injection
     fault_inject_step: int = -1

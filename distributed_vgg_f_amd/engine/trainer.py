@@ -60,7 +60,7 @@ class Trainer:
                                           self.device, compute_dtype,
                                           cfg.seed + self.ctx.rank)
         self.metrics = MetricsLogger(cfg.metrics_path, self.ctx.rank)
-        self.timers = PhaseTimers(self.device, enabled=False)
+        self.timers = PhaseTimers(self.device, enabled=cfg.phase_timers)
         self.step_i = 0
         self.global_batch = cfg.batch_size * self.ctx.world_size
 
@@ -80,13 +80,20 @@ class Trainer:
                 f"fault-inject: rank {self.ctx.rank} dies at step {self.step_i}")
         images, labels = self.pipeline.next()
         self.reducer.prepare_step()
+        tm = self.timers
+        tm.start("fwd")
         logits = self.model(images)
         loss = ops.softmax_xent(logits, labels).sum()
+        tm.stop("fwd")
+        tm.start("bwd")
         loss.backward()
         self.reducer.finish()
+        tm.stop("bwd")
         # loss summed per sample; SUM all-reduce over ranks => scale by 1/GB
+        tm.start("opt")
         self.arena.step(grad_scale=1.0 / self.global_batch)
         self.arena.zero_grad()
+        tm.stop("opt")
         self.step_i += 1
         return loss
 
@@ -108,7 +115,7 @@ class Trainer:
                                  loss=float(last_loss) / cfg.batch_size,
                                  images_per_sec=ips,
                                  ms_per_step=1000.0 * dt / window_steps,
-                                 lr=cfg.lr)
+                                 lr=cfg.lr, **self.timers.collect())
                 window_t0 = time.time()
                 window_steps = 0
             if (cfg.checkpoint_every and cfg.checkpoint_dir and
