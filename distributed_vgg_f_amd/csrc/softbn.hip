@@ -161,7 +161,8 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   int C = x.size(-1);
   i64 R = x.numel() / C;
   auto f32 = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({256, C}, f32), sumsq = at::zeros({256, C}, f32);
+  auto part = at::zeros({2, 256, C}, f32);   // one fill for both partials
+  auto sum = part[0], sumsq = part[1];
   auto mean = at::empty({C}, f32), rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
   // target ~2048 blocks regardless of C (late ResNet layers have small R,
@@ -294,7 +295,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = dy.numel() / C;
   auto f32 = dy.options().dtype(at::kFloat);
-  auto sdyp = at::zeros({256, C}, f32), sdyxp = at::zeros({256, C}, f32);
+  auto bpart = at::zeros({2, 256, C}, f32);
+  auto sdyp = bpart[0], sdyxp = bpart[1];
   auto sdy = at::empty({C}, f32), sdyx = at::empty({C}, f32);
   auto dx = at::empty_like(dy);
   auto dgamma = at::empty({C}, dy.options());
