@@ -75,15 +75,18 @@ class Config:
 
     @staticmethod
     def add_args(p: argparse.ArgumentParser) -> None:
+        # SUPPRESS defaults: only explicitly-passed flags land in the
+        # namespace, so precedence is dataclass default < JSON file < CLI
         for f in dataclasses.fields(Config):
             name = "--" + f.name.replace("_", "-")
             if f.type == "bool" or isinstance(f.default, bool):
                 p.add_argument(name, type=lambda s: s.lower() in ("1", "true", "yes"),
-                               default=f.default)
+                               default=argparse.SUPPRESS)
             elif f.default is None:
-                p.add_argument(name, type=str, default=None)
+                p.add_argument(name, type=str, default=argparse.SUPPRESS)
             else:
-                p.add_argument(name, type=type(f.default), default=f.default)
+                p.add_argument(name, type=type(f.default),
+                               default=argparse.SUPPRESS)
         p.add_argument("--config-json", type=str, default=None,
                        help="JSON file of Config overrides (CLI wins)")
 
@@ -94,7 +97,7 @@ class Config:
             with open(args.config_json) as fh:
                 base = json.load(fh)
         names = {f.name for f in dataclasses.fields(Config)}
-        cfg = {**base}
+        cfg = {k: v for k, v in base.items() if k in names}
         for k, v in vars(args).items():
             if k in names:
                 cfg[k] = v

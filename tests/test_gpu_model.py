@@ -82,3 +82,47 @@ def test_eval_runs():
     t = Trainer(cfg, device=DEV)
     res = t.evaluate(2)
     assert 0.0 <= res["accuracy"] <= 1.0
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from distributed_vgg_f_amd.engine import checkpoint as ckpt
+    cfg = Config(model="vggf", batch_size=4, train_steps=2, log_every=0,
+                 checkpoint_dir=str(tmp_path))
+    t = Trainer(cfg, device=DEV)
+    t.train(2)
+    w_before = t.arena.w32.clone()
+    cfg2 = Config(model="vggf", batch_size=4, log_every=0,
+                  checkpoint_dir=str(tmp_path), resume=True)
+    t2 = Trainer(cfg2, device=DEV)
+    assert t2.step_i == 2
+    assert torch.equal(t2.arena.w32, w_before)
+    # and it keeps training from there
+    loss = float(t2.train_step())
+    assert loss == loss
+
+
+def test_uniform_blocks_gpu_step():
+    cfg = Config(model="vggf", batch_size=4, train_steps=1, log_every=0,
+                 uniform_blocks=True)
+    t = Trainer(cfg, device=DEV)
+    loss = float(t.train_step()) / cfg.batch_size
+    assert loss == loss
+
+
+def test_dropout_model_gpu_step():
+    cfg = Config(model="vggf", batch_size=4, train_steps=1, log_every=0,
+                 dropout=0.5)
+    t = Trainer(cfg, device=DEV)
+    loss = float(t.train_step()) / cfg.batch_size
+    assert loss == loss
+
+
+def test_phase_timers_gpu(tmp_path):
+    mpath = str(tmp_path / "m.jsonl")
+    cfg = Config(model="vggf", batch_size=4, train_steps=2, log_every=2,
+                 phase_timers=True, metrics_path=mpath)
+    t = Trainer(cfg, device=DEV)
+    t.train(2)
+    import json
+    rows = [json.loads(l) for l in open(mpath)]
+    assert any("ms_fwd" in r for r in rows), rows

@@ -89,3 +89,29 @@ def test_vgg16_resnet50_forward_small():
         y = model(torch.randn(1, 64, 64, 3))
         assert y.shape == (1, 1000), name
         y.square().mean().backward()
+
+
+def test_metrics_jsonl(tmp_path):
+    import json
+    mpath = str(tmp_path / "metrics.jsonl")
+    cfg = small_cfg(batch_size=2, train_steps=2, log_every=1, image_size=64,
+                    metrics_path=mpath)
+    t = Trainer(cfg)
+    t.train(2)
+    rows = [json.loads(l) for l in open(mpath)]
+    assert len(rows) >= 2
+    assert all("loss" in r and "images_per_sec" in r for r in rows)
+
+
+def test_config_json_roundtrip(tmp_path):
+    import argparse
+    import json
+    cfgfile = tmp_path / "c.json"
+    cfgfile.write_text(json.dumps({"model": "vgg16", "lr": 0.5}))
+    p = argparse.ArgumentParser()
+    Config.add_args(p)
+    args = p.parse_args(["--config-json", str(cfgfile), "--batch-size", "7"])
+    cfg = Config.from_args(args)
+    assert cfg.model == "vgg16" and cfg.batch_size == 7
+    # precedence: dataclass default < JSON < explicit CLI
+    assert cfg.lr == 0.5
