@@ -44,7 +44,9 @@ at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, int64_t, int64_t,
 // softbn.hip
 std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor&, const at::Tensor&);
 std::vector<at::Tensor> bn_fwd(const at::Tensor&, const at::Tensor&,
-                               const at::Tensor&, double, bool);
+                               const at::Tensor&, double, bool,
+                               const c10::optional<at::Tensor>&,
+                               const c10::optional<at::Tensor>&, double);
 std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&, bool);

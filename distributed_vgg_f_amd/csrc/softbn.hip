@@ -112,8 +112,10 @@ __global__ void bn_reduce_kernel(const u16* __restrict__ x,
 __global__ void bn_stats_kernel(const float* __restrict__ sum,
                                 const float* __restrict__ sumsq,
                                 float* __restrict__ mean,
-                                float* __restrict__ rstd, i64 R, int C,
-                                float eps) {
+                                float* __restrict__ rstd,
+                                float* __restrict__ run_mean,
+                                float* __restrict__ run_var, i64 R, int C,
+                                float eps, float momentum) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float sa = 0.f, qa = 0.f;
@@ -123,9 +125,13 @@ __global__ void bn_stats_kernel(const float* __restrict__ sum,
     qa += sumsq[(i64)t * C + c];
   }
   float m = sa / float(R);
-  float v = qa / float(R) - m * m;
+  float v = fmaxf(qa / float(R) - m * m, 0.f);
   mean[c] = m;
-  rstd[c] = rsqrtf(fmaxf(v, 0.f) + eps);
+  rstd[c] = rsqrtf(v + eps);
+  if (run_mean) {  // running-stat update fused in (eval-mode parity)
+    run_mean[c] = momentum * run_mean[c] + (1.f - momentum) * m;
+    run_var[c] = momentum * run_var[c] + (1.f - momentum) * v;
+  }
 }
 
 template <bool RELU>
@@ -156,7 +162,10 @@ __global__ void bn_apply_kernel(const u16* __restrict__ x,
 }
 
 std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
-                               const at::Tensor& beta, double eps, bool relu) {
+                               const at::Tensor& beta, double eps, bool relu,
+                               const c10::optional<at::Tensor>& run_mean,
+                               const c10::optional<at::Tensor>& run_var,
+                               double momentum) {
   check_bf16(x, "x"); check_bf16(gamma, "gamma"); check_bf16(beta, "beta");
   int C = x.size(-1);
   i64 R = x.numel() / C;
@@ -172,10 +181,12 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(), R, C, rpb);
+  float* rmp = run_mean ? run_mean->data_ptr<float>() : nullptr;
+  float* rvp = run_var ? run_var->data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(bn_stats_kernel, dim3(ceil_div(C, 256)), dim3(256), 0, s,
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(), R, C,
-                     float(eps));
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), rmp, rvp,
+                     R, C, float(eps), float(momentum));
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 total8 = x.numel() / 8;
   if (relu)

@@ -101,7 +101,9 @@ class BatchNorm(nn.Module):
 
     def forward(self, x):
         if self.training:
-            return ops.batchnorm(x, self.weight, self.bias, self.eps, self.relu)
+            return ops.batchnorm(x, self.weight, self.bias, self.eps,
+                                 self.relu, self.running_mean,
+                                 self.running_var, self.momentum)
         rstd = torch.rsqrt(self.running_var.to(x.dtype) + self.eps)
         y = (x - self.running_mean.to(x.dtype)) * rstd * self.weight + self.bias
         return torch.relu(y) if self.relu else y

@@ -65,11 +65,20 @@ def softmax_xent(logits, labels):
     return F.cross_entropy(logits.float(), labels, reduction="none")
 
 
-def batchnorm(x, gamma, beta, eps: float = 1e-5, relu: bool = False):
+def batchnorm(x, gamma, beta, eps: float = 1e-5, relu: bool = False,
+              run_mean=None, run_var=None, momentum: float = 0.9):
     if x.is_cuda:
         from .autograd import BatchNormFunction
-        return BatchNormFunction.apply(x, gamma, beta, eps, relu)
-    return cpu_ref.batchnorm(x, gamma, beta, eps, relu)
+        return BatchNormFunction.apply(x, gamma, beta, eps, relu,
+                                       run_mean, run_var, momentum)
+    y = cpu_ref.batchnorm(x, gamma, beta, eps, relu)
+    if run_mean is not None:
+        with torch.no_grad():
+            m = x.float().mean(dim=(0, 1, 2))
+            v = x.float().var(dim=(0, 1, 2), unbiased=False)
+            run_mean.mul_(momentum).add_(m, alpha=1 - momentum)
+            run_var.mul_(momentum).add_(v, alpha=1 - momentum)
+    return y
 
 
 def add_relu(a, b):

@@ -157,8 +157,10 @@ class BatchNormFunction(torch.autograd.Function):
     """K17 (ResNet-50): training-mode BN over N,H,W with optional fused ReLU."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps: float, relu: bool):
-        y, mean, rstd = _ext().bn_fwd(x, gamma, beta, eps, relu)
+    def forward(ctx, x, gamma, beta, eps: float, relu: bool,
+                run_mean=None, run_var=None, momentum: float = 0.9):
+        y, mean, rstd = _ext().bn_fwd(x, gamma, beta, eps, relu,
+                                      run_mean, run_var, momentum)
         ctx.save_for_backward(x, gamma, mean, rstd,
                               y if relu else torch.empty(0, device=x.device))
         ctx.relu = relu
@@ -170,7 +172,7 @@ class BatchNormFunction(torch.autograd.Function):
         # ReLU mask fused into the BN backward kernels (no separate pass)
         dx, dgamma, dbeta = _ext().bn_bwd(dy.contiguous(), x, gamma, mean,
                                           rstd, y, ctx.relu)
-        return dx, dgamma, dbeta, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 class AddReluFunction(torch.autograd.Function):

@@ -126,3 +126,26 @@ def test_phase_timers_gpu(tmp_path):
     import json
     rows = [json.loads(l) for l in open(mpath)]
     assert any("ms_fwd" in r for r in rows), rows
+
+
+@pytest.mark.parametrize("name,bs", [("vgg16", 2), ("resnet50", 2)])
+def test_model_parity_vs_cpu(name, bs):
+    """model-level T-model parity for the zoo beyond VGG-F: logits cosine
+    between the GPU bf16 HIP path and the CPU fp32 reference."""
+    cfg = Config(model=name, batch_size=bs, seed=11, image_size=96)
+    torch.manual_seed(cfg.seed)
+    model = build_model(cfg)
+    x = torch.randn(bs, 96, 96, 3) * 0.5
+    ref = model(x).float()
+    gm = build_model(cfg)
+    gm.load_state_dict(model.state_dict())
+    gm = gm.to(DEV).bfloat16()
+    # BN buffers must stay fp32 for running-stat kernels
+    for mod in gm.modules():
+        for bn, buf in list(mod._buffers.items()):
+            if buf is not None:
+                mod._buffers[bn] = buf.float()
+    got = gm(x.bfloat16().to(DEV)).float().cpu()
+    cos = torch.nn.functional.cosine_similarity(
+        got.flatten(), ref.flatten(), dim=0)
+    assert float(cos) > 0.95, (name, float(cos))

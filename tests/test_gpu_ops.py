@@ -228,7 +228,8 @@ def test_bn_fwd_bwd():
     torch.manual_seed(39)
     x = rnd(8, 7, 7, 64, scale=2.0)
     g, b = rnd(64, seed=40), rnd(64, seed=41)
-    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, False)
+    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, False,
+                               None, None, 0.9)
     ref = cpu_ref.batchnorm(x.float(), g.float(), b.float())
     check(y, ref, 0.05, "bn_fwd")
     dy = rnd(8, 7, 7, 64, seed=42)
@@ -267,7 +268,8 @@ def test_bn_bwd_fused_relu():
     torch.manual_seed(43)
     x = rnd(4, 5, 5, 32, scale=2.0)
     g, b = rnd(32, seed=44), rnd(32, seed=45)
-    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, True)
+    y, mean, rstd = ext.bn_fwd(x.to(DEV), g.to(DEV), b.to(DEV), 1e-5, True,
+                               None, None, 0.9)
     dy = rnd(4, 5, 5, 32, seed=46)
     dx, dgamma, dbeta = ext.bn_bwd(dy.to(DEV), x.to(DEV), g.to(DEV), mean,
                                    rstd, y, True)
@@ -278,3 +280,17 @@ def test_bn_bwd_fused_relu():
     check(dx, x32.grad, 0.06, "bnrelu_dx")
     check(dgamma, g32.grad, 0.06, "bnrelu_dgamma")
     check(dbeta, b32.grad, 0.06, "bnrelu_dbeta")
+
+
+def test_bn_running_stats_update():
+    torch.manual_seed(60)
+    x = (torch.randn(16, 3, 3, 32) * 2 + 1).bfloat16().to(DEV)
+    g = torch.ones(32).bfloat16().to(DEV)
+    b = torch.zeros(32).bfloat16().to(DEV)
+    rm = torch.zeros(32, device=DEV)
+    rv = torch.ones(32, device=DEV)
+    ext.bn_fwd(x, g, b, 1e-5, False, rm, rv, 0.5)
+    m_ref = x.float().mean(dim=(0, 1, 2))
+    v_ref = x.float().var(dim=(0, 1, 2), unbiased=False)
+    check(rm.bfloat16(), (0.5 * m_ref).cpu(), 0.03, "bn_rm")
+    check(rv.bfloat16(), (0.5 + 0.5 * v_ref).cpu(), 0.03, "bn_rv")
