@@ -62,7 +62,7 @@ template <> struct GldsOk<PlainGather> { static constexpr bool value = true; };
 
 // Implicit-GEMM forward A: virtual A[m=(n,oh,ow)][k=(kh,kw,ci)] = x NHWC.
 // Requires Cin % 8 == 0 (chunks stay inside one filter tap) — Cin<8 first
-// layers use ConvFwdSmallCGather instead. All index divisions are
+// layers are channel-padded to 8 by the launcher. All index divisions are
 // host-precomputed magic multiplies (MDiv).
 struct ConvFwdGather {
   const u16* x;
@@ -140,45 +140,6 @@ struct ConvDgradGather {
   }
 };
 template <> struct GldsOk<ConvDgradGather> { static constexpr bool value = true; };
-
-// Forward A for Cin % 8 != 0 first layers (conv1 Cin=3, ResNet stem):
-// per-element scalar gather straight from NHWC x — no im2col materialized.
-// k-space is Kp = ceil8(KH*KW*Cin); elements beyond Kfull are zero.
-struct ConvFwdSmallCGather {
-  const u16* x;
-  int M, H, W, Cin, stride, pad, KW_, Kfull;
-  MDiv dOHOW, dOW, dCin, dKW;
-  struct Row { const u16* nrow; int ihb, iwb; bool valid; };
-  DEV Row prep(int m) const {
-    u32 n = dOHOW.div(m), r = dOHOW.mod(m, n);
-    u32 oh = dOW.div(r), ow = dOW.mod(r, oh);
-    return {x + (i64)n * H * W * Cin, int(oh) * stride - pad,
-            int(ow) * stride - pad, m < M};
-  }
-  DEV s16x8 load8(const Row& rs, int k) const {
-    s16x8 v = zero8();
-    if (!rs.valid) return v;
-    u32 f = dCin.div(k);
-    int ci = int(dCin.mod(k, f));
-    u32 kh = dKW.div(f);
-    int kw = int(dKW.mod(f, kh));
-    int khi = int(kh);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int kk = k + j;
-      if (kk < Kfull) {
-        int ih = rs.ihb + khi, iw = rs.iwb + kw;
-        if ((unsigned)ih < (unsigned)H && (unsigned)iw < (unsigned)W)
-          v[j] = short(rs.nrow[((i64)ih * W + iw) * Cin + ci]);
-      }
-      if (++ci == Cin) {  // walk (kh,kw,ci) incrementally: no per-elem divs
-        ci = 0;
-        if (++kw == KW_) { kw = 0; ++khi; }
-      }
-    }
-    return v;
-  }
-};
 
 // Wgrad A (Cin % 8 == 0): virtual A[m=(kh,kw,ci)][k=(n,oh,ow-padded)] over
 // xTpad [Cin][N][H+2p][Wp] (halo-padded, zero-filled; +8 elem slack). The
