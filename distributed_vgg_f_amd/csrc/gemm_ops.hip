@@ -255,36 +255,25 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
   if constexpr (GLDS) {
     const u16* zp = zero_page(outB ? outB->options() : c32->options()
                                   .dtype(at::kBFloat16));
+    u16* op = outB ? bfp_mut(*outB) : nullptr;
+    float* cp = c32 ? c32->data_ptr<float>() : nullptr;
+    dim3 gridk(tiles, splitk);
+    dim3 grid1(tiles);
+    // BKT=64 when K allows: nt=1 for the K=64 1x1 convs, half the barriers
+    // for big-K convs (LDS 64 KB -> 2 blocks/CU, glds prefetch covers)
+    bool bk64 = (K % 64 == 0);
+#define GLDS_DISPATCH(EPIC, SPLITC, GRID, CPTR, BPTR)                            do {                                                                           if (bk64)                                                                      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 64>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else                                                                           hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);        } while (0)
     if (splitk > 1) {
-      dim3 grid(tiles, splitk);
-      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 0, true, NFRAG>), grid,
-                         dim3(THREADS), 0, s, ga, gb, nullptr,
-                         c32->data_ptr<float>(), nullptr, zp, M, N, K, 0, K);
+      GLDS_DISPATCH(0, true, gridk, cp, nullptr);
       return;
     }
-    dim3 grid(tiles);
     switch (epi) {
-      case 0:
-        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 0, false, NFRAG>), grid,
-                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
-                           nullptr, nullptr, zp, M, N, K, 0, K);
-        return;
-      case 1:
-        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 1, false, NFRAG>), grid,
-                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
-                           nullptr, bp, zp, M, N, K, 0, K);
-        return;
-      case 2:
-        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 2, false, NFRAG>), grid,
-                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
-                           nullptr, bp, zp, M, N, K, 0, K);
-        return;
-      default:
-        hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, 3, false, NFRAG>), grid,
-                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
-                           nullptr, nullptr, zp, M, N, K, 0, K);
-        return;
+      case 0: GLDS_DISPATCH(0, false, grid1, nullptr, nullptr); return;
+      case 1: GLDS_DISPATCH(1, false, grid1, nullptr, bp); return;
+      case 2: GLDS_DISPATCH(2, false, grid1, nullptr, bp); return;
+      default: GLDS_DISPATCH(3, false, grid1, nullptr, nullptr); return;
     }
+#undef GLDS_DISPATCH
   }
   if (splitk > 1) {
     dim3 grid(tiles, splitk);

@@ -428,14 +428,15 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
 // glds queue before every ds_read — cdna_hip_programming.md section 5
 // ".s-level traps" (a)); __syncthreads() itself emits the vmcnt(0) that
 // completes the in-flight tile.
-template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4>
+template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4,
+          int BKT = BK>
 __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
     GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
     const u16* __restrict__ bias, const u16* __restrict__ zpage, int M, int N,
     int K, int k_lo, int k_hi) {
   constexpr int BN_T = 32 * NFRAG;
-  constexpr int AE = BM * BK;            // A elems per buffer
-  constexpr int BE = BN_T * BK;
+  constexpr int AE = BM * BKT;           // A elems per buffer
+  constexpr int BE = BN_T * BKT;
   __shared__ u16 smem[2][AE + BE];
 
   const int tiles_n = (N + BN_T - 1) / BN_T;
@@ -445,8 +446,8 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
   const int m0 = tm * BM, n0 = tn * BN_T;
   if (SPLITK) {
     int nchunk = gridDim.y;
-    int kt = (K + BK - 1) / BK;
-    int span = (kt + nchunk - 1) / nchunk * BK;
+    int kt = (K + BKT - 1) / BKT;
+    int span = (kt + nchunk - 1) / nchunk * BKT;
     k_lo = blockIdx.y * span;
     k_hi = min(K, k_lo + span);
     if (k_lo >= k_hi) return;
@@ -465,8 +466,8 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
 #pragma unroll
   for (int g = 0; g < AGI; ++g) {
     int off = (wave * AGI + g) * 512 + lane * 8;
-    rsa[g] = ga.prep(m0 + (off >> 5));
-    ak8[g] = off & 31;
+    rsa[g] = ga.prep(m0 + off / BKT);
+    ak8[g] = off % BKT;
     alds[g] = &smem[0][(wave * AGI + g) * 512];
   }
   typename GB::Row rsb[BGI];
@@ -475,8 +476,8 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
 #pragma unroll
   for (int g = 0; g < BGI; ++g) {
     int off = (wave * BGI + g) * 512 + lane * 8;
-    rsb[g] = gb.prep(n0 + (off >> 5));
-    bk8[g] = off & 31;
+    rsb[g] = gb.prep(n0 + off / BKT);
+    bk8[g] = off % BKT;
     blds[g] = &smem[0][AE + (wave * BGI + g) * 512];
   }
   constexpr int BUF = AE + BE;
@@ -489,33 +490,36 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
 #pragma unroll
     for (int j = 0; j < NFRAG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int nt = (k_hi - k_lo + BK - 1) / BK;
+  const int nt = (k_hi - k_lo + BKT - 1) / BKT;
   STAGE_GLDS(0, k_lo);
   __syncthreads();          // vmcnt(0) inside: tile 0 landed
 
   int p = 0;
   for (int tile = 0; tile < nt; ++tile) {
-    if (tile + 1 < nt) STAGE_GLDS(p ^ 1, k_lo + (tile + 1) * BK);
+    if (tile + 1 < nt) STAGE_GLDS(p ^ 1, k_lo + (tile + 1) * BKT);
     const int kfrag = (lane >> 4) * 8;
-    bf16x8 af[4], bf[NFRAG];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int row = wm + i * 16 + (lane & 15);
-      af[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(
-          &smem[p][row * BK + kfrag]));
+    for (int ks = 0; ks < BKT / 32; ++ks) {
+      bf16x8 af[4], bf[NFRAG];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = wm + i * 16 + (lane & 15);
+        af[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            &smem[p][row * BKT + ks * 32 + kfrag]));
+      }
+#pragma unroll
+      for (int j = 0; j < NFRAG; ++j) {
+        int col = wn + j * 16 + (lane & 15);
+        bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            &smem[p][AE + col * BKT + ks * 32 + kfrag]));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < NFRAG; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
     }
-#pragma unroll
-    for (int j = 0; j < NFRAG; ++j) {
-      int col = wn + j * 16 + (lane & 15);
-      bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
-          &smem[p][AE + col * BK + kfrag]));
-    }
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < NFRAG; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf[j], acc[i][j], 0, 0, 0);
     __syncthreads();        // drains next tile's glds; frees buf p
     p ^= 1;
   }
