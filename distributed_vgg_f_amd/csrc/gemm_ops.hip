@@ -259,9 +259,10 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     float* cp = c32 ? c32->data_ptr<float>() : nullptr;
     dim3 gridk(tiles, splitk);
     dim3 grid1(tiles);
-    // BKT=64 when K allows: nt=1 for the K=64 1x1 convs, half the barriers
-    // for big-K convs (LDS 64 KB -> 2 blocks/CU, glds prefetch covers)
-    bool bk64 = (K % 64 == 0);
+    // A/B verdict: BK=64 (64 KB LDS, 2 blocks/CU) loses 2-7% on every
+    // model incl. the K=64 1x1 single-tile case — occupancy beats barrier
+    // count again. Template retained for future tuning.
+    bool bk64 = false;
 #define GLDS_DISPATCH(EPIC, SPLITC, GRID, CPTR, BPTR)                            do {                                                                           if (bk64)                                                                      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 64>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else                                                                           hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);        } while (0)
     if (splitk > 1) {
       GLDS_DISPATCH(0, true, gridk, cp, nullptr);
