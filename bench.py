@@ -28,6 +28,9 @@ def main():
     p.add_argument("--model", type=str, default="vggf")
     p.add_argument("--bucket-mb", type=int, default=25)
     p.add_argument("--overlap-comm", type=int, default=1)
+    p.add_argument("--use-graph", type=int, default=1,
+                   help="hipGraph-capture the step (single-GPU; auto-off "
+                        "when distributed, eager fallback on failure)")
     args = p.parse_args()
 
     from distributed_vgg_f_amd.config import Config
@@ -38,7 +41,8 @@ def main():
     cfg = Config(model=args.model, batch_size=args.batch_size,
                  precision="bf16" if use_gpu else "fp32",
                  log_every=0, bucket_mb=args.bucket_mb,
-                 overlap_comm=bool(args.overlap_comm))
+                 overlap_comm=bool(args.overlap_comm),
+                 use_graph=bool(args.use_graph))
     ctx = init_distributed(cfg.backend)
     world = ctx.world_size
     if world > 1 and args.gpus != world:

@@ -61,6 +61,8 @@ class Config:
     log_every: int = 10
     metrics_path: Optional[str] = None  # JSONL metrics
     phase_timers: bool = False          # hipEvent per-phase ms (fwd/bwd/opt)
+    use_graph: bool = False             # hipGraph-capture the train step
+    #                                     (single-GPU; falls back to eager)
 
     # debug / fault injection
     fault_inject_step: int = -1

@@ -69,6 +69,53 @@ class Trainer:
             if self.ctx.is_chief and self.step_i:
                 self.metrics.log(event="resumed", step=self.step_i)
 
+        self._graph = None
+        if cfg.use_graph:
+            self._try_capture_graph()
+
+    # ------------------------------------------------------------------ #
+    def _graph_body(self):
+        """The captured step: H2D + fwd + bwd + fused SGD + grad zero."""
+        self._gimg.copy_(self._gimg_h, non_blocking=True)
+        self._glab.copy_(self._glab_h, non_blocking=True)
+        logits = self.model(self._gimg)
+        loss = ops.softmax_xent(logits, self._glab).sum()
+        loss.backward()
+        self.arena.step(grad_scale=1.0 / self.global_batch)
+        self.arena.zero_grad()
+        return loss
+
+    def _try_capture_graph(self) -> bool:
+        """Capture the whole train step into a hipGraph (launch-bound inner
+        loop -> one replay per step). Single-GPU only this round: RCCL
+        collectives stay out of capture. Falls back to eager on any error."""
+        if self.device.type != "cuda" or self.ctx.world_size > 1:
+            return False
+        try:
+            self._gimg_h = self.pipeline.images[0]     # pinned host
+            self._glab_h = self.pipeline.labels[0]
+            self._gimg = torch.empty(self._gimg_h.shape,
+                                     dtype=self._gimg_h.dtype,
+                                     device=self.device)
+            self._glab = torch.empty(self._glab_h.shape, dtype=torch.long,
+                                     device=self.device)
+            for _ in range(3):                         # allocator warmup
+                self._graph_body()
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._gloss = self._graph_body()
+            self._graph = g
+            if self.ctx.is_chief:
+                self.metrics.log(event="hipgraph_captured")
+            return True
+        except Exception as e:  # pragma: no cover - hardware dependent
+            import sys
+            print(f"[trainer] hipGraph capture failed, eager fallback: {e}",
+                  file=sys.stderr)
+            self._graph = None
+            return False
+
     # ------------------------------------------------------------------ #
     def train_step(self) -> torch.Tensor:
         """One synchronous-SGD step over the global batch. Returns the
@@ -78,6 +125,10 @@ class Trainer:
                 cfg.fault_inject_rank == self.ctx.rank):
             raise RuntimeError(
                 f"fault-inject: rank {self.ctx.rank} dies at step {self.step_i}")
+        if self._graph is not None:
+            self._graph.replay()
+            self.step_i += 1
+            return self._gloss
         images, labels = self.pipeline.next()
         self.reducer.prepare_step()
         tm = self.timers
