@@ -74,38 +74,60 @@ class Trainer:
             self._try_capture_graph()
 
     # ------------------------------------------------------------------ #
-    def _graph_body(self):
-        """The captured step: H2D + fwd + bwd + fused SGD + grad zero."""
-        self._gimg.copy_(self._gimg_h, non_blocking=True)
-        self._glab.copy_(self._glab_h, non_blocking=True)
-        logits = self.model(self._gimg)
-        loss = ops.softmax_xent(logits, self._glab).sum()
+    def _graph_body(self, i: int):
+        """Captured step i: compute on buffer i while the OTHER buffer's
+        H2D runs on a forked copy stream (cross-step prefetch: graphs are
+        replayed alternately, so replay k's copy feeds replay k+1)."""
+        main = torch.cuda.current_stream()
+        ev_fork = torch.cuda.Event()
+        ev_join = torch.cuda.Event()
+        ev_fork.record(main)
+        self._gcs.wait_event(ev_fork)
+        with torch.cuda.stream(self._gcs):
+            self._gimg[1 - i].copy_(self._gimg_h[1 - i], non_blocking=True)
+            self._glab[1 - i].copy_(self._glab_h[1 - i], non_blocking=True)
+            ev_join.record(self._gcs)
+        logits = self.model(self._gimg[i])
+        loss = ops.softmax_xent(logits, self._glab[i]).sum()
         loss.backward()
         self.arena.step(grad_scale=1.0 / self.global_batch)
         self.arena.zero_grad()
+        main.wait_event(ev_join)
         return loss
 
     def _try_capture_graph(self) -> bool:
-        """Capture the whole train step into a hipGraph (launch-bound inner
-        loop -> one replay per step). Single-GPU only this round: RCCL
+        """Capture the train step into TWO ping-pong hipGraphs (compute on
+        buffer i, prefetch buffer 1-i on a copy stream) so H2D overlaps
+        compute across replays. Single-GPU only this round: RCCL
         collectives stay out of capture. Falls back to eager on any error."""
         if self.device.type != "cuda" or self.ctx.world_size > 1:
             return False
         try:
-            self._gimg_h = self.pipeline.images[0]     # pinned host
-            self._glab_h = self.pipeline.labels[0]
-            self._gimg = torch.empty(self._gimg_h.shape,
-                                     dtype=self._gimg_h.dtype,
-                                     device=self.device)
-            self._glab = torch.empty(self._glab_h.shape, dtype=torch.long,
-                                     device=self.device)
-            for _ in range(3):                         # allocator warmup
-                self._graph_body()
+            self._gimg_h = [self.pipeline.images[0], self.pipeline.images[1]]
+            self._glab_h = [self.pipeline.labels[0], self.pipeline.labels[1]]
+            self._gimg = [torch.empty(h.shape, dtype=h.dtype,
+                                      device=self.device)
+                          for h in self._gimg_h]
+            self._glab = [torch.empty(h.shape, dtype=torch.long,
+                                      device=self.device)
+                          for h in self._glab_h]
+            self._gcs = torch.cuda.Stream(self.device)
+            for i in (0, 1):
+                self._gimg[i].copy_(self._gimg_h[i], non_blocking=True)
+                self._glab[i].copy_(self._glab_h[i], non_blocking=True)
+            for i in (0, 1, 0):                      # allocator warmup
+                self._graph_body(i)
             torch.cuda.synchronize(self.device)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._gloss = self._graph_body()
-            self._graph = g
+            self._graphs, self._glosses = [], []
+            pool = torch.cuda.graph_pool_handle()
+            for i in (0, 1):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g, pool=pool):
+                    loss = self._graph_body(i)
+                self._graphs.append(g)
+                self._glosses.append(loss)
+            self._graph = True
+            self._gi = 0
             if self.ctx.is_chief:
                 self.metrics.log(event="hipgraph_captured")
             return True
@@ -126,9 +148,11 @@ class Trainer:
             raise RuntimeError(
                 f"fault-inject: rank {self.ctx.rank} dies at step {self.step_i}")
         if self._graph is not None:
-            self._graph.replay()
+            i = self._gi
+            self._gi ^= 1
+            self._graphs[i].replay()
             self.step_i += 1
-            return self._gloss
+            return self._glosses[i]
         images, labels = self.pipeline.next()
         self.reducer.prepare_step()
         tm = self.timers
