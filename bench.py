@@ -28,9 +28,10 @@ def main():
     p.add_argument("--model", type=str, default="vggf")
     p.add_argument("--bucket-mb", type=int, default=25)
     p.add_argument("--overlap-comm", type=int, default=1)
-    p.add_argument("--use-graph", type=int, default=1,
-                   help="hipGraph-capture the step (single-GPU; auto-off "
-                        "when distributed, eager fallback on failure)")
+    p.add_argument("--use-graph", type=int, default=0,
+                   help="hipGraph-capture the step (single-GPU only; A/B: "
+                        "wins ~5%% at bs<=64, loses ~10%% at bs=256 -> "
+                        "default off; see DESIGN.md)")
     args = p.parse_args()
 
     from distributed_vgg_f_amd.config import Config
