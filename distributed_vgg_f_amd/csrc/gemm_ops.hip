@@ -34,6 +34,20 @@ static void check_bf16(const at::Tensor& t, const char* name) {
 
 static inline int ceil8(int v) { return (v + 7) & ~7; }
 
+// runtime tuning knobs (benchmark sweeps; defaults are the shipped values)
+static int env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
+static int nfrag_threshold() {           // N <= thresh -> 64-wide tile
+  static int t = env_int("DVGG_NFRAG_THRESH", 64);
+  return t;
+}
+static int sk_scale() {                  // percent scale on split-K targets
+  static int t = env_int("DVGG_SK_SCALE", 100);
+  return t;
+}
+
 // ------------------------------------------------------------ transpose2d
 // [R][C] -> [C][Rp] (Rp >= R, zero-padded cols), bf16, LDS-tiled.
 __global__ void transpose2d_kernel(const u16* __restrict__ in,
@@ -312,7 +326,7 @@ template <class GA, class GB>
 static void launch_gemm(const GA& ga, const GB& gb, at::Tensor* outB,
                         at::Tensor* c32, const at::Tensor* bias, int epi,
                         int M, int N, int K, int splitk) {
-  if (N <= 64)
+  if (N <= nfrag_threshold())
     launch_gemm_t<GA, GB, 2>(ga, gb, outB, c32, bias, epi, M, N, K, splitk);
   else
     launch_gemm_t<GA, GB, 4>(ga, gb, outB, c32, bias, epi, M, N, K, splitk);
@@ -344,13 +358,14 @@ static void launch_gemm_tn_t(const PlainGather& ga, const PlainGather& gb,
 static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
                            at::Tensor* outB, at::Tensor* c32, int M, int N,
                            int K, int splitk) {
-  if (N <= 64)
+  if (N <= nfrag_threshold())
     launch_gemm_tn_t<2>(ga, gb, outB, c32, M, N, K, splitk);
   else
     launch_gemm_tn_t<4>(ga, gb, outB, c32, M, N, K, splitk);
 }
 
 static int pick_splitk(int tiles, int K, int target_blocks = 512) {
+  target_blocks = target_blocks * sk_scale() / 100;
   if (tiles >= target_blocks) return 1;
   int kt = std::max(1, K / BK);
   int want = std::min(kt, std::max(1, target_blocks / std::max(1, tiles)));
