@@ -294,3 +294,19 @@ def test_bn_running_stats_update():
     v_ref = x.float().var(dim=(0, 1, 2), unbiased=False)
     check(rm.bfloat16(), (0.5 * m_ref).cpu(), 0.03, "bn_rm")
     check(rv.bfloat16(), (0.5 + 0.5 * v_ref).cpu(), 0.03, "bn_rv")
+
+
+def test_wgrad_splitk_stress_repeat():
+    """split-K atomic accumulation under varying block
+    execution order — repeated runs must all stay within tolerance of the
+    reference (catches order-dependent bugs in the k-slicing)."""
+    torch.manual_seed(70)
+    N, H, W, Ci, Co, k, s, p = 2, 13, 13, 256, 256, 3, 1, 1
+    x = rnd(N, H, W, Ci, scale=0.5)
+    dy = rnd(N, H, W, Co, scale=0.5)
+    w32 = torch.zeros(k, k, Ci, Co, requires_grad=True)
+    cpu_ref.conv2d(x.float(), w32, None, s, p).backward(dy.float())
+    xg, dyg = x.to(DEV), dy.to(DEV)
+    for rep in range(5):
+        dw = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
+        check(dw, w32.grad, 0.06, f"wgrad_stress_rep{rep}")
