@@ -106,6 +106,36 @@ __global__ void nhwc2cnhw_kernel(const u16* __restrict__ in,
   }
 }
 
+// dyTS builder for the glds wgrad: KW pre-shifted transposed copies of the
+// (masked) dy. Copy kw holds dy[n][oh][ow][co] at
+// dyTS[kw][co][n][oh][ow + kw - pad + 8]; everything else stays zero.
+__global__ void dyts_kernel(const u16* __restrict__ dy, u16* __restrict__ out,
+                            int N, int OH, int OW, int C, int OWq, int KW,
+                            int pad, i64 copy_stride) {
+  __shared__ u16 tile[32][33];
+  int nh = blockIdx.x;
+  int w0 = blockIdx.y * 32, c0 = blockIdx.z * 32;
+  int n = nh / OH, oh = nh % OH;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int iw = w0 + ty + i * 8, c = c0 + tx;
+    tile[ty + i * 8][tx] = (iw < OW && c < C)
+        ? dy[(((i64)n * OH + oh) * OW + iw) * C + c] : u16(0);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int c = c0 + ty + i * 8, iw = w0 + tx;
+    if (c < C && iw < OW) {
+      u16 v = tile[tx][ty + i * 8];
+      i64 base = (((i64)c * N + n) * OH + oh) * OWq + iw + 8 - pad;
+      for (int kw = 0; kw < KW; ++kw)
+        out[kw * copy_stride + base + kw] = v;
+    }
+  }
+}
+
 // Cached, zeroed, slack-padded buffers for the wgrad transposes: the
 // interior is fully rewritten every call and the zero halo is never
 // written, so the (expensive at bs=256) zero-fill happens once per shape.
@@ -289,7 +319,7 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
       default: GLDS_DISPATCH(3, false, grid1, nullptr, nullptr); return;
     }
 #undef GLDS_DISPATCH
-  }
+  } else {   // register-staged path (never instantiated for glds-only gathers)
   if (splitk > 1) {
     dim3 grid(tiles, splitk);
     hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, NFRAG>), grid,
@@ -319,6 +349,7 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
                            nullptr, nullptr, M, N, K, 0, K);
         break;
     }
+  }
   }
 }
 
@@ -467,6 +498,45 @@ static at::Tensor pad_channels8(const at::Tensor& x) {
   return out;
 }
 
+// glds wgrad (stride==1, Cin %% 128 == 0): substituted-reduction implicit
+// GEMM with both operands staged by global_load_lds (see gemm_tile.h).
+static bool wgrad_glds_ok(int Cin, int stride, int KW, int pad) {
+  return stride == 1 && Cin % 128 == 0 && KW <= 8 && pad <= 8 &&
+         env_int("DVGG_WGRAD_GLDS", 1);
+}
+
+static void wgrad_glds(const at::Tensor& x, const at::Tensor& dym, int KH,
+                       int KW, int pad, at::Tensor& dw2d, int Cout) {
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int OH = dym.size(1), OW = dym.size(2);
+  int Kfull = KH * KW * Cin;
+  int Wk = ceil8(W);
+  int Kred = N * OH * Wk;
+  int Hp = H + 2 * pad;
+  int Wg = ceil8(W) + 16;
+  auto xT = nhwc2cnhw_pad(x, Hp, Wg, pad, 8, 'g');   // cols [8, 8+W)
+  int OWq = ceil8(Wk + 16) + 8;
+  i64 cstride = (i64)Cout * N * OH * OWq;
+  char key[96];
+  snprintf(key, sizeof key, "s%d_%d_%d_%d_%d_%d", x.get_device(), KW, Cout,
+           N, OH, OWq);
+  auto dyTS = cached_zero_buffer(key, (i64)KW * cstride, x.options())
+      .narrow(0, 0, (i64)KW * cstride);
+  dim3 tgrid(N * OH, ceil_div(OW, 32), ceil_div(Cout, 32));
+  hipLaunchKernelGGL(dyts_kernel, tgrid, dim3(256), 0, cur_stream(),
+                     bfp(dym), bfp_mut(dyTS), N, OH, OW, Cout, OWq, KW, pad,
+                     cstride);
+  MDiv dOHWk{u32(OH * Wk)}, dWk{u32(Wk)};
+  WgradXgGather ga{bfp(xT), Kfull, Cin, N, Hp, Wg, OH, Wk, Kred,
+                   MDiv(u32(Cin)), MDiv(u32(KW)), dOHWk, dWk};
+  WgradDygGather gb{bfp(dyTS), Cout, N, OH, OWq, Kred, cstride, dOHWk, dWk};
+  int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
+  int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+}
+
 // ------------------------------------------------------------------ conv
 static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
                            int& OH, int& OW) {
@@ -562,6 +632,11 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
 
+  if (wgrad_glds_ok(Cin, stride, KW, pad)) {
+    auto dyc = dy.contiguous();
+    wgrad_glds(x, dyc, KH, KW, pad, dw2d, Cout);
+    return dw;
+  }
   // Padding Cin<8 to 8 was tried and measured 3.4x SLOWER for wgrad
   // (zero-channel waste x scalar stride-4 staging); the direct NHWC
   // gather below stays for Cin < 8.
@@ -687,9 +762,23 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     }
   }
 
-  // ---- wgrad (reuses dyT from the preamble)
+  // ---- wgrad
   auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
+  if (wgrad_glds_ok(Cin, stride, KW, pad)) {
+    // dym is already ReLU-masked when relu (preamble wrote it); dyT from
+    // the preamble is unused on this path
+    auto dymc = dym.contiguous();
+    wgrad_glds(x, dymc, KH, KW, pad, dw2d, Cout);
+    at::Tensor db;
+    if (need_db) {
+      db = at::empty({Cout}, dy.options());
+      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                         dim3(256), 0, s, db32.data_ptr<float>(),
+                         bfp_mut(db), Cout);
+    }
+    return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
+  }
   int Kred = N * OH * OWp;
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);

@@ -239,6 +239,64 @@ struct WgradXDirectGather {
   }
 };
 
+// glds-capable wgrad gathers (stride==1, Cin % 128 == 0: one filter tap
+// per 128-row M-tile). The reduction is substituted to ow' = ow + kw - pad
+// so the A side reads x at ALIGNED columns (x stored at column iw+8 in
+// xTg [Cin][N][H+2p][Wg]); the per-tap column shift is baked into KW
+// pre-shifted zero-padded copies of the transposed dy
+// (dyTS [KW][Cout][N][OH][OWq]: copy kw holds dy[ow] at column
+// ow + kw - pad + 8). The kernel passes the block-uniform tap to B.
+struct WgradXgGather {
+  const u16* xT;           // [Cin][N][Hp][Wg], column offset +8, zero halo
+  int Mrows, Cin, N, Hp, Wg, OH, Wk, Klim;   // Wk = ceil8(W): k-space width
+  MDiv dCin, dKW, dOHWk, dWk;
+  struct Row { const u16* cib; int kh; bool valid; };
+  DEV Row prep(int m) const {
+    u32 f = dCin.div(m), ci = dCin.mod(m, f);
+    u32 kh = dKW.div(f);
+    return {xT + (i64)ci * N * Hp * Wg, int(kh), m < Mrows};
+  }
+  DEV int block_tap(int m0) const {   // kw of this M-tile (tile-uniform)
+    u32 f = dCin.div(m0);
+    u32 kh = dKW.div(f);
+    return int(dKW.mod(f, kh));
+  }
+  DEV const u16* ptr8(const Row& rs, int k, const u16* zpage) const {
+    if (!rs.valid || k >= Klim) return zpage;
+    u32 n = dOHWk.div(k), r = dOHWk.mod(k, n);
+    u32 oh = dWk.div(r), owp = dWk.mod(r, oh);
+    return rs.cib + ((i64)n * Hp + int(oh) + rs.kh) * Wg + owp + 8;
+  }
+};
+template <> struct GldsOk<WgradXgGather> { static constexpr bool value = true; };
+
+struct WgradDygGather {
+  const u16* dyTS;         // [KW][Cout][N][OH][OWq]
+  int Cout, N, OH, OWq, Klim;
+  i64 copy_stride;         // Cout*N*OH*OWq elems
+  MDiv dOHWk, dWk;         // SAME k-space decomposition as the A side
+  struct Row { i64 off; bool valid; };
+  DEV Row prep(int m) const {
+    return {(i64)m * N * OH * OWq, m < Cout};
+  }
+  DEV const u16* ptr8_tap(const Row& rs, int k, int tap,
+                          const u16* zpage) const {
+    if (!rs.valid || k >= Klim) return zpage;
+    u32 n = dOHWk.div(k), r = dOHWk.mod(k, n);
+    u32 oh = dWk.div(r), owp = dWk.mod(r, oh);
+    return dyTS + (i64)tap * copy_stride + rs.off +
+           ((i64)n * OH + int(oh)) * OWq + owp + 8;
+  }
+  DEV const u16* ptr8(const Row& rs, int k, const u16* zpage) const {
+    return ptr8_tap(rs, k, 0, zpage);
+  }
+};
+template <> struct GldsOk<WgradDygGather> { static constexpr bool value = true; };
+
+template <class G> struct HasTap { static constexpr bool value = false; };
+template <> struct HasTap<WgradXgGather> { static constexpr bool value = true; };
+template <> struct HasTap<WgradDygGather> { static constexpr bool value = true; };
+
 // Wgrad B: virtual Bt[n'=co][k=(n,oh,ow-padded)] over dyTpad
 // [Cout][N][OH][OWp] (zero-filled pad, 16B rows) — clean vector loads.
 struct WgradDyGather {
@@ -418,6 +476,16 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   }
 }
 
+// tap-aware B-source helper for the glds kernel (wgrad: block-uniform tap)
+template <class GB>
+DEV const u16* gb_src(const GB& gb, const typename GB::Row& r, int k,
+                      int tap, const u16* zpage) {
+  if constexpr (HasTap<GB>::value)
+    return gb.ptr8_tap(r, k, tap, zpage);
+  else
+    return gb.ptr8(r, k, zpage);
+}
+
 // ------------------------------------------------- glds-staged variant
 // For operands whose chunks are single 16B loads (ConvFwd/ConvDgrad/Plain):
 // stage via __builtin_amdgcn_global_load_lds with PER-LANE source addresses
@@ -481,8 +549,11 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
     blds[g] = &smem[0][AE + (wave * BGI + g) * 512];
   }
   constexpr int BUF = AE + BE;
+  int tap = 0;
+  if constexpr (HasTap<GA>::value) tap = ga.block_tap(m0);
+  (void)tap;
 
-#define STAGE_GLDS(pbuf, kb)                                                   do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < AGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + ak8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (alds[g] + (pbuf) * BUF), 16, 0, 0);                               _Pragma("unroll")                                                            for (int g = 0; g < BGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb.ptr8(rsb[g], (kb) + bk8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (blds[g] + (pbuf) * BUF), 16, 0, 0);                             } while (0)
+#define STAGE_GLDS(pbuf, kb)                                                   do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < AGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + ak8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (alds[g] + (pbuf) * BUF), 16, 0, 0);                               _Pragma("unroll")                                                            for (int g = 0; g < BGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb[g], (kb) + bk8[g], tap, zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (blds[g] + (pbuf) * BUF), 16, 0, 0);                             } while (0)
 
   f32x4 acc[4][NFRAG];
 #pragma unroll
