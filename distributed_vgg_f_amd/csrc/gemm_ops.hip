@@ -502,7 +502,7 @@ static at::Tensor pad_channels8(const at::Tensor& x) {
 // GEMM with both operands staged by global_load_lds (see gemm_tile.h).
 static bool wgrad_glds_ok(int Cin, int stride, int KW, int pad) {
   return stride == 1 && Cin % 128 == 0 && KW <= 8 && pad <= 8 &&
-         env_int("DVGG_WGRAD_GLDS", 1);
+         env_int("DVGG_WGRAD_GLDS", 0);
 }
 
 static void wgrad_glds(const at::Tensor& x, const at::Tensor& dym, int KH,
