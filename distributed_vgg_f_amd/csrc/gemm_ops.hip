@@ -395,6 +395,24 @@ static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
     launch_gemm_tn_t<4>(ga, gb, outB, c32, M, N, K, splitk);
 }
 
+// wgrad-dedicated launcher: BM=256/512-thread tile when Cout >= 128
+// (halves the dy B-panel re-reads — wgrad's binding constraint).
+template <class GA, class GB>
+static void launch_gemm_wgrad(const GA& ga, const GB& gb, at::Tensor* c32,
+                              int M, int N, int K, int splitk) {
+  auto s = cur_stream();
+  if (N >= 128 && M >= 256 && env_int("DVGG_WGRAD_BM256", 1)) {
+    int tiles = ceil_div(M, 256) * ceil_div(N, 128);
+    dim3 grid(tiles, splitk);
+    hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, 4, false, false, 256,
+                                    512>), grid, dim3(512), 0, s, ga, gb,
+                       nullptr, c32->data_ptr<float>(), nullptr, M, N, K, 0,
+                       K);
+    return;
+  }
+  launch_gemm(ga, gb, nullptr, c32, nullptr, 0, M, N, K, splitk);
+}
+
 static int pick_splitk(int tiles, int K, int target_blocks = 512) {
   target_blocks = target_blocks * sk_scale() / 100;
   if (tiles >= target_blocks) return 1;
@@ -659,7 +677,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
                     MDiv(u32(OWp))};
-    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+    launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
   return dw;
@@ -796,7 +814,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
                     MDiv(u32(OWp))};
-    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+    launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
 

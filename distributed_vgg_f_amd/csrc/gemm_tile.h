@@ -321,20 +321,24 @@ struct WgradDyGather {
 // wgrad, where both x and dy have the reduction as their OUTER dim); the
 // transpose happens during LDS staging (scalar ds_writes) instead of a
 // separate global transpose pass, and the reduction dim needs no padding.
+// BM_T/THREADS_T: 128/256 (4 waves, 2x2) or 256/512 (8 waves, 4x2 — same
+// per-wave 64-row tile; halves the B-panel re-read count for wgrad, whose
+// dy operand is re-read once per M-tile).
 template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4,
-          bool TNA = false, bool TNB = false>
-__global__ __launch_bounds__(THREADS) void gemm_kernel(
+          bool TNA = false, bool TNB = false, int BM_T = BM,
+          int THREADS_T = THREADS>
+__global__ __launch_bounds__(THREADS_T) void gemm_kernel(
     GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
     const u16* __restrict__ bias, int M, int N, int K, int k_lo, int k_hi) {
   constexpr int BN_T = 32 * NFRAG;
-  __shared__ u16 As[2][BM * LDS_STRIDE];
+  __shared__ u16 As[2][BM_T * LDS_STRIDE];
   __shared__ u16 Bs[2][BN_T * LDS_STRIDE];
 
   const int tiles_n = (N + BN_T - 1) / BN_T;
-  const int tiles_m = (M + BM - 1) / BM;
+  const int tiles_m = (M + BM_T - 1) / BM_T;
   u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
   const int tm = wg / tiles_n, tn = wg % tiles_n;
-  const int m0 = tm * BM, n0 = tn * BN_T;
+  const int m0 = tm * BM_T, n0 = tn * BN_T;
   if (SPLITK) {
     int nchunk = gridDim.y;
     int kt = (K + BK - 1) / BK;
@@ -351,14 +355,14 @@ __global__ __launch_bounds__(THREADS) void gemm_kernel(
   // staging assignment: chunk = 8 bf16; A: BM*BK/2048 chunks per thread,
   // B: BN_T*BK/2048; threads tile rows first (k-contiguous operands) or
   // k-rows first (TN operands: chunk = 8 columns of one reduction row).
-  constexpr int CPT_A = BM * BK / 8 / THREADS;      // 2 at BK=32
-  constexpr int CPT_B = BN_T * BK / 8 / THREADS;    // 2 / 1
+  constexpr int CPT_A = BM_T * BK / 8 / THREADS_T;  // 2 at BK=32
+  constexpr int CPT_B = BN_T * BK / 8 / THREADS_T;  // 2 / 1
   constexpr int TPR_A = (BK / 8) / CPT_A;           // threads per A row
   constexpr int TPR_B = (BK / 8) / CPT_B;
   const int arow = t / TPR_A, ak8 = (t % TPR_A) * (CPT_A * 8);
   const int brow = t / TPR_B, bk8 = (t % TPR_B) * (CPT_B * 8);
   // TN assignments: chunk id -> (local k, col8)
-  const int atk = (t * CPT_A) / (BM / 8), atc = ((t * CPT_A) % (BM / 8)) * 8;
+  const int atk = (t * CPT_A) / (BM_T / 8), atc = ((t * CPT_A) % (BM_T / 8)) * 8;
   const int btk = (t * CPT_B) / (BN_T / 8), btc = ((t * CPT_B) % (BN_T / 8)) * 8;
 
   const auto rsa = ga.prep(TNA ? 0 : m0 + arow);
