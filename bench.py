@@ -12,7 +12,6 @@ Multi (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
 
 import argparse
 import json
-import os
 import sys
 import time
 

@@ -15,8 +15,8 @@ from __future__ import annotations
 import argparse
 import dataclasses
 import json
-from dataclasses import dataclass, field
-from typing import List, Optional
+from dataclasses import dataclass
+from typing import Optional
 
 
 @dataclass

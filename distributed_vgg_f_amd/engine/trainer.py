@@ -9,7 +9,6 @@ backward, one fused SGD over the arenas.
 
 from __future__ import annotations
 
-import os
 import time
 from typing import Optional
 

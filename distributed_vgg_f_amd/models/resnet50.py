@@ -8,7 +8,6 @@ the canonical graph has conv->BN->ReLU.
 
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 
 from .. import ops

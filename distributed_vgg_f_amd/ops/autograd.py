@@ -17,8 +17,6 @@ Fusion contract with the kernels:
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from . import native

@@ -13,7 +13,6 @@ tensors never touch this module.
 from __future__ import annotations
 
 import importlib
-import os
 
 _ext = None
 _tried = False

@@ -29,7 +29,7 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 
-from ..engine.arena import ArenaOptimizer, Bucket
+from ..engine.arena import ArenaOptimizer
 
 
 @dataclass

@@ -2,8 +2,6 @@
 """setup.py — `python setup.py build_ext --inplace` builds the in-tree HIP
 extension via build_hip.py (raw hipcc for gfx950; no hipify)."""
 
-import sys
-
 from setuptools import setup
 from setuptools.command.build_ext import build_ext as _build_ext
 

@@ -2,7 +2,6 @@
 config[0]), loss decreases under the arena optimizer, param count matches
 SURVEY.md section 2.3 (60,834,536)."""
 
-import pytest
 import torch
 
 from distributed_vgg_f_amd.config import Config

@@ -2,8 +2,6 @@
 (T-unit/T-grad of SURVEY.md section 4.2). These define the contract the HIP
 kernels are tested against in the gpu-marked tests."""
 
-import numpy as np
-import pytest
 import torch
 
 from distributed_vgg_f_amd.ops import cpu_ref
