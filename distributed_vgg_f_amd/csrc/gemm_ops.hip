@@ -8,6 +8,7 @@
 
 #include "common.h"
 #include "gemm_tile.h"
+#include "gemm8.h"
 
 namespace dvgg {
 
@@ -307,6 +308,41 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     // model incl. the K=64 1x1 single-tile case — occupancy beats barrier
     // count again. Template retained for future tuning.
     bool bk64 = false;
+    // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B)
+    if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 &&
+        env_int("DVGG_GEMM8", 1)) {
+      int tiles8 = ceil_div(M, G8_BM) * ceil_div(N, G8_BN);
+      if (splitk > 1) {
+        dim3 g8(tiles8, splitk);
+        hipLaunchKernelGGL((gemm8_kernel<GA, GB, 0, true>), g8,
+                           dim3(G8_THREADS), 0, s, ga, gb, nullptr, cp,
+                           nullptr, zp, M, N, K, 0, K);
+        return;
+      }
+      dim3 g8(tiles8);
+      switch (epi) {
+        case 0:
+          hipLaunchKernelGGL((gemm8_kernel<GA, GB, 0, false>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr,
+                             nullptr, zp, M, N, K, 0, K);
+          return;
+        case 1:
+          hipLaunchKernelGGL((gemm8_kernel<GA, GB, 1, false>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr, bp,
+                             zp, M, N, K, 0, K);
+          return;
+        case 2:
+          hipLaunchKernelGGL((gemm8_kernel<GA, GB, 2, false>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr, bp,
+                             zp, M, N, K, 0, K);
+          return;
+        default:
+          hipLaunchKernelGGL((gemm8_kernel<GA, GB, 3, false>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr,
+                             nullptr, zp, M, N, K, 0, K);
+          return;
+      }
+    }
 #define GLDS_DISPATCH(EPIC, SPLITC, GRID, CPTR, BPTR)                            do {                                                                           if (bk64)                                                                      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 64>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else                                                                           hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);        } while (0)
     if (splitk > 1) {
       GLDS_DISPATCH(0, true, gridk, cp, nullptr);
