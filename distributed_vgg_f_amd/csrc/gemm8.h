@@ -7,11 +7,10 @@
 // (64 MFMA per K-tile per wave), fp32 accumulation (128 acc VGPRs/lane).
 // LDS: 2 buffers x (A 32 KB + B 32 KB) = 128 KB, one workgroup per CU.
 //
-// The LDS image uses the st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5):
-// 128-B rows read by ds_read_b128 would otherwise be an 8-way bank
-// conflict (Guideline 4). global_load_lds writes lane-linear, so the
-// swizzle is applied to the per-lane SOURCE element index and repeated on
-// the read side (section 5.4 rule 21; the map is an involution).
+// v2: counted-vmcnt k-half pipeline — each K-tile staged as two 32-KB
+// k-halves issued two phases before first read; raw s_barrier + counted
+// s_waitcnt vmcnt(4) keep prefetches in flight across barriers. LDS rows
+// are 64 B ([row][32] per half): ds_read_b128 residual conflict <= 4-way.
 #pragma once
 
 #include "common.h"
@@ -48,47 +47,36 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
   const int lane = t & 63, wave = t >> 6;
   const int wm = (wave >> 2) * 128, wn = (wave & 3) * 64;
 
-  // glds slots: operand bytes = 32 KB = 4 instrs/wave x 8 waves x 1 KB.
-  // Linear dest byte off = (wave*4+g)*1024 + lane*16; the SOURCE element
-  // comes from the swizzled position so the LDS image is st_16x32.
+  // K-half pipelined staging (v2): each K-tile's operands are staged as
+  // two k-halves (A 256x32 + B 256x32 = 32 KB, 4 glds per wave each).
+  // S(T,h) is issued during phase (T-1,h), i.e. TWO phases before its
+  // first read, and phases wait with a COUNTED s_waitcnt vmcnt(4) + raw
+  // s_barrier — prefetched halves stay in flight across barriers
+  // (cdna_hip_programming.md section 5 "Pipelining across barriers";
+  // __syncthreads would drain the glds queue to vmcnt(0) every phase).
+  constexpr int HE = G8_BM * 32;              // elems per operand k-half
   typename GA::Row rsa[4];
   typename GB::Row rsb[4];
   int ak8[4], bk8[4];
-  u16* alds[4];
-  u16* blds[4];
+  int aoff[4], boff[4];
 #pragma unroll
   for (int g = 0; g < 4; ++g) {
-    int off = (wave * 4 + g) * 1024 + lane * 16;   // dest byte (linear)
-    int soff = sw16x32(off);                       // source element position
-    int row = soff / 128, k8 = (soff % 128) / 2;
+    int off = (wave * 4 + g) * 1024 + lane * 16;   // byte in 16 KB half
+    int row = off / 64, k8 = (off % 64) / 2;
     rsa[g] = ga.prep(m0 + row);
     ak8[g] = k8;
-    alds[g] = &smem[0][0] + off / 2;
+    aoff[g] = off / 2;                             // elem offset in A half
     rsb[g] = gb.prep(n0 + row);
     bk8[g] = k8;
-    blds[g] = &smem[0][0] + AE + off / 2;
+    boff[g] = off / 2;
   }
   int tap = 0;
   if constexpr (HasTap<GA>::value) tap = ga.block_tap(m0);
   (void)tap;
 
-#define G8_STAGE(pbuf, kb)                                                   \
-  do {                                                                       \
-    _Pragma("unroll")                                                        \
-    for (int g = 0; g < 4; ++g)                                              \
-      __builtin_amdgcn_global_load_lds(                                      \
-          (const __attribute__((address_space(1))) unsigned int*)            \
-              ga.ptr8(rsa[g], (kb) + ak8[g], zpage),                         \
-          (__attribute__((address_space(3))) unsigned int*)                  \
-              (alds[g] + (i64)(pbuf) * 2 * AE), 16, 0, 0);                   \
-    _Pragma("unroll")                                                        \
-    for (int g = 0; g < 4; ++g)                                              \
-      __builtin_amdgcn_global_load_lds(                                      \
-          (const __attribute__((address_space(1))) unsigned int*)            \
-              gb_src(gb, rsb[g], (kb) + bk8[g], tap, zpage),                 \
-          (__attribute__((address_space(3))) unsigned int*)                  \
-              (blds[g] + (i64)(pbuf) * 2 * AE), 16, 0, 0);                   \
-  } while (0)
+  // LDS: [buf][khalf][A(16KB) | B(16KB)]
+#define G8_HALF(pbuf, h) (&smem[0][0] + ((i64)(pbuf) * 4 + (h) * 2) * HE)
+#define G8_STAGE(pbuf, h, kb)                                                  do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + (h) * 32 + ak8[g], zpage),                        (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + aoff[g]), 16, 0, 0);                           _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb[g], (kb) + (h) * 32 + bk8[g], tap, zpage),                (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + HE + boff[g]), 16, 0, 0);                    } while (0)
 
   f32x4 acc[8][4];
 #pragma unroll
@@ -97,42 +85,84 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nt = (k_hi - k_lo + G8_BK - 1) / G8_BK;
-  G8_STAGE(0, k_lo);
-  __syncthreads();
+  const int kfrag = (lane >> 4) * 8;
 
-  const int kfrag = (lane >> 4) * 8;            // 8 k-elems per lane
+#define G8_PHASE(pbuf, h, WAITN)                                               do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i) {                                                  int row = wm + i * 16 + (lane & 15);                                         af[i] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));              }                                                                            _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j) {                                                  int col = wn + j * 16 + (lane & 15);                                         bf[j] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));              }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i)                                                    _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             } while (0)
+
+  // prologue: T0 both halves in flight
+  G8_STAGE(0, 0, k_lo);
+  G8_STAGE(0, 1, k_lo);
   int p = 0;
-  for (int tile = 0; tile < nt; ++tile) {
-    if (tile + 1 < nt) G8_STAGE(p ^ 1, k_lo + (tile + 1) * G8_BK);
-    const u16* As = &smem[p][0];
-    const u16* Bs = &smem[p][AE];
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+  for (int tile = 0; tile < nt - 1; ++tile) {
+    // phase (T,0): read T.h0; stage (T+1).h0
+    {
+      // issue next stage FIRST? No: stage after the wait+frag-reads so the
+      // count at the NEXT phase boundary stays uniform.
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const u16* As_ = G8_HALF(p, 0);
+      const u16* Bs_ = As_ + HE;
       bf16x8 af[8], bf[4];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         int row = wm + i * 16 + (lane & 15);
-        int byte = sw16x32(row * 128 + (ks * 32 + kfrag) * 2);
         af[i] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(As + byte / 2));
+            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
       }
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         int col = wn + j * 16 + (lane & 15);
-        int byte = sw16x32(col * 128 + (ks * 32 + kfrag) * 2);
         bf[j] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(Bs + byte / 2));
+            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
       }
+      G8_STAGE(p ^ 1, 0, k_lo + (tile + 1) * G8_BK);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 8; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();
+    // phase (T,1): read T.h1; stage (T+1).h1
+    {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const u16* As_ = G8_HALF(p, 1);
+      const u16* Bs_ = As_ + HE;
+      bf16x8 af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int row = wm + i * 16 + (lane & 15);
+        af[i] = as_bf16x8(
+            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int col = wn + j * 16 + (lane & 15);
+        bf[j] = as_bf16x8(
+            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
+      }
+      G8_STAGE(p ^ 1, 1, k_lo + (tile + 1) * G8_BK);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
     p ^= 1;
   }
+  // final tile: h0 waits for h1's stage still outstanding (vmcnt 4),
+  // h1 drains everything (vmcnt 0)
+  G8_PHASE(p, 0, 4);
+  G8_PHASE(p, 1, 0);
+#undef G8_PHASE
+#undef G8_STAGE
+#undef G8_HALF
 #undef G8_STAGE
 
   // epilogue
