@@ -174,6 +174,12 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
 #undef G8_STAGE
 
   // epilogue
+  float bvj[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int col = n0 + wn + j * 16 + (lane & 15);
+    bvj[j] = (EPI == 1 || EPI == 2) && col < N ? bf2f(bias[col]) : 0.f;
+  }
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
@@ -181,7 +187,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
     for (int j = 0; j < 4; ++j) {
       int col = n0 + wn + j * 16 + (lane & 15);
       if (col >= N) continue;
-      float bv = (EPI == 1 || EPI == 2) ? bf2f(bias[col]) : 0.f;
+      float bv = bvj[j];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = row_base + r;

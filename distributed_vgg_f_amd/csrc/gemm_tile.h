@@ -452,6 +452,12 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
   }
 
   // ---------------------------------------------------------- epilogue
+  float bvj[NFRAG];
+#pragma unroll
+  for (int j = 0; j < NFRAG; ++j) {
+    int col = n0 + wn + j * 16 + (lane & 15);
+    bvj[j] = (EPI == 1 || EPI == 2) && col < N ? bf2f(bias[col]) : 0.f;
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
@@ -459,7 +465,7 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
     for (int j = 0; j < NFRAG; ++j) {
       int col = n0 + wn + j * 16 + (lane & 15);
       if (col >= N) continue;
-      float bv = (EPI == 1 || EPI == 2) ? bf2f(bias[col]) : 0.f;
+      float bv = bvj[j];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = row_base + r;
@@ -601,6 +607,12 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
 #undef STAGE_GLDS
 
   // epilogue (same as gemm_kernel)
+  float bvj[NFRAG];
+#pragma unroll
+  for (int j = 0; j < NFRAG; ++j) {
+    int col = n0 + wn + j * 16 + (lane & 15);
+    bvj[j] = (EPI == 1 || EPI == 2) && col < N ? bf2f(bias[col]) : 0.f;
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
@@ -608,7 +620,7 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
     for (int j = 0; j < NFRAG; ++j) {
       int col = n0 + wn + j * 16 + (lane & 15);
       if (col >= N) continue;
-      float bv = (EPI == 1 || EPI == 2) ? bf2f(bias[col]) : 0.f;
+      float bv = bvj[j];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = row_base + r;
