@@ -62,11 +62,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
 #pragma unroll
   for (int g = 0; g < 4; ++g) {
     int off = (wave * 4 + g) * 1024 + lane * 16;   // byte in 16 KB half
-    int row = off / 64, slot = (off % 64) / 16;
-    // conflict-free read swizzle: slot ^ ((row + row/4) & 3); the glds
-    // dest is lane-linear, so the swizzle goes on the SOURCE k-position
-    // (section 5.4 rule 21) and is repeated on the read side below.
-    int k8 = (slot ^ ((row + (row >> 2)) & 3)) * 8;
+    int row = off / 64, k8 = (off % 64) / 2;
     rsa[g] = ga.prep(m0 + row);
     ak8[g] = k8;
     aoff[g] = off / 2;                             // elem offset in A half
@@ -91,7 +87,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
   const int nt = (k_hi - k_lo + G8_BK - 1) / G8_BK;
   const int kfrag = (lane >> 4) * 8;
 
-#define G8_PHASE(pbuf, h, WAITN)                                               do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i) {                                                  int row = wm + i * 16 + (lane & 15);                                         int sl = ((kfrag >> 3) ^ ((row + (row >> 2)) & 3)) * 8;                      af[i] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(As_ + row * 32 + sl));              }                                                                            _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j) {                                                  int col = wn + j * 16 + (lane & 15);                                         int sl = ((kfrag >> 3) ^ ((col + (col >> 2)) & 3)) * 8;                      bf[j] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + sl));              }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i)                                                    _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             } while (0)
+#define G8_PHASE(pbuf, h, WAITN)                                               do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i) {                                                  int row = wm + i * 16 + (lane & 15);                                         af[i] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));              }                                                                            _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j) {                                                  int col = wn + j * 16 + (lane & 15);                                         bf[j] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));              }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i)                                                    _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             } while (0)
 
   // prologue: T0 both halves in flight
   G8_STAGE(0, 0, k_lo);
@@ -104,24 +100,22 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
       // count at the NEXT phase boundary stays uniform.
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      G8_STAGE(p ^ 1, 0, k_lo + (tile + 1) * G8_BK);
       const u16* As_ = G8_HALF(p, 0);
       const u16* Bs_ = As_ + HE;
       bf16x8 af[8], bf[4];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         int row = wm + i * 16 + (lane & 15);
-        int sl = ((kfrag >> 3) ^ ((row + (row >> 2)) & 3)) * 8;
         af[i] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(As_ + row * 32 + sl));
+            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
       }
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         int col = wn + j * 16 + (lane & 15);
-        int sl = ((kfrag >> 3) ^ ((col + (col >> 2)) & 3)) * 8;
         bf[j] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + sl));
+            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
       }
+      G8_STAGE(p ^ 1, 0, k_lo + (tile + 1) * G8_BK);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 8; ++i)
@@ -135,24 +129,22 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
     {
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      G8_STAGE(p ^ 1, 1, k_lo + (tile + 1) * G8_BK);
       const u16* As_ = G8_HALF(p, 1);
       const u16* Bs_ = As_ + HE;
       bf16x8 af[8], bf[4];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         int row = wm + i * 16 + (lane & 15);
-        int sl = ((kfrag >> 3) ^ ((row + (row >> 2)) & 3)) * 8;
         af[i] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(As_ + row * 32 + sl));
+            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
       }
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         int col = wn + j * 16 + (lane & 15);
-        int sl = ((kfrag >> 3) ^ ((col + (col >> 2)) & 3)) * 8;
         bf[j] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + sl));
+            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
       }
+      G8_STAGE(p ^ 1, 1, k_lo + (tile + 1) * G8_BK);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 8; ++i)
