@@ -21,7 +21,7 @@ OUT = os.path.join(ROOT, "distributed_vgg_f_amd", "_hipops.so")
 
 SOURCES = ["gemm_ops.hip", "mem.hip", "lrnpool.hip", "softbn.hip",
            "hipops.cpp"]
-HEADERS = ["common.h", "gemm_tile.h"]
+HEADERS = ["common.h", "gemm_tile.h", "gemm8.h"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

@@ -310,7 +310,7 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     bool bk64 = false;
     // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B)
     if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 &&
-        env_int("DVGG_GEMM8", 1)) {
+        env_int("DVGG_GEMM8", 0)) {  // experimental: pipeline race, see DESIGN.md
       int tiles8 = ceil_div(M, G8_BM) * ceil_div(N, G8_BN);
       if (splitk > 1) {
         dim3 g8(tiles8, splitk);
