@@ -54,14 +54,19 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
   // s_barrier — prefetched halves stay in flight across barriers
   // (cdna_hip_programming.md section 5 "Pipelining across barriers";
   // __syncthreads would drain the glds queue to vmcnt(0) every phase).
+  // A 16 KB k-half = 1024 sixteen-byte lane-loads = 512 threads x 2 groups
+  // (the v2 bug: this loop ran g<4 as in the full-tile v1 staging, so off
+  // covered 32 KB — rows 256..511 gathered garbage and aoff/boff overwrote
+  // the neighbouring halves. One half-stage is 4 glds/lane, which is what
+  // the vmcnt(4) phase waits below are counted for.)
   constexpr int HE = G8_BM * 32;              // elems per operand k-half
-  typename GA::Row rsa[4];
-  typename GB::Row rsb[4];
-  int ak8[4], bk8[4];
-  int aoff[4], boff[4];
+  typename GA::Row rsa[2];
+  typename GB::Row rsb[2];
+  int ak8[2], bk8[2];
+  int aoff[2], boff[2];
 #pragma unroll
-  for (int g = 0; g < 4; ++g) {
-    int off = (wave * 4 + g) * 1024 + lane * 16;   // byte in 16 KB half
+  for (int g = 0; g < 2; ++g) {
+    int off = (wave * 2 + g) * 1024 + lane * 16;   // byte in 16 KB half
     int row = off / 64, k8 = (off % 64) / 2;
     rsa[g] = ga.prep(m0 + row);
     ak8[g] = k8;
@@ -76,7 +81,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
 
   // LDS: [buf][khalf][A(16KB) | B(16KB)]
 #define G8_HALF(pbuf, h) (&smem[0][0] + ((i64)(pbuf) * 4 + (h) * 2) * HE)
-#define G8_STAGE(pbuf, h, kb)                                                  do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + (h) * 32 + ak8[g], zpage),                        (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + aoff[g]), 16, 0, 0);                           _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb[g], (kb) + (h) * 32 + bk8[g], tap, zpage),                (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + HE + boff[g]), 16, 0, 0);                    } while (0)
+#define G8_STAGE(pbuf, h, kb)                                                  do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < 2; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + (h) * 32 + ak8[g], zpage),                        (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + aoff[g]), 16, 0, 0);                           _Pragma("unroll")                                                            for (int g = 0; g < 2; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb[g], (kb) + (h) * 32 + bk8[g], tap, zpage),                (__attribute__((address_space(3))) unsigned int*)                                (G8_HALF(pbuf, h) + HE + boff[g]), 16, 0, 0);                    } while (0)
 
   f32x4 acc[8][4];
 #pragma unroll
