@@ -308,10 +308,22 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     // model incl. the K=64 1x1 single-tile case — occupancy beats barrier
     // count again. Template retained for future tuning.
     bool bk64 = false;
-    // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B)
-    if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 &&
-        env_int("DVGG_GEMM8", 0)) {  // experimental: pipeline race, see DESIGN.md
-      int tiles8 = ceil_div(M, G8_BM) * ceil_div(N, G8_BN);
+    // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B).
+    // Measured routing gates (VGG-F/VGG-16/ResNet A/B, profiles/):
+    //  * fill: at 1 workgroup/CU the deep tile needs enough blocks to
+    //    cover the chip — VGG-F conv3-5 (169 tiles) lose ~1.5%, VGG-16's
+    //    196-tile conv5 wins, so the line sits at 192 (DVGG_G8_MINTILES);
+    //  * contiguity: scattered A gathers (Cin=64 -> one tap per BK tile,
+    //    VGG-F conv2) lose to the 4-blocks/CU 128^2 kernel, which hides
+    //    gather latency with occupancy; dense K runs >= 128 win ~10-15%
+    //    per call (VGG-16 fwd/dgrad) (DVGG_G8_MINCONTIG).
+    // (split-K stays on the 128^2 kernel: gemm8's per-element fp32-atomic
+    // split-K epilogue measured -5% on VGG-F's fc layers.)
+    int tiles8 = ceil_div(M, G8_BM) * ceil_div(N, G8_BN);
+    if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 && splitk == 1 &&
+        env_int("DVGG_GEMM8", 1) &&
+        ga.kcontig() >= env_int("DVGG_G8_MINCONTIG", 128) &&
+        tiles8 >= env_int("DVGG_G8_MINTILES", 192)) {
       if (splitk > 1) {
         dim3 g8(tiles8, splitk);
         hipLaunchKernelGGL((gemm8_kernel<GA, GB, 0, true>), g8,

@@ -53,6 +53,7 @@ struct PlainGather {
   DEV const u16* ptr8(const Row& r, int k, const u16* zpage) const {
     return (r.valid && k < cols) ? r.rp + k : zpage;
   }
+  int kcontig() const { return cols; }  // host: dense K run for g8 gate
 };
 
 // which gathers can stage via global_load_lds (single 16B load or a zero
@@ -92,6 +93,7 @@ struct ConvFwdGather {
       return rs.nrow + ((i64)ih * W + iw) * Cin + c0;
     return zpage;
   }
+  int kcontig() const { return Cin; }   // host: K run inside one tap
 };
 template <> struct GldsOk<ConvFwdGather> { static constexpr bool value = true; };
 
@@ -138,6 +140,7 @@ struct ConvDgradGather {
     ok = ok && oh < OH && ow < OW;
     return ok ? rs.nrow + ((i64)oh * OW + ow) * Cout + c0 : zpage;
   }
+  int kcontig() const { return Cout; }  // host: K run inside one tap
 };
 template <> struct GldsOk<ConvDgradGather> { static constexpr bool value = true; };
 
@@ -267,6 +270,7 @@ struct WgradXgGather {
     u32 oh = dWk.div(r), owp = dWk.mod(r, oh);
     return rs.cib + ((i64)n * Hp + int(oh) + rs.kh) * Wg + owp + 8;
   }
+  int kcontig() const { return 0; }   // host: keep wgrad off the g8 route
 };
 template <> struct GldsOk<WgradXgGather> { static constexpr bool value = true; };
 
