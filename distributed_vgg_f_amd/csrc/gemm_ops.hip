@@ -107,6 +107,49 @@ __global__ void nhwc2cnhw_kernel(const u16* __restrict__ in,
   }
 }
 
+// Vectorized (b128 both sides) NHWC -> C,N,Hp,Wp transpose for C % 8 == 0.
+// 64(out-col) x 64(c) tile per block, 8x fewer global memory instructions
+// than the u16 path (which measured ~8x off bandwidth speed-of-light on
+// VGG-16: 223 us vs ~26 us for a 103 MB dy). The tile is indexed by OUT
+// column wp (so 16B stores are always aligned regardless of the halo
+// offset co); the source read is shifted by -co, and out-of-image
+// positions store explicit zeros into the (pre-zeroed) halo columns.
+// LDS slot XOR swizzle keeps both phases <= 2-way conflicted.
+__global__ void nhwc2cnhw8_kernel(const u16* __restrict__ in,
+                                  u16* __restrict__ out, int N, int H, int W,
+                                  int C, int Hp, int Wp, int ro, int co) {
+  __shared__ u16 tile8[64 * 64];
+  int nh = blockIdx.x;
+  int wb = blockIdx.y * 64, c0 = blockIdx.z * 64;
+  int n = nh / H, ih = nh % H;
+  int t = threadIdx.x;
+  const u16* inb = in + ((i64)n * H + ih) * W * C;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int wloc = i * 32 + (t >> 3), c8 = t & 7;
+    int iw = wb + wloc - co, c = c0 + c8 * 8;
+    s16x8 v = zero8();
+    if ((unsigned)iw < (unsigned)W && c < C)
+      v = *reinterpret_cast<const s16x8*>(inb + (i64)iw * C + c);
+    *reinterpret_cast<s16x8*>(&tile8[wloc * 64 + ((c8 ^ (wloc & 7)) * 8)]) = v;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int c = j * 32 + (t & 31), wv = (t >> 5) * 8;
+    int oc = c0 + c;
+    if (oc >= C || wb + wv >= Wp) continue;
+    union { s16x8 v; u16 e[8]; } pk;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int w = wv + k;
+      pk.e[k] = tile8[w * 64 + ((c >> 3) ^ (w & 7)) * 8 + (c & 7)];
+    }
+    *reinterpret_cast<s16x8*>(
+        out + (((i64)oc * N + n) * Hp + (ih + ro)) * Wp + wb + wv) = pk.v;
+  }
+}
+
 // dyTS builder for the glds wgrad: KW pre-shifted transposed copies of the
 // (masked) dy. Copy kw holds dy[n][oh][ow][co] at
 // dyTS[kw][co][n][oh][ow + kw - pad + 8]; everything else stays zero.
@@ -159,9 +202,15 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
            C, N, Hp, Wp, ro, co);
   auto out = cached_zero_buffer(key, numel, x.options())
       .narrow(0, 0, numel).view({C, N, Hp, Wp});
-  dim3 grid(N * H, ceil_div(W, 32), ceil_div(C, 32));
-  hipLaunchKernelGGL(nhwc2cnhw_kernel, grid, dim3(256), 0, cur_stream(),
-                     bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
+  if (C % 8 == 0 && Wp % 8 == 0) {
+    dim3 grid(N * H, ceil_div(Wp, 64), ceil_div(C, 64));
+    hipLaunchKernelGGL(nhwc2cnhw8_kernel, grid, dim3(256), 0, cur_stream(),
+                       bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
+  } else {
+    dim3 grid(N * H, ceil_div(W, 32), ceil_div(C, 32));
+    hipLaunchKernelGGL(nhwc2cnhw_kernel, grid, dim3(256), 0, cur_stream(),
+                       bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
+  }
   return out;
 }
 
@@ -180,6 +229,80 @@ __global__ void bias_slot_reduce_kernel(const float* __restrict__ db32p,
 #pragma unroll 8
   for (int s = 0; s < 256; ++s) a += db32p[(i64)s * C + c];
   db[c] = f2bf(a);
+}
+
+// Vectorized preamble (see bwd_preamble_kernel below for semantics):
+// b128 dy/y loads, b128 dym stores, 64x64 transpose tile with the same
+// XOR-slot LDS layout as nhwc2cnhw8_kernel, per-thread fp32 db partials
+// reduced through LDS to one atomic per channel per block. Requires
+// C % 8 == 0 and Wp % 8 == 0 (always true: C is a conv Cout).
+template <bool RELU, bool WRITE_DYM, bool DB>
+__global__ void bwd_preamble8_kernel(const u16* __restrict__ dy,
+                                     const u16* __restrict__ y,
+                                     u16* __restrict__ dym,
+                                     u16* __restrict__ dyT,
+                                     float* __restrict__ db32, int N, int H,
+                                     int W, int C, int Wp) {
+  __shared__ u16 tile8[64 * 64];
+  __shared__ float dbp[32][64];
+  int nh = blockIdx.x;
+  int wb = blockIdx.y * 64, c0 = blockIdx.z * 64;
+  int n = nh / H, ih = nh % H;
+  int t = threadIdx.x;
+  const i64 rbase = ((i64)n * H + ih) * W * C;
+  float dba[8];
+  if (DB)
+#pragma unroll
+    for (int k = 0; k < 8; ++k) dba[k] = 0.f;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int wloc = i * 32 + (t >> 3), c8 = t & 7;
+    int iw = wb + wloc, c = c0 + c8 * 8;
+    union { s16x8 v; u16 e[8]; } pk;
+    pk.v = zero8();
+    if (iw < W && c < C) {
+      i64 o = rbase + (i64)iw * C + c;
+      pk.v = *reinterpret_cast<const s16x8*>(dy + o);
+      if (RELU) {
+        union { s16x8 v; u16 e[8]; } yv;
+        yv.v = *reinterpret_cast<const s16x8*>(y + o);
+#pragma unroll
+        for (int k = 0; k < 8; ++k)
+          if (!(bf2f(yv.e[k]) > 0.f)) pk.e[k] = 0;
+      }
+      if (WRITE_DYM) *reinterpret_cast<s16x8*>(dym + o) = pk.v;
+      if (DB)
+#pragma unroll
+        for (int k = 0; k < 8; ++k) dba[k] += bf2f(pk.e[k]);
+    }
+    *reinterpret_cast<s16x8*>(&tile8[wloc * 64 + ((c8 ^ (wloc & 7)) * 8)]) =
+        pk.v;
+  }
+  if (DB) {
+#pragma unroll
+    for (int k = 0; k < 8; ++k) dbp[t >> 3][(t & 7) * 8 + k] = dba[k];
+  }
+  __syncthreads();
+  if (DB && t < 64 && c0 + t < C) {
+    float a = 0.f;
+#pragma unroll
+    for (int r = 0; r < 32; ++r) a += dbp[r][t];
+    atomicAdd(&db32[(i64)(blockIdx.x & 255) * C + c0 + t], a);
+  }
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int c = j * 32 + (t & 31), wv = (t >> 5) * 8;
+    int oc = c0 + c;
+    if (oc >= C || wb + wv >= Wp) continue;
+    union { s16x8 v; u16 e[8]; } pk;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      int w = wv + k;
+      pk.e[k] = tile8[w * 64 + ((c >> 3) ^ (w & 7)) * 8 + (c & 7)];
+    }
+    *reinterpret_cast<s16x8*>(
+        dyT + (((i64)oc * N + n) * H + ih) * Wp + wb + wv) = pk.v;
+  }
 }
 
 // ------------------------- fused conv-backward preamble (K10+K11+K16)
@@ -771,35 +894,31 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   at::Tensor dym = (relu && need_dx) ? at::empty_like(dy) : dy;
   auto db32 = need_db ? at::zeros({256, Cout}, dy.options().dtype(at::kFloat))
                       : at::Tensor();
-  dim3 pgrid(N * OH, ceil_div(OW, 32), ceil_div(Cout, 32));
   const u16* yp = relu ? bfp(y) : nullptr;
   float* dbp = need_db ? db32.data_ptr<float>() : nullptr;
   u16* dymp = (relu && need_dx) ? bfp_mut(dym) : nullptr;
   u16* dyTp = bfp_mut(dyT);
-  if (relu && need_dx && need_db)
-    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, true>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
-  else if (relu && need_dx)
-    hipLaunchKernelGGL((bwd_preamble_kernel<true, true, false>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
-  else if (relu && need_db)
-    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, true>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
-  else if (relu)
-    hipLaunchKernelGGL((bwd_preamble_kernel<true, false, false>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
-  else if (need_db)
-    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, true>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
-  else
-    hipLaunchKernelGGL((bwd_preamble_kernel<false, false, false>), pgrid,
-                       dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N, OH,
-                       OW, Cout, OWp);
+  bool vec8 = (Cout % 8 == 0) && (OWp % 8 == 0);
+  dim3 pgrid(N * OH, ceil_div(vec8 ? OWp : OW, vec8 ? 64 : 32),
+             ceil_div(Cout, vec8 ? 64 : 32));
+#define DVGG_PRE_LAUNCH(R, DM, DBF)                                          \
+  do {                                                                       \
+    if (vec8)                                                                \
+      hipLaunchKernelGGL((bwd_preamble8_kernel<R, DM, DBF>), pgrid,          \
+                         dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N,   \
+                         OH, OW, Cout, OWp);                                 \
+    else                                                                     \
+      hipLaunchKernelGGL((bwd_preamble_kernel<R, DM, DBF>), pgrid,           \
+                         dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N,   \
+                         OH, OW, Cout, OWp);                                 \
+  } while (0)
+  if (relu && need_dx && need_db)      DVGG_PRE_LAUNCH(true, true, true);
+  else if (relu && need_dx)            DVGG_PRE_LAUNCH(true, true, false);
+  else if (relu && need_db)            DVGG_PRE_LAUNCH(true, false, true);
+  else if (relu)                       DVGG_PRE_LAUNCH(true, false, false);
+  else if (need_db)                    DVGG_PRE_LAUNCH(false, false, true);
+  else                                 DVGG_PRE_LAUNCH(false, false, false);
+#undef DVGG_PRE_LAUNCH
   // note: when relu && !need_dx, dgrad is skipped so dym (== dy) is unused;
   // the masked values only exist in dyT. wgrad/db see masked dy either way.
 
