@@ -574,6 +574,22 @@ static void launch_gemm_wgrad(const GA& ga, const GB& gb, at::Tensor* c32,
   auto s = cur_stream();
   if (N >= 128 && M >= 256 && env_int("DVGG_WGRAD_BM256", 1)) {
     int tiles = ceil_div(M, 256) * ceil_div(N, 128);
+    // A/B verdict: XSLAB measured WORSE (VGG-F 5.48->5.59, VGG-16
+    // 32.6->34.5) — either the linear-id%8 XCD mapping assumption fails
+    // here or slab skew/atomic finalize costs dominate; default off,
+    // revisit with PMC L2 counters in round 2.
+    if (env_int("DVGG_WGRAD_XSLAB", 0)) {
+      // slab-major grid + slab count a multiple of 8 => every tile of one
+      // k-slab runs on one XCD; the dy B-slab is HBM-fetched once and
+      // re-read out of that XCD's L2 (see gemm_kernel XSLAB note).
+      int sk8 = std::max(8, (splitk + 7) / 8 * 8);
+      dim3 grid(sk8, tiles);
+      hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, 4, false, false, 256,
+                                      512, true>), grid, dim3(512), 0, s, ga,
+                         gb, nullptr, c32->data_ptr<float>(), nullptr, M, N,
+                         K, 0, K);
+      return;
+    }
     dim3 grid(tiles, splitk);
     hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, 4, false, false, 256,
                                     512>), grid, dim3(512), 0, s, ga, gb,

@@ -328,9 +328,15 @@ struct WgradDyGather {
 // BM_T/THREADS_T: 128/256 (4 waves, 2x2) or 256/512 (8 waves, 4x2 — same
 // per-wave 64-row tile; halves the B-panel re-read count for wgrad, whose
 // dy operand is re-read once per M-tile).
+// XSLAB (split-K only): grid is (slabs, tiles) instead of (tiles, slabs),
+// so the hardware's round-robin XCD assignment (linear workgroup id % 8,
+// x-fastest) pins every M/N-tile of one k-slab to ONE XCD when the slab
+// count is a multiple of 8. Each B k-slab is then fetched from HBM once
+// and re-read by the other tiles out of that XCD's L2 — the wgrad
+// B-panel re-read fix (DESIGN.md "known headroom" 1).
 template <class GA, class GB, int EPI, bool SPLITK, int NFRAG = 4,
           bool TNA = false, bool TNB = false, int BM_T = BM,
-          int THREADS_T = THREADS>
+          int THREADS_T = THREADS, bool XSLAB = false>
 __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
     GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
     const u16* __restrict__ bias, int M, int N, int K, int k_lo, int k_hi) {
@@ -340,14 +346,16 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
 
   const int tiles_n = (N + BN_T - 1) / BN_T;
   const int tiles_m = (M + BM_T - 1) / BM_T;
-  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  u32 wg = XSLAB ? blockIdx.y
+                 : xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
   const int tm = wg / tiles_n, tn = wg % tiles_n;
   const int m0 = tm * BM_T, n0 = tn * BN_T;
   if (SPLITK) {
-    int nchunk = gridDim.y;
+    int nchunk = XSLAB ? gridDim.x : gridDim.y;
+    int slab = XSLAB ? blockIdx.x : blockIdx.y;
     int kt = (K + BK - 1) / BK;
     int span = (kt + nchunk - 1) / nchunk * BK;
-    k_lo = blockIdx.y * span;
+    k_lo = slab * span;
     k_hi = min(K, k_lo + span);
     if (k_lo >= k_hi) return;  // empty slice: c32 is pre-zeroed
   }
