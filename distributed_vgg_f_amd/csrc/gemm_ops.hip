@@ -572,6 +572,21 @@ template <class GA, class GB>
 static void launch_gemm_wgrad(const GA& ga, const GB& gb, at::Tensor* c32,
                               int M, int N, int K, int splitk) {
   auto s = cur_stream();
+  if (N >= 128 && M >= 512 && env_int("DVGG_WGRAD_BM512", 0)) {
+    // A/B verdict: WORSE on every model (VGG-F 5.27->5.95, VGG-16
+    // 33.0->36.9, ResNet 45.9->47.4) — 1 block/CU (100 KB LDS) loses more
+    // latency hiding than the quartered dy B-panel re-reads buy, the same
+    // pattern as the BK=64 and XSLAB experiments. Kept for reference.
+    // 512-row tile, 16 waves: quarters the dy B-panel re-read count
+    // (tiles_m) at the cost of 1 block/CU (100 KB LDS).
+    int tiles = ceil_div(M, 512) * ceil_div(N, 128);
+    dim3 grid(tiles, splitk);
+    hipLaunchKernelGGL((gemm_kernel<GA, GB, 0, true, 4, false, false, 512,
+                                    1024>), grid, dim3(1024), 0, s, ga, gb,
+                       nullptr, c32->data_ptr<float>(), nullptr, M, N, K, 0,
+                       K);
+    return;
+  }
   if (N >= 128 && M >= 256 && env_int("DVGG_WGRAD_BM256", 1)) {
     int tiles = ceil_div(M, 256) * ceil_div(N, 128);
     // A/B verdict: XSLAB measured WORSE (VGG-F 5.48->5.59, VGG-16

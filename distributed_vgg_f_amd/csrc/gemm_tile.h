@@ -368,11 +368,15 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
   // B: BN_T*BK/2048; threads tile rows first (k-contiguous operands) or
   // k-rows first (TN operands: chunk = 8 columns of one reduction row).
   constexpr int CPT_A = BM_T * BK / 8 / THREADS_T;  // 2 at BK=32
-  constexpr int CPT_B = BN_T * BK / 8 / THREADS_T;  // 2 / 1
+  // B side may have fewer chunks than threads (BM_T=512/1024 threads):
+  // then CPT_B=1 and only threads with brow < BN_T stage B.
+  constexpr int CPT_BN = BN_T * BK / 8;             // total B chunks
+  constexpr int CPT_B = CPT_BN >= THREADS_T ? CPT_BN / THREADS_T : 1;
   constexpr int TPR_A = (BK / 8) / CPT_A;           // threads per A row
   constexpr int TPR_B = (BK / 8) / CPT_B;
   const int arow = t / TPR_A, ak8 = (t % TPR_A) * (CPT_A * 8);
   const int brow = t / TPR_B, bk8 = (t % TPR_B) * (CPT_B * 8);
+  const bool bstage = CPT_BN >= THREADS_T || t < CPT_BN;
   // TN assignments: chunk id -> (local k, col8)
   const int atk = (t * CPT_A) / (BM_T / 8), atc = ((t * CPT_A) % (BM_T / 8)) * 8;
   const int btk = (t * CPT_B) / (BN_T / 8), btc = ((t * CPT_B) % (BN_T / 8)) * 8;
@@ -394,6 +398,7 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
   for (int i = 0; i < CPT_A; ++i)
     ra[i] = TNA ? ga.load8(ga.prep(k_lo + atk), m0 + atc + i * 8)
                 : ga.load8(rsa, k_lo + ak8 + i * 8);
+  if (bstage)
 #pragma unroll
   for (int i = 0; i < CPT_B; ++i)
     rb[i] = TNB ? gb.load8(gb.prep(k_lo + btk), n0 + btc + i * 8)
@@ -413,6 +418,7 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
             ra[i];
       }
     }
+    if (bstage)
 #pragma unroll
     for (int i = 0; i < CPT_B; ++i) {
       if (TNB) {
@@ -432,6 +438,7 @@ __global__ __launch_bounds__(THREADS_T) void gemm_kernel(
       for (int i = 0; i < CPT_A; ++i)
         ra[i] = TNA ? ga.load8(ga.prep(kb + atk), m0 + atc + i * 8)
                     : ga.load8(rsa, kb + ak8 + i * 8);
+      if (bstage)
 #pragma unroll
       for (int i = 0; i < CPT_B; ++i)
         rb[i] = TNB ? gb.load8(gb.prep(kb + btk), n0 + btc + i * 8)
