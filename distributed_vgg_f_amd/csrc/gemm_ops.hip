@@ -14,7 +14,8 @@ namespace dvgg {
 
 // defined in mem.hip
 at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
-at::Tensor bias_grad(const at::Tensor&);
+at::Tensor bias_grad(const at::Tensor&,
+                     const c10::optional<at::Tensor>& = c10::nullopt);
 
 static inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -671,14 +672,17 @@ at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
   return dx;
 }
 
-at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy) {
+at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy,
+                        const c10::optional<at::Tensor>& dw_out) {
   check_bf16(x, "x"); check_bf16(dy, "dy");
   int M = x.size(0), K = x.size(1), N = dy.size(1);
   TORCH_CHECK(dy.size(0) == M, "bad fc wgrad shape");
   // TN-staged GEMM: x [M][K] and dy [M][N] consumed directly (reduction =
   // their outer dim, transposed in LDS during staging) — no global
   // transpose pass, no reduction padding.
-  auto dw = at::empty({K, N}, x.options());
+  // dw_out: write straight into the caller's grad-arena view (skips the
+  // autograd AccumulateGrad add; see engine/arena.py direct-grad contract).
+  auto dw = dw_out ? dw_out->view({K, N}) : at::empty({K, N}, x.options());
   int Mred = ((M + BK - 1) / BK) * BK;   // k-loop granularity (rows pred'd)
   PlainGather ga{bfp(x), M, K, K};       // rows = reduction
   PlainGather gb{bfp(dy), M, N, N};
@@ -890,7 +894,9 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
 std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                                    const at::Tensor& x, const at::Tensor& w,
                                    int64_t stride_, int64_t pad_, bool relu,
-                                   bool need_dx, bool need_db) {
+                                   bool need_dx, bool need_db,
+                                   const c10::optional<at::Tensor>& dw_out,
+                                   const c10::optional<at::Tensor>& db_out) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   int stride = int(stride_), pad = int(pad_);
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
@@ -908,9 +914,10 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor dx1;
     if (need_dx)
       dx1 = linear_dgrad(dy2, w.view({Cin, Cout})).view({N, H, W, Cin});
-    auto dw1 = linear_wgrad(x.view({M, Cin}), dy2).view({1, 1, Cin, Cout});
+    auto dw1 = linear_wgrad(x.view({M, Cin}), dy2, dw_out)
+        .view({1, 1, Cin, Cout});
     at::Tensor db1;
-    if (need_db) db1 = bias_grad(dy2);
+    if (need_db) db1 = bias_grad(dy2, db_out);
     return {dx1, dw1, db1};
   }
 
@@ -979,7 +986,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   }
 
   // ---- wgrad
-  auto dw = at::empty({KH, KW, Cin, Cout}, x.options());
+  auto dw = dw_out ? *dw_out : at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
   if (wgrad_glds_ok(Cin, stride, KW, pad)) {
     // dym is already ReLU-masked when relu (preamble wrote it); dyT from
@@ -988,7 +995,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     wgrad_glds(x, dymc, KH, KW, pad, dw2d, Cout);
     at::Tensor db;
     if (need_db) {
-      db = at::empty({Cout}, dy.options());
+      db = db_out ? *db_out : at::empty({Cout}, dy.options());
       hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
                          dim3(256), 0, s, db32.data_ptr<float>(),
                          bfp_mut(db), Cout);
@@ -1018,7 +1025,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
 
   at::Tensor db;
   if (need_db) {
-    db = at::empty({Cout}, dy.options());
+    db = db_out ? *db_out : at::empty({Cout}, dy.options());
     hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
                        dim3(256), 0, s, db32.data_ptr<float>(), bfp_mut(db),
                        Cout);

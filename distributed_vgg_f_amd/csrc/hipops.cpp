@@ -9,7 +9,8 @@ namespace dvgg {
 at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&,
                       const c10::optional<at::Tensor>&, bool);
 at::Tensor linear_dgrad(const at::Tensor&, const at::Tensor&);
-at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&);
+at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&,
+                        const c10::optional<at::Tensor>&);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&,
                       const c10::optional<at::Tensor>&, int64_t, int64_t,
                       bool);
@@ -19,12 +20,14 @@ at::Tensor conv2d_wgrad(const at::Tensor&, const at::Tensor&, int64_t,
                         int64_t, int64_t, int64_t);
 std::vector<at::Tensor> conv2d_bwd(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
-                                   int64_t, int64_t, bool, bool, bool);
+                                   int64_t, int64_t, bool, bool, bool,
+                                   const c10::optional<at::Tensor>&,
+                                   const c10::optional<at::Tensor>&);
 
 // mem.hip
 at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
 at::Tensor add_relu_fwd(const at::Tensor&, const at::Tensor&);
-at::Tensor bias_grad(const at::Tensor&);
+at::Tensor bias_grad(const at::Tensor&, const c10::optional<at::Tensor>&);
 void sgd_step(at::Tensor&, at::Tensor&, const at::Tensor&, at::Tensor&,
               double, double, double, double);
 std::vector<at::Tensor> dropout_fwd(const at::Tensor&, double, int64_t,
@@ -49,7 +52,9 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor&, const at::Tensor&,
                                const c10::optional<at::Tensor>&, double);
 std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
-                               const at::Tensor&, const at::Tensor&, bool);
+                               const at::Tensor&, const at::Tensor&, bool,
+                               const c10::optional<at::Tensor>&,
+                               const c10::optional<at::Tensor>&);
 
 }  // namespace dvgg
 
@@ -57,14 +62,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for distributed_vgg_f_amd";
   m.def("linear_fwd", &dvgg::linear_fwd);
   m.def("linear_dgrad", &dvgg::linear_dgrad);
-  m.def("linear_wgrad", &dvgg::linear_wgrad);
+  m.def("linear_wgrad", &dvgg::linear_wgrad, py::arg("x"), py::arg("dy"),
+        py::arg("dw_out") = c10::nullopt);
   m.def("conv2d_fwd", &dvgg::conv2d_fwd);
   m.def("conv2d_dgrad", &dvgg::conv2d_dgrad);
   m.def("conv2d_wgrad", &dvgg::conv2d_wgrad);
-  m.def("conv2d_bwd", &dvgg::conv2d_bwd);
+  m.def("conv2d_bwd", &dvgg::conv2d_bwd, py::arg("dy"), py::arg("y"),
+        py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("pad"),
+        py::arg("relu"), py::arg("need_dx"), py::arg("need_db"),
+        py::arg("dw_out") = c10::nullopt, py::arg("db_out") = c10::nullopt);
   m.def("relu_bwd", &dvgg::relu_bwd);
   m.def("add_relu_fwd", &dvgg::add_relu_fwd);
-  m.def("bias_grad", &dvgg::bias_grad);
+  m.def("bias_grad", &dvgg::bias_grad, py::arg("dy"),
+        py::arg("db_out") = c10::nullopt);
   m.def("sgd_step", &dvgg::sgd_step);
   m.def("dropout_fwd", &dvgg::dropout_fwd);
   m.def("dropout_bwd", &dvgg::dropout_bwd);
@@ -74,5 +84,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_bwd", &dvgg::maxpool_bwd);
   m.def("softmax_xent_fwd", &dvgg::softmax_xent_fwd);
   m.def("bn_fwd", &dvgg::bn_fwd);
-  m.def("bn_bwd", &dvgg::bn_bwd);
+  m.def("bn_bwd", &dvgg::bn_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("gamma"), py::arg("mean"), py::arg("rstd"), py::arg("yr"),
+        py::arg("relu"), py::arg("dg_out") = c10::nullopt,
+        py::arg("db_out") = c10::nullopt);
 }

@@ -123,7 +123,8 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
   if (i < n) out[i] = f2bf(in[i]);
 }
 
-at::Tensor bias_grad(const at::Tensor& dy) {
+at::Tensor bias_grad(const at::Tensor& dy,
+                     const c10::optional<at::Tensor>& db_out) {
   check_bf16(dy, "dy");
   int C = dy.size(-1);
   i64 R = dy.numel() / C;
@@ -132,7 +133,7 @@ at::Tensor bias_grad(const at::Tensor& dy) {
   dim3 grid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0, cur_stream(),
                      bfp(dy), db32.data_ptr<float>(), R, C, rpb);
-  auto db = at::empty({C}, dy.options());
+  auto db = db_out ? *db_out : at::empty({C}, dy.options());
   hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(ceil_div(C, 256)), dim3(256),
                      0, cur_stream(), db32.data_ptr<float>(), bfp_mut(db), C);
   return db;

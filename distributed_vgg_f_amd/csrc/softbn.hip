@@ -327,7 +327,9 @@ __global__ void bn_bwd_apply_kernel(const u16* __restrict__ dy,
 std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
                                const at::Tensor& gamma, const at::Tensor& mean,
                                const at::Tensor& rstd, const at::Tensor& yr,
-                               bool relu) {
+                               bool relu,
+                               const c10::optional<at::Tensor>& dg_out,
+                               const c10::optional<at::Tensor>& db_out) {
   check_bf16(dy, "dy"); check_bf16(x, "x");
   int C = dy.size(-1);
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
@@ -338,8 +340,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto coef = at::empty({3, C}, f32);
   auto cA = coef[0], cB = coef[1], cD = coef[2];
   auto dx = at::empty_like(dy);
-  auto dgamma = at::empty({C}, dy.options());
-  auto dbeta = at::empty({C}, dy.options());
+  auto dgamma = dg_out ? *dg_out : at::empty({C}, dy.options());
+  auto dbeta = db_out ? *db_out : at::empty({C}, dy.options());
   int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
   auto s = cur_stream();
   dim3 rgrid(ceil_div(C, 256), ceil_div(R, (i64)rpb));

@@ -84,7 +84,21 @@ class ArenaOptimizer:
             self.w32[ofs:ofs + n].copy_(p.detach().float().reshape(-1).to(device))
             view = self.wc[ofs:ofs + n].view(p.shape)
             newp = nn.Parameter(view, requires_grad=True)
-            newp.grad = self.g[ofs:ofs + n].view(p.shape)
+            gview = self.g[ofs:ofs + n].view(p.shape)
+            if device.type == "cuda":
+                # direct-grad contract (GPU): the backward kernels write the
+                # gradient straight into the arena view (_dvgg_gbuf) and
+                # return it; .grad starts None each step (detach_grads) so
+                # AccumulateGrad STORES the returned view instead of adding
+                # — no per-param add kernel, no arena zeroing. NOTE: grads
+                # OVERWRITE, so multi-backward accumulation between steps
+                # is not supported on the GPU path (nothing uses it).
+                newp._dvgg_gbuf = gview
+            else:
+                # CPU path (cpu_ref composite ops differentiate through
+                # torch autograd): classic .grad views + AccumulateGrad
+                # adds + zero_grad() per step.
+                newp.grad = gview
             owner = model.get_submodule(name.rsplit(".", 1)[0]) if "." in name \
                 else model
             owner._parameters[name.rsplit(".", 1)[-1]] = newp
@@ -116,6 +130,12 @@ class ArenaOptimizer:
 
     def zero_grad(self) -> None:
         self.g.zero_()
+
+    def detach_grads(self) -> None:
+        """Direct-grad mode (GPU): drop .grad so the next backward's
+        AccumulateGrad stores the arena view returned by our kernels."""
+        for s in self.slots:
+            s.param.grad = None
 
     def step(self, grad_scale: float = 1.0, lr: Optional[float] = None) -> None:
         ops.sgd_momentum_step(self.w32, self.wc, self.g, self.m32,

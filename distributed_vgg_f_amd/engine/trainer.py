@@ -88,9 +88,9 @@ class Trainer:
             ev_join.record(self._gcs)
         logits = self.model(self._gimg[i])
         loss = ops.softmax_xent(logits, self._glab[i]).sum()
+        self.arena.detach_grads()   # direct-grad: kernels write the arena
         loss.backward()
         self.arena.step(grad_scale=1.0 / self.global_batch)
-        self.arena.zero_grad()
         main.wait_event(ev_join)
         return loss
 
@@ -160,13 +160,18 @@ class Trainer:
         loss = ops.softmax_xent(logits, labels).sum()
         tm.stop("fwd")
         tm.start("bwd")
+        if self.device.type == "cuda":
+            # direct-grad: backward kernels overwrite the arena views and
+            # AccumulateGrad stores them back as .grad (no adds, no zeroing)
+            self.arena.detach_grads()
         loss.backward()
         self.reducer.finish()
         tm.stop("bwd")
         # loss summed per sample; SUM all-reduce over ranks => scale by 1/GB
         tm.start("opt")
         self.arena.step(grad_scale=1.0 / self.global_batch)
-        self.arena.zero_grad()
+        if self.device.type != "cuda":
+            self.arena.zero_grad()
         tm.stop("opt")
         self.step_i += 1
         return loss

@@ -35,6 +35,9 @@ class ConvFunction(torch.autograd.Function):
         ctx.save_for_backward(x, w, y if relu else torch.empty(0, device=x.device))
         ctx.stride, ctx.pad, ctx.relu = stride, pad, relu
         ctx.has_bias = bias is not None
+        # direct-grad: arena views attached by ArenaOptimizer (engine/arena.py)
+        ctx.gw = getattr(w, "_dvgg_gbuf", None)
+        ctx.gb = getattr(bias, "_dvgg_gbuf", None) if bias is not None else None
         return y
 
     @staticmethod
@@ -45,9 +48,11 @@ class ConvFunction(torch.autograd.Function):
         # fused path: one pass over dy produces the ReLU-masked dy, the
         # wgrad transpose and the bias grad, then the dgrad/wgrad GEMMs run
         # on the shared intermediates (csrc conv2d_bwd: K10/K11 fused into
-        # the K2/K3 preamble)
+        # the K2/K3 preamble). dw/db land directly in the grad arena when
+        # the params carry _dvgg_gbuf views (skips AccumulateGrad adds).
         dx, dw, db = _ext().conv2d_bwd(dy.contiguous(), y, x, w, ctx.stride,
-                                       ctx.pad, ctx.relu, need_dx, need_db)
+                                       ctx.pad, ctx.relu, need_dx, need_db,
+                                       ctx.gw, ctx.gb)
         return (dx if need_dx else None, dw, db if need_db else None,
                 None, None, None)
 
@@ -61,6 +66,8 @@ class LinearFunction(torch.autograd.Function):
         ctx.save_for_backward(x, w, y if relu else torch.empty(0, device=x.device))
         ctx.relu = relu
         ctx.has_bias = bias is not None
+        ctx.gw = getattr(w, "_dvgg_gbuf", None)
+        ctx.gb = getattr(bias, "_dvgg_gbuf", None) if bias is not None else None
         return y
 
     @staticmethod
@@ -73,9 +80,9 @@ class LinearFunction(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = _ext().linear_dgrad(dy, w)      # dy @ w^T
         if ctx.needs_input_grad[1]:
-            dw = _ext().linear_wgrad(x, dy)      # x^T @ dy
+            dw = _ext().linear_wgrad(x, dy, ctx.gw)  # x^T @ dy, into arena
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = _ext().bias_grad(dy)
+            db = _ext().bias_grad(dy, ctx.gb)
         return dx, dw, db, None
 
 
@@ -162,6 +169,8 @@ class BatchNormFunction(torch.autograd.Function):
         ctx.save_for_backward(x, gamma, mean, rstd,
                               y if relu else torch.empty(0, device=x.device))
         ctx.relu = relu
+        ctx.gg = getattr(gamma, "_dvgg_gbuf", None)
+        ctx.gb = getattr(beta, "_dvgg_gbuf", None)
         return y
 
     @staticmethod
@@ -169,7 +178,7 @@ class BatchNormFunction(torch.autograd.Function):
         x, gamma, mean, rstd, y = ctx.saved_tensors
         # ReLU mask fused into the BN backward kernels (no separate pass)
         dx, dgamma, dbeta = _ext().bn_bwd(dy.contiguous(), x, gamma, mean,
-                                          rstd, y, ctx.relu)
+                                          rstd, y, ctx.relu, ctx.gg, ctx.gb)
         return dx, dgamma, dbeta, None, None, None, None, None
 
 
