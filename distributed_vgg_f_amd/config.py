@@ -69,7 +69,8 @@ class Config:
     fault_inject_rank: int = -1
 
     # data
-    data: str = "synthetic"            # synthetic ImageNet-shaped tensors
+    data: str = "synthetic"            # synthetic | records (NPZ shards, R5)
+    data_dir: Optional[str] = None     # shard directory for data=records
     channels: int = 3
 
     def scaled_lr(self, world_size: int) -> float:

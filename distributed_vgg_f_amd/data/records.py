@@ -1,0 +1,172 @@
+"""Sharded NPZ record-file input pipeline (SURVEY.md §2.1 R5 real-data
+equivalent).
+
+The reference's R5 feeds real image batches; its dataset/decoder stack is
+unverified (SURVEY.md marks it [UNVERIFIED]) and this image has no network
+and no JPEG decoder (no PIL/torchvision), so the rebuild's on-disk format
+is *decoded* sharded NPZ records — the moral equivalent of TFRecord shards
+for this environment:
+
+    data_dir/
+      train-00000.npz   # images: [N,H,W,C] uint8 (or float32), labels: [N]
+      train-00001.npz
+      ...
+      val-00000.npz     # optional eval split
+
+Semantics:
+  * rank sharding: shard files are dealt round-robin by rank (world_size
+    aware), so ranks read disjoint data — the all-reduce then averages
+    over the true global batch, matching the reference's between-graph
+    data split (SURVEY.md §2.2 DP row).
+  * per-epoch shuffling: a seeded permutation of (shard, index) per epoch;
+    deterministic given (seed, epoch, rank).
+  * a background prefetch thread decodes/converts the next batch into
+    PINNED host memory while the GPU computes; next() hands out tensors
+    with the same interface as SyntheticPipeline (H2D overlap included).
+  * uint8 images are normalized to [-1, 1] in the conversion step
+    (float inputs are passed through unchanged).
+
+`write_shards()` is the companion writer (used by tests and by
+tools/make_records.py to convert arrays into shards).
+"""
+
+from __future__ import annotations
+
+import glob
+import os
+import queue
+import threading
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+
+def write_shards(images: np.ndarray, labels: np.ndarray, out_dir: str,
+                 shard_size: int = 1024, split: str = "train") -> List[str]:
+    """Write [N,H,W,C] images + [N] labels as NPZ shards; returns paths."""
+    assert images.ndim == 4 and len(labels) == len(images)
+    os.makedirs(out_dir, exist_ok=True)
+    paths = []
+    for si, ofs in enumerate(range(0, len(images), shard_size)):
+        path = os.path.join(out_dir, f"{split}-{si:05d}.npz")
+        np.savez(path, images=images[ofs:ofs + shard_size],
+                 labels=labels[ofs:ofs + shard_size])
+        paths.append(path)
+    return paths
+
+
+class RecordPipeline:
+    """Epoch-shuffled, rank-sharded reader over NPZ shards with a
+    prefetch thread and pinned-memory staging."""
+
+    def __init__(self, data_dir: str, batch_size: int,
+                 device: torch.device, dtype: torch.dtype,
+                 split: str = "train", rank: int = 0, world_size: int = 1,
+                 seed: int = 0, prefetch: int = 2):
+        pattern = os.path.join(data_dir, f"{split}-*.npz")
+        shards = sorted(glob.glob(pattern))
+        if not shards:
+            raise FileNotFoundError(
+                f"no record shards matching {pattern}; write them with "
+                "distributed_vgg_f_amd.data.records.write_shards / "
+                "tools/make_records.py")
+        # deal whole shards round-robin by rank; every rank needs >= 1
+        mine = shards[rank::world_size] if world_size > 1 else shards
+        if not mine:
+            mine = [shards[rank % len(shards)]]
+        self.shards = mine
+        self.batch_size = batch_size
+        self.device = device
+        self.dtype = dtype
+        self.seed = seed
+        self.rank = rank
+        self._pin = device.type == "cuda"
+        self._epoch = 0
+        self._q: "queue.Queue" = queue.Queue(maxsize=prefetch)
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._producer, daemon=True)
+        self._thread.start()
+        if self._pin:
+            self.copy_stream = torch.cuda.Stream(device)
+        else:
+            self.copy_stream = None
+        self._inflight: Optional[Tuple[torch.Tensor, torch.Tensor,
+                                       Optional[torch.cuda.Event]]] = None
+
+    # ---------------- producer (background thread, CPU) ----------------
+    def _iter_epoch(self, epoch: int):
+        rng = np.random.default_rng((self.seed, epoch, self.rank))
+        order = rng.permutation(len(self.shards))
+        carry_img: List[np.ndarray] = []
+        carry_lab: List[np.ndarray] = []
+        carried = 0
+        for si in order:
+            with np.load(self.shards[si]) as z:
+                img, lab = z["images"], z["labels"]
+            perm = rng.permutation(len(img))
+            img, lab = img[perm], lab[perm]
+            ofs = 0
+            while ofs < len(img):
+                take = min(self.batch_size - carried, len(img) - ofs)
+                carry_img.append(img[ofs:ofs + take])
+                carry_lab.append(lab[ofs:ofs + take])
+                carried += take
+                ofs += take
+                if carried == self.batch_size:
+                    yield (np.concatenate(carry_img),
+                           np.concatenate(carry_lab))
+                    carry_img, carry_lab, carried = [], [], 0
+        # drop the ragged tail (reference-style fixed batch shapes)
+
+    def _convert(self, img: np.ndarray, lab: np.ndarray):
+        t = torch.from_numpy(np.ascontiguousarray(img))
+        if t.dtype == torch.uint8:
+            t = t.to(torch.float32).div_(127.5).sub_(1.0)
+        t = t.to(self.dtype)
+        l = torch.from_numpy(np.ascontiguousarray(lab)).to(torch.long)
+        if self._pin:
+            t = t.pin_memory()
+            l = l.pin_memory()
+        return t, l
+
+    def _producer(self):
+        while not self._stop.is_set():
+            got = False
+            for img, lab in self._iter_epoch(self._epoch):
+                got = True
+                item = self._convert(img, lab)
+                while not self._stop.is_set():
+                    try:
+                        self._q.put(item, timeout=0.25)
+                        break
+                    except queue.Full:
+                        continue
+                if self._stop.is_set():
+                    return
+            self._epoch += 1
+            if not got:   # shards smaller than one batch: loop the data
+                raise RuntimeError("record shards hold fewer samples than "
+                                   "one batch; lower --batch-size")
+
+    # ---------------- consumer (training loop) ----------------
+    def next(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        himg, hlab = self._q.get()
+        if not self._pin:
+            return himg, hlab
+        with torch.cuda.stream(self.copy_stream):
+            dimg = himg.to(self.device, non_blocking=True)
+            dlab = hlab.to(self.device, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(self.copy_stream)
+        torch.cuda.current_stream().wait_event(ev)
+        return dimg, dlab
+
+    def close(self):
+        self._stop.set()
+        try:
+            while True:
+                self._q.get_nowait()
+        except queue.Empty:
+            pass
+        self._thread.join(timeout=2.0)

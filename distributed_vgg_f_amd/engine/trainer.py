@@ -16,7 +16,7 @@ import torch
 
 from .. import ops
 from ..config import Config
-from ..data import SyntheticPipeline
+from ..data import RecordPipeline, SyntheticPipeline
 from ..models import build_model
 from ..parallel.dist import DistContext, GradReducer, broadcast_init, \
     init_distributed
@@ -54,10 +54,17 @@ class Trainer:
                                     cfg.weight_decay, cfg.bucket_mb)
         broadcast_init(self.arena, self.ctx)
         self.reducer = GradReducer(self.arena, self.ctx, cfg.overlap_comm)
-        self.pipeline = SyntheticPipeline(cfg.batch_size, cfg.image_size,
-                                          cfg.channels, cfg.num_classes,
-                                          self.device, compute_dtype,
-                                          cfg.seed + self.ctx.rank)
+        if cfg.data == "records":
+            self.pipeline = RecordPipeline(cfg.data_dir, cfg.batch_size,
+                                           self.device, compute_dtype,
+                                           rank=self.ctx.rank,
+                                           world_size=self.ctx.world_size,
+                                           seed=cfg.seed)
+        else:
+            self.pipeline = SyntheticPipeline(cfg.batch_size, cfg.image_size,
+                                              cfg.channels, cfg.num_classes,
+                                              self.device, compute_dtype,
+                                              cfg.seed + self.ctx.rank)
         self.metrics = MetricsLogger(cfg.metrics_path, self.ctx.rank)
         self.timers = PhaseTimers(self.device, enabled=cfg.phase_timers)
         self.step_i = 0
@@ -101,6 +108,8 @@ class Trainer:
         collectives stay out of capture. Falls back to eager on any error."""
         if self.device.type != "cuda" or self.ctx.world_size > 1:
             return False
+        if not isinstance(self.pipeline, SyntheticPipeline):
+            return False   # capture pins fixed host buffers (synthetic pool)
         try:
             self._gimg_h = [self.pipeline.images[0], self.pipeline.images[1]]
             self._glab_h = [self.pipeline.labels[0], self.pipeline.labels[1]]
