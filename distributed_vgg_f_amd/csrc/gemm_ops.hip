@@ -151,6 +151,49 @@ __global__ void nhwc2cnhw8_kernel(const u16* __restrict__ in,
   }
 }
 
+// Phase-plane transpose for strided small-Cin wgrad (VGG-F conv1):
+// x [N,H,W,C] -> xP [C][s][s][N][Hq][Wqp] with ih = s*ihq + ph,
+// iw = s*iwq + pw (s = stride, pad = 0). One block per (n, ih): stage the
+// W*C row in LDS, then emit C*s phase rows of Wq elems with b128 stores.
+// Turns the stride-s wgrad A gather into the stride-1 funnel gather
+// (WgradXGather phl mode) — the scalar XDirect path measured ~87 TF on
+// conv1, this path removes the stride-s scatter entirely.
+__global__ void nhwc2phase_kernel(const u16* __restrict__ in,
+                                  u16* __restrict__ out, int N, int H, int W,
+                                  int C, int s, int Wq, int Hq, int Wqp) {
+  extern __shared__ u16 rowbuf[];          // W*C elems (host-checked <= 4096)
+  int nh = blockIdx.x;
+  int n = nh / H, ih = nh % H;
+  int t = threadIdx.x;
+  int elems = W * C;
+  const u16* src = in + (i64)nh * elems;
+  for (int e8 = t * 8; e8 < elems; e8 += blockDim.x * 8) {
+    if (e8 + 8 <= elems)
+      *reinterpret_cast<s16x8*>(&rowbuf[e8]) =
+          *reinterpret_cast<const s16x8*>(src + e8);
+    else
+      for (int j = e8; j < elems; ++j) rowbuf[j] = src[j];
+  }
+  __syncthreads();
+  int ph = ih % s, ihq = ih / s;
+  int rows = C * s;                        // (c, pw) phase rows
+  int chunks = (Wq + 7) / 8;
+  for (int job = t; job < rows * chunks; job += blockDim.x) {
+    int r = job / chunks, q = job % chunks;
+    int c = r / s, pw = r % s;
+    union { s16x8 v; u16 e[8]; } pk;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int iwq = q * 8 + j;
+      int iw = iwq * s + pw;
+      pk.e[j] = (iwq < Wq && iw < W) ? rowbuf[iw * C + c] : u16(0);
+    }
+    i64 plane = ((i64)c * s + ph) * s + pw;
+    *reinterpret_cast<s16x8*>(
+        out + ((plane * N + n) * Hq + ihq) * Wqp + q * 8) = pk.v;
+  }
+}
+
 // dyTS builder for the glds wgrad: KW pre-shifted transposed copies of the
 // (masked) dy. Copy kw holds dy[n][oh][ow][co] at
 // dyTS[kw][co][n][oh][ow + kw - pad + 8]; everything else stays zero.
@@ -212,6 +255,24 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
     hipLaunchKernelGGL(nhwc2cnhw_kernel, grid, dim3(256), 0, cur_stream(),
                        bfp(x), bfp_mut(out), N, H, W, C, Hp, Wp, ro, co);
   }
+  return out;
+}
+
+// Host wrapper for nhwc2phase_kernel: cached pre-zeroed planes buffer.
+static at::Tensor nhwc2phase(const at::Tensor& x, int s, int Hq, int Wqp) {
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int Wq = ceil_div(W, s);
+  TORCH_CHECK(W * C <= 4096, "phase transpose row too large");
+  i64 numel = (i64)C * s * s * N * Hq * Wqp;
+  char key[96];
+  snprintf(key, sizeof key, "q%d_%d_%d_%d_%d_%d", x.get_device(), C, s, N,
+           Hq, Wqp);
+  auto out = cached_zero_buffer(key, numel, x.options())
+      .narrow(0, 0, numel).view({C * s * s, N, Hq, Wqp});
+  dim3 grid(N * H);
+  hipLaunchKernelGGL(nhwc2phase_kernel, grid, dim3(256),
+                     (W * C + 8) * sizeof(u16), cur_stream(), bfp(x),
+                     bfp_mut(out), N, H, W, C, s, Wq, Hq, Wqp);
   return out;
 }
 
@@ -871,7 +932,21 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-  if (Cin % 8 != 0) {
+  bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
+                  pad == 0 && ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
+                  env_int("DVGG_WGRAD_PHASE", 1);
+  if (phase_ok) {
+    // stride-phase planes: the A gather becomes stride-1 funnel loads
+    int phl = stride == 4 ? 2 : 1;
+    int KHq = (KH - 1) / stride, KWq = (KW - 1) / stride;
+    int Hq = std::max(ceil_div(H, stride), OH + KHq + 1);
+    int Wqp = ceil8(std::max(ceil_div(W, stride), OWp + KWq));
+    auto xP = nhwc2phase(x, stride, Hq, Wqp);
+    WgradXGather ga{bfp(xP), Kfull, Cin, N, Hq, Wqp, OH, OWp, 1, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp)), phl};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  } else if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
                           MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
@@ -882,7 +957,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');  // [Cin][N][Hp][Wp]
     WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
-                    MDiv(u32(OWp))};
+                    MDiv(u32(OWp)), 0};
     launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
@@ -1007,7 +1082,21 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-  if (Cin % 8 != 0) {
+  bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
+                  pad == 0 && ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
+                  env_int("DVGG_WGRAD_PHASE", 1);
+  if (phase_ok) {
+    // stride-phase planes: the A gather becomes stride-1 funnel loads
+    int phl = stride == 4 ? 2 : 1;
+    int KHq = (KH - 1) / stride, KWq = (KW - 1) / stride;
+    int Hq = std::max(ceil_div(H, stride), OH + KHq + 1);
+    int Wqp = ceil8(std::max(ceil_div(W, stride), OWp + KWq));
+    auto xP = nhwc2phase(x, stride, Hq, Wqp);
+    WgradXGather ga{bfp(xP), Kfull, Cin, N, Hq, Wqp, OH, OWp, 1, Kred,
+                    MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
+                    MDiv(u32(OWp)), phl};
+    launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
+  } else if (Cin % 8 != 0) {
     WgradXDirectGather ga{bfp(x), Kfull, Cin, N, H, W, OH, OWp, stride,
                           pad, Kred, MDiv(u32(Cin)), MDiv(u32(KW)),
                           MDiv(u32(OH * OWp)), MDiv(u32(OWp))};
@@ -1018,7 +1107,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
     WgradXGather ga{bfp(xT), Kfull, Cin, N, Hp, Wp, OH, OWp, stride, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
-                    MDiv(u32(OWp))};
+                    MDiv(u32(OWp)), 0};
     launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);

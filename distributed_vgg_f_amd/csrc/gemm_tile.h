@@ -147,22 +147,32 @@ template <> struct GldsOk<ConvDgradGather> { static constexpr bool value = true;
 // Wgrad A (Cin % 8 == 0): virtual A[m=(kh,kw,ci)][k=(n,oh,ow-padded)] over
 // xTpad [Cin][N][H+2p][Wp] (halo-padded, zero-filled; +8 elem slack). The
 // k-space pads OW to OWp (mult of 8) so chunks never cross rows.
+// phl (phase-plane mode, stride 2/4 small-Cin wgrad): xT holds
+// [Cin][s][s] stride-phase planes of x (s = 1<<phl), each [N][Hp][Wp];
+// ih = s*ihq + ph, iw = s*iwq + pw. A row (kh,kw,ci) then reads plane
+// (ci, kh&(s-1), kw&(s-1)) at row offsets (kh>>phl, kw>>phl) — a STRIDE-1
+// gather, so the b128+alignbit funnel path applies to strided convs
+// (VGG-F conv1 s=4). phl=0 is the classic [Cin][N][Hp][Wp] layout.
 struct WgradXGather {
   const u16* xT;
   int Mrows, Cin, N, Hp, Wp, OH, OWp, stride, Klim;
   MDiv dCin, dKW, dOHOWp, dOWp;
+  int phl;
   struct Row { const u16* cib; int kh, kw; bool valid; };
   DEV Row prep(int m) const {
     u32 f = dCin.div(m), ci = dCin.mod(m, f);
     u32 kh = dKW.div(f), kw = dKW.mod(f, kh);
-    return {xT + (i64)ci * N * Hp * Wp, int(kh), int(kw), m < Mrows};
+    u32 s1 = (1u << phl) - 1u;
+    u32 plane = ((ci << phl | (kh & s1)) << phl) | (kw & s1);
+    return {xT + (i64)plane * N * Hp * Wp, int(kh >> phl), int(kw >> phl),
+            m < Mrows};
   }
   DEV s16x8 load8(const Row& rs, int k) const {
     if (!rs.valid || k >= Klim) return zero8();
     u32 n = dOHOWp.div(k), r = dOHOWp.mod(k, n);
     u32 oh = dOWp.div(r), ow = dOWp.mod(r, oh);
     s16x8 v;
-    if (stride == 1 && dKW.d <= 8) {
+    if (stride == 1 && int(dKW.d - 1) >> phl < 8) {
       // The row base (cib + ...*Wp) is 16B-aligned and ow is a multiple of
       // 8, so the only misalignment is +kw elements (kw < 8). Read the
       // aligned 32B window with TWO b128 loads (the kernel is VMEM-issue
