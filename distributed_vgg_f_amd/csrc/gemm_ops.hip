@@ -160,7 +160,8 @@ __global__ void nhwc2cnhw8_kernel(const u16* __restrict__ in,
 // conv1, this path removes the stride-s scatter entirely.
 __global__ void nhwc2phase_kernel(const u16* __restrict__ in,
                                   u16* __restrict__ out, int N, int H, int W,
-                                  int C, int s, int Wq, int Hq, int Wqp) {
+                                  int C, int s, int Wq, int Hq, int Wqp,
+                                  int pad) {
   extern __shared__ u16 rowbuf[];          // W*C elems (host-checked <= 4096)
   int nh = blockIdx.x;
   int n = nh / H, ih = nh % H;
@@ -175,7 +176,7 @@ __global__ void nhwc2phase_kernel(const u16* __restrict__ in,
       for (int j = e8; j < elems; ++j) rowbuf[j] = src[j];
   }
   __syncthreads();
-  int ph = ih % s, ihq = ih / s;
+  int ph = (ih + pad) % s, ihq = (ih + pad) / s;   // halo folded in
   int rows = C * s;                        // (c, pw) phase rows
   int chunks = (Wq + 7) / 8;
   for (int job = t; job < rows * chunks; job += blockDim.x) {
@@ -185,8 +186,9 @@ __global__ void nhwc2phase_kernel(const u16* __restrict__ in,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int iwq = q * 8 + j;
-      int iw = iwq * s + pw;
-      pk.e[j] = (iwq < Wq && iw < W) ? rowbuf[iw * C + c] : u16(0);
+      int iw = iwq * s + pw - pad;
+      pk.e[j] = (iwq < Wq && iw >= 0 && iw < W) ? rowbuf[iw * C + c]
+                                                : u16(0);
     }
     i64 plane = ((i64)c * s + ph) * s + pw;
     *reinterpret_cast<s16x8*>(
@@ -259,20 +261,21 @@ static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
 }
 
 // Host wrapper for nhwc2phase_kernel: cached pre-zeroed planes buffer.
-static at::Tensor nhwc2phase(const at::Tensor& x, int s, int Hq, int Wqp) {
+static at::Tensor nhwc2phase(const at::Tensor& x, int s, int Hq, int Wqp,
+                             int pad) {
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
-  int Wq = ceil_div(W, s);
+  int Wq = ceil_div(W + 2 * pad, s);
   TORCH_CHECK(W * C <= 4096, "phase transpose row too large");
   i64 numel = (i64)C * s * s * N * Hq * Wqp;
   char key[96];
-  snprintf(key, sizeof key, "q%d_%d_%d_%d_%d_%d", x.get_device(), C, s, N,
-           Hq, Wqp);
+  snprintf(key, sizeof key, "q%d_%d_%d_%d_%d_%d_%d", x.get_device(), C, s, N,
+           Hq, Wqp, pad);
   auto out = cached_zero_buffer(key, numel, x.options())
       .narrow(0, 0, numel).view({C * s * s, N, Hq, Wqp});
   dim3 grid(N * H);
   hipLaunchKernelGGL(nhwc2phase_kernel, grid, dim3(256),
                      (W * C + 8) * sizeof(u16), cur_stream(), bfp(x),
-                     bfp_mut(out), N, H, W, C, s, Wq, Hq, Wqp);
+                     bfp_mut(out), N, H, W, C, s, Wq, Hq, Wqp, pad);
   return out;
 }
 
@@ -933,15 +936,16 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
-                  pad == 0 && ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
+                  ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);
   if (phase_ok) {
     // stride-phase planes: the A gather becomes stride-1 funnel loads
+    // (padding is folded into the plane layout as a row/col offset)
     int phl = stride == 4 ? 2 : 1;
     int KHq = (KH - 1) / stride, KWq = (KW - 1) / stride;
-    int Hq = std::max(ceil_div(H, stride), OH + KHq + 1);
-    int Wqp = ceil8(std::max(ceil_div(W, stride), OWp + KWq));
-    auto xP = nhwc2phase(x, stride, Hq, Wqp);
+    int Hq = std::max(ceil_div(H + 2 * pad, stride), OH + KHq + 1);
+    int Wqp = ceil8(std::max(ceil_div(W + 2 * pad, stride), OWp + KWq));
+    auto xP = nhwc2phase(x, stride, Hq, Wqp, pad);
     WgradXGather ga{bfp(xP), Kfull, Cin, N, Hq, Wqp, OH, OWp, 1, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
                     MDiv(u32(OWp)), phl};
@@ -1083,15 +1087,16 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
   auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
-                  pad == 0 && ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
+                  ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);
   if (phase_ok) {
     // stride-phase planes: the A gather becomes stride-1 funnel loads
+    // (padding is folded into the plane layout as a row/col offset)
     int phl = stride == 4 ? 2 : 1;
     int KHq = (KH - 1) / stride, KWq = (KW - 1) / stride;
-    int Hq = std::max(ceil_div(H, stride), OH + KHq + 1);
-    int Wqp = ceil8(std::max(ceil_div(W, stride), OWp + KWq));
-    auto xP = nhwc2phase(x, stride, Hq, Wqp);
+    int Hq = std::max(ceil_div(H + 2 * pad, stride), OH + KHq + 1);
+    int Wqp = ceil8(std::max(ceil_div(W + 2 * pad, stride), OWp + KWq));
+    auto xP = nhwc2phase(x, stride, Hq, Wqp, pad);
     WgradXGather ga{bfp(xP), Kfull, Cin, N, Hq, Wqp, OH, OWp, 1, Kred,
                     MDiv(u32(Cin)), MDiv(u32(KW)), MDiv(u32(OH * OWp)),
                     MDiv(u32(OWp)), phl};
