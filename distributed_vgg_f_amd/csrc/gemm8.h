@@ -92,80 +92,32 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
   const int nt = (k_hi - k_lo + G8_BK - 1) / G8_BK;
   const int kfrag = (lane >> 4) * 8;
 
-#define G8_PHASE(pbuf, h, WAITN)                                               do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i) {                                                  int row = wm + i * 16 + (lane & 15);                                         af[i] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));              }                                                                            _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j) {                                                  int col = wn + j * 16 + (lane & 15);                                         bf[j] = as_bf16x8(                                                               *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));              }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 0; i < 8; ++i)                                                    _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             } while (0)
+// v3 fine interleave: bf[0..3]+af[0..1] read up front, then each 8-MFMA
+// fragment group issues the NEXT af pair before a counted lgkmcnt wait —
+// LDS reads for group g+1 run under group g's MFMAs instead of bunching
+// all 12 reads (and one big wait) ahead of all 32 MFMAs. The asm memory
+// clobbers pin the issue order; ds reads retire in order so lgkmcnt(2)
+// exactly covers "everything but the pair just issued".
+#define G8_RD(dst, base, idx16) \
+  dst = as_bf16x8(*reinterpret_cast<const s16x8*>((base) + (idx16)))
+#define G8_PHASE(pbuf, h, WAITN, STAGE_STMT)                                   do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8_RD(bf[j], Bs_, (wn + j * 16 + (lane & 15)) * 32 + kfrag);               G8_RD(af[0], As_, (wm + (lane & 15)) * 32 + kfrag);                          G8_RD(af[1], As_, (wm + 16 + (lane & 15)) * 32 + kfrag);                     asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8_RD(af[2 * g + 2], As_,                                                        (wm + (2 * g + 2) * 16 + (lane & 15)) * 32 + kfrag);                   G8_RD(af[2 * g + 3], As_,                                                        (wm + (2 * g + 3) * 16 + (lane & 15)) * 32 + kfrag);                   asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 2 * g; i < 2 * g + 2; ++i)                                        _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
 
   // prologue: T0 both halves in flight
   G8_STAGE(0, 0, k_lo);
   G8_STAGE(0, 1, k_lo);
   int p = 0;
   for (int tile = 0; tile < nt - 1; ++tile) {
-    // phase (T,0): read T.h0; stage (T+1).h0
-    {
-      // issue next stage FIRST? No: stage after the wait+frag-reads so the
-      // count at the NEXT phase boundary stays uniform.
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
-      const u16* As_ = G8_HALF(p, 0);
-      const u16* Bs_ = As_ + HE;
-      bf16x8 af[8], bf[4];
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        int row = wm + i * 16 + (lane & 15);
-        af[i] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        int col = wn + j * 16 + (lane & 15);
-        bf[j] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
-      }
-      G8_STAGE(p ^ 1, 0, k_lo + (tile + 1) * G8_BK);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf[j], acc[i][j], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-    }
-    // phase (T,1): read T.h1; stage (T+1).h1
-    {
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
-      const u16* As_ = G8_HALF(p, 1);
-      const u16* Bs_ = As_ + HE;
-      bf16x8 af[8], bf[4];
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        int row = wm + i * 16 + (lane & 15);
-        af[i] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(As_ + row * 32 + kfrag));
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        int col = wn + j * 16 + (lane & 15);
-        bf[j] = as_bf16x8(
-            *reinterpret_cast<const s16x8*>(Bs_ + col * 32 + kfrag));
-      }
-      G8_STAGE(p ^ 1, 1, k_lo + (tile + 1) * G8_BK);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf[j], acc[i][j], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-    }
+    // phase (T,h): read T.h (fine-interleaved); stage (T+1).h under it
+    G8_PHASE(p, 0, 4, G8_STAGE(p ^ 1, 0, k_lo + (tile + 1) * G8_BK));
+    G8_PHASE(p, 1, 4, G8_STAGE(p ^ 1, 1, k_lo + (tile + 1) * G8_BK));
     p ^= 1;
   }
   // final tile: h0 waits for h1's stage still outstanding (vmcnt 4),
   // h1 drains everything (vmcnt 0)
-  G8_PHASE(p, 0, 4);
-  G8_PHASE(p, 1, 0);
+  G8_PHASE(p, 0, 4, );
+  G8_PHASE(p, 1, 0, );
 #undef G8_PHASE
+#undef G8_RD
 #undef G8_STAGE
 #undef G8_HALF
 #undef G8_STAGE

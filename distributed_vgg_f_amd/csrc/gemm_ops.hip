@@ -539,8 +539,9 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B).
     // Measured routing gates (VGG-F/VGG-16/ResNet A/B, profiles/):
     //  * fill: at 1 workgroup/CU the deep tile needs enough blocks to
-    //    cover the chip — VGG-F conv3-5 (169 tiles) lose ~1.5%, VGG-16's
-    //    196-tile conv5 wins, so the line sits at 192 (DVGG_G8_MINTILES);
+    //    cover the chip; with the v3 fine interleave the break-even
+    //    dropped to ~96 tiles (VGG-F conv3-5 at 169 tiles now WIN:
+    //    5.04 -> 4.89 ms) (DVGG_G8_MINTILES);
     //  * contiguity: scattered A gathers (Cin=64 -> one tap per BK tile,
     //    VGG-F conv2) lose to the 4-blocks/CU 128^2 kernel, which hides
     //    gather latency with occupancy; dense K runs >= 128 win ~10-15%
@@ -551,7 +552,7 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 && splitk == 1 &&
         env_int("DVGG_GEMM8", 1) &&
         ga.kcontig() >= env_int("DVGG_G8_MINCONTIG", 128) &&
-        tiles8 >= env_int("DVGG_G8_MINTILES", 192)) {
+        tiles8 >= env_int("DVGG_G8_MINTILES", 96)) {
       if (splitk > 1) {
         dim3 g8(tiles8, splitk);
         hipLaunchKernelGGL((gemm8_kernel<GA, GB, 0, true>), g8,
