@@ -536,6 +536,8 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     // model incl. the K=64 1x1 single-tile case — occupancy beats barrier
     // count again. Template retained for future tuning.
     bool bk64 = false;
+    // 3-buffer counted-vmcnt ring variant (A/B; see gemm_glds3_kernel)
+    bool g3 = env_int("DVGG_GLDS3", 1);  // measured: VGG-F -1.5%, others flat+
     // 256x256 8-wave deep tile for wide-N shapes (DVGG_GEMM8 to A/B).
     // Measured routing gates (VGG-F/VGG-16/ResNet A/B, profiles/):
     //  * fill: at 1 workgroup/CU the deep tile needs enough blocks to
@@ -584,7 +586,7 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
           return;
       }
     }
-#define GLDS_DISPATCH(EPIC, SPLITC, GRID, CPTR, BPTR)                            do {                                                                           if (bk64)                                                                      hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 64>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else                                                                           hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);        } while (0)
+#define GLDS_DISPATCH(EPIC, SPLITC, GRID, CPTR, BPTR)                            do {                                                                           if (g3)                                                                        hipLaunchKernelGGL((gemm_glds3_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else if (bk64)                                                                 hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 64>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);          else                                                                           hipLaunchKernelGGL((gemm_glds_kernel<GA, GB, EPIC, SPLITC, NFRAG,                                                 32>), GRID, dim3(THREADS), 0,                              s, ga, gb, op, CPTR, BPTR, zp, M, N, K, 0, K);        } while (0)
     if (splitk > 1) {
       GLDS_DISPATCH(0, true, gridk, cp, nullptr);
       return;

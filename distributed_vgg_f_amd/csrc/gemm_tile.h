@@ -667,4 +667,165 @@ __global__ __launch_bounds__(THREADS) void gemm_glds_kernel(
   }
 }
 
+// 3-buffer-ring + counted-vmcnt variant of gemm_glds_kernel (DVGG_GLDS3).
+// The 2-buffer kernel drains vmcnt(0) at every tile barrier (the glds
+// prefetch only spans ONE tile's MFMAs, so each tile eats most of the HBM
+// latency and relies on 4-block occupancy to hide it). Here stages are
+// issued TWO k-tiles ahead into a 48 KB ring (3 blocks/CU for NFRAG=4),
+// the per-tile wait is a counted vmcnt(4), one raw barrier per tile
+// fences the ring (write target (t+2)%3 was last read at t-1, two
+// barriers back), and the LDS reads are fine-interleaved under the MFMA
+// row groups with counted lgkmcnt waits (cf. gemm8 v3).
+template <class GA, class GB, int EPI, bool SPLITK, int NFRAG, int BKT = 32>
+__global__ __launch_bounds__(THREADS) void gemm_glds3_kernel(
+    GA ga, GB gb, u16* __restrict__ C, float* __restrict__ c32,
+    const u16* __restrict__ bias, const u16* __restrict__ zpage, int M, int N,
+    int K, int k_lo, int k_hi) {
+  constexpr int BN_T = 32 * NFRAG;
+  constexpr int AE = BM * BKT;
+  constexpr int BE = BN_T * BKT;
+  __shared__ u16 smem[3][AE + BE];
+
+  const int tiles_n = (N + BN_T - 1) / BN_T;
+  const int tiles_m = (M + BM - 1) / BM;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int tm = wg / tiles_n, tn = wg % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN_T;
+  if (SPLITK) {
+    int nchunk = gridDim.y;
+    int kt = (K + BKT - 1) / BKT;
+    int span = (kt + nchunk - 1) / nchunk * BKT;
+    k_lo = blockIdx.y * span;
+    k_hi = min(K, k_lo + span);
+    if (k_lo >= k_hi) return;
+  }
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
+
+  constexpr int AGI = AE * 2 / 1024 / 4;
+  constexpr int BGI = BE * 2 / 1024 / 4;
+  typename GA::Row rsa[AGI];
+  int ak8[AGI];
+  int aoff[AGI];
+#pragma unroll
+  for (int g = 0; g < AGI; ++g) {
+    int off = (wave * AGI + g) * 512 + lane * 8;
+    rsa[g] = ga.prep(m0 + off / BKT);
+    ak8[g] = off % BKT;
+    aoff[g] = (wave * AGI + g) * 512;
+  }
+  typename GB::Row rsb[BGI];
+  int bk8[BGI];
+  int boff[BGI];
+#pragma unroll
+  for (int g = 0; g < BGI; ++g) {
+    int off = (wave * BGI + g) * 512 + lane * 8;
+    rsb[g] = gb.prep(n0 + off / BKT);
+    bk8[g] = off % BKT;
+    boff[g] = AE + (wave * BGI + g) * 512;
+  }
+  constexpr int BUF = AE + BE;
+  int tap = 0;
+  if constexpr (HasTap<GA>::value) tap = ga.block_tap(m0);
+  (void)tap;
+
+#define STAGE_G3(pbuf, kb)                                                     do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < AGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + ak8[g], zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (&smem[0][0] + (i64)(pbuf) * BUF + aoff[g]), 16, 0, 0);            _Pragma("unroll")                                                            for (int g = 0; g < BGI; ++g)                                                  __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb[g], (kb) + bk8[g], tap, zpage),                                   (__attribute__((address_space(3))) unsigned int*)                                (&smem[0][0] + (i64)(pbuf) * BUF + boff[g]), 16, 0, 0);          } while (0)
+
+  f32x4 acc[4][NFRAG];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = (k_hi - k_lo + BKT - 1) / BKT;
+  const int kfrag = (lane >> 4) * 8;
+  STAGE_G3(0, k_lo);
+  if (nt > 1) STAGE_G3(1, k_lo + BKT);
+
+  int p = 0;
+  for (int tile = 0; tile < nt; ++tile) {
+    // keep exactly ONE stage (the newest, AGI+BGI ops) in flight across
+    // the wait — NFRAG=2 stages are 3 ops, not 4
+    if (tile + 1 < nt) {
+      if constexpr (AGI + BGI == 3)
+        asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+      else if constexpr (AGI + BGI == 4)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const u16* As_ = &smem[0][0] + (i64)p * BUF;
+    const u16* Bs_ = As_ + AE;
+#pragma unroll
+    for (int ks = 0; ks < BKT / 32; ++ks) {
+      bf16x8 af[4], bf[NFRAG];
+#pragma unroll
+      for (int j = 0; j < NFRAG; ++j) {
+        int col = wn + j * 16 + (lane & 15);
+        bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            Bs_ + col * BKT + ks * 32 + kfrag));
+      }
+      af[0] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          As_ + (wm + (lane & 15)) * BKT + ks * 32 + kfrag));
+      asm volatile("" ::: "memory");
+      if (ks == 0 && tile + 2 < nt)
+        STAGE_G3((tile + 2) % 3, k_lo + (tile + 2) * BKT);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        if (i < 3) {
+          af[i + 1] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+              As_ + (wm + (i + 1) * 16 + (lane & 15)) * BKT + ks * 32 +
+              kfrag));
+          asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int j = 0; j < NFRAG; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+    p = p + 1 == 3 ? 0 : p + 1;
+  }
+#undef STAGE_G3
+
+  // epilogue (same as gemm_glds_kernel)
+  float bvj[NFRAG];
+#pragma unroll
+  for (int j = 0; j < NFRAG; ++j) {
+    int col = n0 + wn + j * 16 + (lane & 15);
+    bvj[j] = (EPI == 1 || EPI == 2) && col < N ? bf2f(bias[col]) : 0.f;
+  }
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = n0 + wn + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = bvj[j];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row_base + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r];
+        if (SPLITK) {
+          atomicAdd(&c32[(i64)row * N + col], v);
+        } else {
+          v += bv;
+          if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
+          C[(i64)row * N + col] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
 }  // namespace dvgg
