@@ -27,7 +27,9 @@ from .metrics import MetricsLogger, PhaseTimers
 
 def _device(ctx: DistContext) -> torch.device:
     if torch.cuda.is_available():
-        return torch.device("cuda", ctx.local_rank)
+        # tolerate more ranks than GPUs (single-GPU multi-rank smoke tests)
+        return torch.device("cuda",
+                            ctx.local_rank % max(torch.cuda.device_count(), 1))
     return torch.device("cpu")
 
 
