@@ -216,17 +216,34 @@ class Trainer:
             ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i)
 
     # ------------------------------------------------------------------ #
+    def _eval_pipeline(self):
+        """R7 held-out data: the val-* record shards when data=records and
+        they exist; otherwise the training pipeline (synthetic)."""
+        if self.cfg.data == "records":
+            import glob as _glob
+            import os as _os
+            if _glob.glob(_os.path.join(self.cfg.data_dir, "val-*.npz")):
+                if not hasattr(self, "_val_pipeline"):
+                    dtype = self.arena.compute_dtype
+                    self._val_pipeline = RecordPipeline(
+                        self.cfg.data_dir, self.cfg.batch_size, self.device,
+                        dtype, split="val", rank=self.ctx.rank,
+                        world_size=self.ctx.world_size, seed=self.cfg.seed)
+                return self._val_pipeline
+        return self.pipeline
+
     @torch.no_grad()
     def evaluate(self, steps: Optional[int] = None) -> dict:
-        """R7: accuracy over held-out (synthetic) batches from the latest
-        checkpoint state."""
+        """R7: accuracy over held-out batches (val record shards when
+        available, else the synthetic pipeline) from the current state."""
         cfg = self.cfg
         steps = steps if steps is not None else cfg.eval_steps
+        pipeline = self._eval_pipeline()
         self.model.eval()
         correct = torch.zeros((), dtype=torch.long, device=self.device)
         total = 0
         for _ in range(steps):
-            images, labels = self.pipeline.next()
+            images, labels = pipeline.next()
             logits = self.model(images)
             correct += (logits.float().argmax(1) == labels).sum()
             total += labels.numel()

@@ -89,3 +89,22 @@ def test_trainer_on_records_cpu(tmp_path):
     l1 = t.train_step()
     assert float(l0) == float(l0) and float(l1) == float(l1)  # finite
     t.pipeline.close()
+
+
+def test_eval_uses_val_split(tmp_path):
+    rng = np.random.default_rng(1)
+    img = rng.integers(0, 256, (32, 64, 64, 3), dtype=np.uint8)
+    lab = np.zeros(32, dtype=np.int64)
+    write_shards(img, lab, str(tmp_path), shard_size=16, split="train")
+    # val split: all label 3 so accuracy is measurable against predictions
+    write_shards(img, np.full(32, 3, dtype=np.int64), str(tmp_path),
+                 shard_size=16, split="val")
+    cfg = Config(model="vggf", image_size=64, num_classes=10, batch_size=8,
+                 precision="fp32", train_steps=1, eval_steps=2,
+                 data="records", data_dir=str(tmp_path))
+    t = Trainer(cfg)
+    res = t.evaluate()
+    assert res["samples"] == 16
+    assert t._eval_pipeline() is t._val_pipeline   # val shards were picked
+    t.pipeline.close()
+    t._val_pipeline.close()
