@@ -310,3 +310,38 @@ def test_wgrad_splitk_stress_repeat():
     for rep in range(5):
         dw = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
         check(dw, w32.grad, 0.06, f"wgrad_stress_rep{rep}")
+
+
+# ---- kernel-variant regression guards: the routing env knobs select
+# between gemm_glds_kernel (2-buffer), gemm_glds3_kernel (counted ring,
+# default) and gemm8_kernel (deep tile); every variant must stay
+# numerically correct even while the default routing evolves.
+VARIANT_ENVS = [
+    {"DVGG_GLDS3": "0"},                                   # legacy 2-buffer
+    {"DVGG_GLDS3": "1"},                                   # counted ring
+    {"DVGG_GEMM8": "1", "DVGG_G8_MINTILES": "0",
+     "DVGG_G8_MINCONTIG": "0"},                            # deep tile, forced
+    {"DVGG_GEMM8": "0"},                                   # deep tile off
+]
+
+
+@pytest.mark.parametrize("env", VARIANT_ENVS,
+                         ids=["glds2", "glds3", "gemm8-on", "gemm8-off"])
+def test_conv_fwd_dgrad_variants(env, monkeypatch):
+    import os
+    for k_, v_ in env.items():
+        monkeypatch.setenv(k_, v_)
+    torch.manual_seed(21)
+    N, H, W, Ci, Co, k, s, p = 2, 13, 13, 256, 256, 3, 1, 1
+    x = rnd(N, H, W, Ci, scale=0.5)
+    w = rnd(k, k, Ci, Co, scale=1.0 / math.sqrt(k * k * Ci))
+    b = rnd(Co, scale=0.1)
+    y = ext.conv2d_fwd(x.to(DEV), w.to(DEV), b.to(DEV), s, p, True)
+    ref = cpu_ref.conv2d(x.float(), w.float(), b.float(), s, p, relu=True)
+    check(y, ref, 0.05, "variant_fwd")
+    dy = rnd(N, H, W, Co, scale=0.5)
+    dx = ext.conv2d_dgrad(dy.to(DEV), w.to(DEV), H, W, s, p)
+    xr = x.float().requires_grad_(True)
+    yr = cpu_ref.conv2d(xr, w.float(), None, s, p)
+    yr.backward(dy.float())
+    check(dx, xr.grad, 0.05, "variant_dgrad")
