@@ -50,8 +50,9 @@ def main():
               file=sys.stderr)
     n_gpus = world if world > 1 else 1
 
-    device = torch.device("cuda", ctx.local_rank) if use_gpu \
-        else torch.device("cpu")
+    device = torch.device(
+        "cuda", ctx.local_rank % max(torch.cuda.device_count(), 1)) \
+        if use_gpu else torch.device("cpu")
     t = Trainer(cfg, ctx=ctx, device=device)
 
     def barrier_sync():
@@ -69,8 +70,9 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks
-    et = torch.tensor([elapsed], dtype=torch.float64)
+    # MAX over ranks (RCCL needs the tensor on the GPU)
+    et = torch.tensor([elapsed], dtype=torch.float64,
+                      device=device if ctx.backend == "nccl" else "cpu")
     if world > 1:
         torch.distributed.all_reduce(et, op=torch.distributed.ReduceOp.MAX)
     elapsed = float(et[0])
