@@ -85,7 +85,6 @@ def test_eval_runs():
 
 
 def test_checkpoint_roundtrip_gpu(tmp_path):
-    from distributed_vgg_f_amd.engine import checkpoint as ckpt
     cfg = Config(model="vggf", batch_size=4, train_steps=2, log_every=0,
                  checkpoint_dir=str(tmp_path))
     t = Trainer(cfg, device=DEV)
