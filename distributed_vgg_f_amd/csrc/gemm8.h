@@ -154,4 +154,108 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
   }
 }
 
+
+// 256x128 (BN=128) flavor of the deep tile for N=128 layers (VGG-16
+// conv2_x, ResNet layer2 3x3): 8 waves as 4(M) x 2(N), per-wave 64x64 =
+// 4x4 fragments, BK=64 staged as two k-halves of (A 16 KB + B 8 KB),
+// 2 ring buffers = 96 KB LDS at 1 workgroup/CU. Same counted-vmcnt
+// pipeline and fine interleave as gemm8_kernel v3; one half-stage is
+// 3 glds/lane (2 A groups + 1 B group), hence the vmcnt(3) waits.
+template <class GA, class GB, int EPI>
+__global__ __launch_bounds__(G8_THREADS) void gemm8n_kernel(
+    GA ga, GB gb, u16* __restrict__ C, const u16* __restrict__ bias,
+    const u16* __restrict__ zpage, int M, int N, int K) {
+  constexpr int BNN = 128;
+  constexpr int HEA = G8_BM * 32;             // A elems per k-half (16 KB)
+  constexpr int HEB = BNN * 32;               // B elems per k-half (8 KB)
+  constexpr int HB = HEA + HEB;
+  __shared__ u16 smem[2][2 * HB];
+
+  const int tiles_n = (N + BNN - 1) / BNN;
+  const int tiles_m = (M + G8_BM - 1) / G8_BM;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int tm = wg / tiles_n, tn = wg % tiles_n;
+  const int m0 = tm * G8_BM, n0 = tn * BNN;
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * 64;
+
+  typename GA::Row rsa[2];
+  int ak8[2], aoff[2];
+#pragma unroll
+  for (int g = 0; g < 2; ++g) {
+    int off = (wave * 2 + g) * 1024 + lane * 16;   // byte in 16 KB A half
+    rsa[g] = ga.prep(m0 + off / 64);
+    ak8[g] = (off % 64) / 2;
+    aoff[g] = off / 2;
+  }
+  typename GB::Row rsb;
+  int bk8, boff;
+  {
+    int off = wave * 1024 + lane * 16;             // byte in 8 KB B half
+    rsb = gb.prep(n0 + off / 64);
+    bk8 = (off % 64) / 2;
+    boff = off / 2;
+  }
+  int tap = 0;
+  if constexpr (HasTap<GA>::value) tap = ga.block_tap(m0);
+  (void)tap;
+
+#define G8N_HALF(pbuf, h) (&smem[0][0] + ((i64)(pbuf) * 2 + (h)) * HB)
+#define G8N_STAGE(pbuf, h, kb)                                                 do {                                                                           _Pragma("unroll")                                                            for (int g = 0; g < 2; ++g)                                                    __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          ga.ptr8(rsa[g], (kb) + (h) * 32 + ak8[g], zpage),                        (__attribute__((address_space(3))) unsigned int*)                                (G8N_HALF(pbuf, h) + aoff[g]), 16, 0, 0);                          __builtin_amdgcn_global_load_lds(                                                (const __attribute__((address_space(1))) unsigned int*)                          gb_src(gb, rsb, (kb) + (h) * 32 + bk8, tap, zpage),                      (__attribute__((address_space(3))) unsigned int*)                                (G8N_HALF(pbuf, h) + HEA + boff), 16, 0, 0);                     } while (0)
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = (K + G8_BK - 1) / G8_BK;
+  const int kfrag = (lane >> 4) * 8;
+
+#define G8N_RD(dst, base, idx16)                                               dst = as_bf16x8(*reinterpret_cast<const s16x8*>((base) + (idx16)))
+#define G8N_PHASE(pbuf, h, WAITN, STAGE_STMT)                                  do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8N_HALF(pbuf, h);                                          const u16* Bs_ = As_ + HEA;                                                  bf16x8 af[4], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8N_RD(bf[j], Bs_, (wn + j * 16 + (lane & 15)) * 32 + kfrag);              G8N_RD(af[0], As_, (wm + (lane & 15)) * 32 + kfrag);                         asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8N_RD(af[g + 1], As_,                                                           (wm + (g + 1) * 16 + (lane & 15)) * 32 + kfrag);                       asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[g][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[g], bf[j], acc[g][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
+
+  G8N_STAGE(0, 0, 0);
+  G8N_STAGE(0, 1, 0);
+  int p = 0;
+  for (int tile = 0; tile < nt - 1; ++tile) {
+    G8N_PHASE(p, 0, 3, G8N_STAGE(p ^ 1, 0, (tile + 1) * G8_BK));
+    G8N_PHASE(p, 1, 3, G8N_STAGE(p ^ 1, 1, (tile + 1) * G8_BK));
+    p ^= 1;
+  }
+  G8N_PHASE(p, 0, 3, );
+  G8N_PHASE(p, 1, 0, );
+#undef G8N_PHASE
+#undef G8N_RD
+#undef G8N_STAGE
+#undef G8N_HALF
+
+  // epilogue
+  float bvj[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int col = n0 + wn + j * 16 + (lane & 15);
+    bvj[j] = (EPI == 1 || EPI == 2) && col < N ? bf2f(bias[col]) : 0.f;
+  }
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = bvj[j];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row_base + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (EPI == 2 || EPI == 3) v = v > 0.f ? v : 0.f;
+        C[(i64)row * N + col] = f2bf(v);
+      }
+    }
+  }
+}
+
 }  // namespace dvgg

@@ -550,6 +550,39 @@ static void launch_gemm_t(const GA& ga, const GB& gb, at::Tensor* outB,
     //    per call (VGG-16 fwd/dgrad) (DVGG_G8_MINCONTIG).
     // (split-K stays on the 128^2 kernel: gemm8's per-element fp32-atomic
     // split-K epilogue measured -5% on VGG-F's fc layers.)
+    // BN=128 deep-tile flavor for N in [128, 256) (DVGG_GEMM8N; off by
+    // default pending a perf A/B — numerics-validated): VGG-16 conv2_x
+    // and ResNet layer2 shapes.
+    if (NFRAG == 4 && N >= 128 && N < 256 && M >= 256 && K >= 128 &&
+        splitk == 1 && env_int("DVGG_GEMM8N", 0) &&
+        ga.kcontig() >= env_int("DVGG_G8_MINCONTIG", 128) &&
+        ceil_div(M, G8_BM) * ceil_div(N, 128) >=
+            env_int("DVGG_G8_MINTILES", 96)) {
+      int tiles8n = ceil_div(M, G8_BM) * ceil_div(N, 128);
+      dim3 g8(tiles8n);
+      switch (epi) {
+        case 0:
+          hipLaunchKernelGGL((gemm8n_kernel<GA, GB, 0>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr,
+                             zp, M, N, K);
+          return;
+        case 1:
+          hipLaunchKernelGGL((gemm8n_kernel<GA, GB, 1>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, bp, zp, M,
+                             N, K);
+          return;
+        case 2:
+          hipLaunchKernelGGL((gemm8n_kernel<GA, GB, 2>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, bp, zp, M,
+                             N, K);
+          return;
+        default:
+          hipLaunchKernelGGL((gemm8n_kernel<GA, GB, 3>), g8,
+                             dim3(G8_THREADS), 0, s, ga, gb, op, nullptr,
+                             zp, M, N, K);
+          return;
+      }
+    }
     int tiles8 = ceil_div(M, G8_BM) * ceil_div(N, G8_BN);
     if (NFRAG == 4 && N >= 256 && M >= 256 && K >= 128 && splitk == 1 &&
         env_int("DVGG_GEMM8", 1) &&

@@ -322,17 +322,23 @@ VARIANT_ENVS = [
     {"DVGG_GEMM8": "1", "DVGG_G8_MINTILES": "0",
      "DVGG_G8_MINCONTIG": "0"},                            # deep tile, forced
     {"DVGG_GEMM8": "0"},                                   # deep tile off
+    {"DVGG_GEMM8N": "1", "DVGG_G8_MINTILES": "0",
+     "DVGG_G8_MINCONTIG": "0"},                            # BN=128 deep tile
 ]
 
 
 @pytest.mark.parametrize("env", VARIANT_ENVS,
-                         ids=["glds2", "glds3", "gemm8-on", "gemm8-off"])
+                         ids=["glds2", "glds3", "gemm8-on", "gemm8-off",
+                              "gemm8n"])
 def test_conv_fwd_dgrad_variants(env, monkeypatch):
     import os
     for k_, v_ in env.items():
         monkeypatch.setenv(k_, v_)
     torch.manual_seed(21)
-    N, H, W, Ci, Co, k, s, p = 2, 13, 13, 256, 256, 3, 1, 1
+    if env.get("DVGG_GEMM8N") == "1":
+        N, H, W, Ci, Co, k, s, p = 2, 16, 16, 128, 128, 3, 1, 1
+    else:
+        N, H, W, Ci, Co, k, s, p = 2, 13, 13, 256, 256, 3, 1, 1
     x = rnd(N, H, W, Ci, scale=0.5)
     w = rnd(k, k, Ci, Co, scale=1.0 / math.sqrt(k * k * Ci))
     b = rnd(Co, scale=0.1)
