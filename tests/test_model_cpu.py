@@ -114,3 +114,19 @@ def test_config_json_roundtrip(tmp_path):
     assert cfg.model == "vgg16" and cfg.batch_size == 7
     # precedence: dataclass default < JSON < explicit CLI
     assert cfg.lr == 0.5
+
+
+def test_vggf_dropout_train_vs_eval():
+    # dropout active in train mode (masks some activations), identity in eval
+    cfg = small_cfg(batch_size=2, dropout=0.5)
+    torch.manual_seed(0)
+    from distributed_vgg_f_amd.models import build_model
+    m = build_model(cfg)
+    x = torch.randn(2, cfg.image_size, cfg.image_size, 3)
+    m.eval()
+    y_eval1 = m(x)
+    y_eval2 = m(x)
+    assert torch.equal(y_eval1, y_eval2)      # eval: deterministic identity
+    m.train()
+    y_tr = m(x)
+    assert not torch.allclose(y_tr, y_eval1)  # train: masked path differs
