@@ -21,8 +21,8 @@ import torch
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-size", type=int, default=256)
     p.add_argument("--model", type=str, default="vggf")
     p.add_argument("--bucket-mb", type=int, default=25)
@@ -64,11 +64,30 @@ def main():
     for _ in range(args.warmup):
         t.train_step()
     barrier_sync()
+    # chunk boundary events on the compute stream: per-chunk ms without
+    # touching the timed region (variance visibility)
+    nchunks = min(args.steps, 10)
+    bounds = sorted({round(i * args.steps / nchunks)
+                     for i in range(nchunks + 1)})
+    evs = []
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
+        if use_gpu and i in bounds:
+            e = torch.cuda.Event(enable_timing=True)
+            e.record()
+            evs.append(e)
         t.train_step()
+    if use_gpu and args.steps in bounds:
+        e = torch.cuda.Event(enable_timing=True)
+        e.record()
+        evs.append(e)
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    chunk_ms = []
+    if use_gpu and len(evs) >= 2:
+        steps_per = [b - a for a, b in zip(bounds, bounds[1:])]
+        chunk_ms = [evs[i].elapsed_time(evs[i + 1]) / steps_per[i]
+                    for i in range(len(evs) - 1)]
 
     # MAX over ranks (RCCL needs the tensor on the GPU)
     et = torch.tensor([elapsed], dtype=torch.float64,
@@ -92,6 +111,9 @@ def main():
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(ms, 3),
+            "ms_per_step_min": round(min(chunk_ms), 3) if chunk_ms else None,
+            "ms_per_step_mean": round(sum(chunk_ms) / len(chunk_ms), 3)
+            if chunk_ms else None,
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,

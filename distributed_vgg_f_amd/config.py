@@ -39,6 +39,9 @@ class Config:
     # training
     batch_size: int = 256              # per-GPU
     lr: float = 0.01
+    scale_lr: bool = False             # linear LR scaling: lr *= world_size
+    #                                    (global batch grows with DP width;
+    #                                     off by default = reference behavior)
     momentum: float = 0.9
     weight_decay: float = 0.0
     train_steps: int = 100
@@ -74,7 +77,7 @@ class Config:
     channels: int = 3
 
     def scaled_lr(self, world_size: int) -> float:
-        return self.lr
+        return self.lr * world_size if self.scale_lr else self.lr
 
     @staticmethod
     def add_args(p: argparse.ArgumentParser) -> None:

@@ -1084,12 +1084,16 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
            OWp);
   auto dyT = cached_zero_buffer(key, tnumel, dy.options())
       .narrow(0, 0, tnumel).view({Cout, N, OH, OWp});
-  at::Tensor dym = (relu && need_dx) ? at::empty_like(dy) : dy;
+  // dym must be masked whenever a downstream consumer reads it: dgrad
+  // (need_dx) or the glds wgrad path (which consumes dym, not dyT)
+  bool glds_path = wgrad_glds_ok(Cin, stride, KW, pad);
+  bool write_dym = relu && (need_dx || glds_path);
+  at::Tensor dym = write_dym ? at::empty_like(dy) : dy;
   auto db32 = need_db ? at::zeros({256, Cout}, dy.options().dtype(at::kFloat))
                       : at::Tensor();
   const u16* yp = relu ? bfp(y) : nullptr;
   float* dbp = need_db ? db32.data_ptr<float>() : nullptr;
-  u16* dymp = (relu && need_dx) ? bfp_mut(dym) : nullptr;
+  u16* dymp = write_dym ? bfp_mut(dym) : nullptr;
   u16* dyTp = bfp_mut(dyT);
   bool vec8 = (Cout % 8 == 0) && (OWp % 8 == 0);
   dim3 pgrid(N * OH, ceil_div(vec8 ? OWp : OW, vec8 ? 64 : 32),
@@ -1105,14 +1109,14 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                          dim3(256), 0, s, bfp(dy), yp, dymp, dyTp, dbp, N,   \
                          OH, OW, Cout, OWp);                                 \
   } while (0)
-  if (relu && need_dx && need_db)      DVGG_PRE_LAUNCH(true, true, true);
-  else if (relu && need_dx)            DVGG_PRE_LAUNCH(true, true, false);
+  if (relu && write_dym && need_db)    DVGG_PRE_LAUNCH(true, true, true);
+  else if (relu && write_dym)          DVGG_PRE_LAUNCH(true, true, false);
   else if (relu && need_db)            DVGG_PRE_LAUNCH(true, false, true);
   else if (relu)                       DVGG_PRE_LAUNCH(true, false, false);
   else if (need_db)                    DVGG_PRE_LAUNCH(false, false, true);
   else                                 DVGG_PRE_LAUNCH(false, false, false);
 #undef DVGG_PRE_LAUNCH
-  // note: when relu && !need_dx, dgrad is skipped so dym (== dy) is unused;
+  // note: when relu && !write_dym, dym (== dy) is unused downstream;
   // the masked values only exist in dyT. wgrad/db see masked dy either way.
 
   // ---- dgrad
@@ -1143,9 +1147,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   // ---- wgrad
   auto dw = dw_out ? *dw_out : at::empty({KH, KW, Cin, Cout}, x.options());
   auto dw2d = dw.view({Kfull, Cout});
-  if (wgrad_glds_ok(Cin, stride, KW, pad)) {
-    // dym is already ReLU-masked when relu (preamble wrote it); dyT from
-    // the preamble is unused on this path
+  if (glds_path) {
+    // dym is ReLU-masked whenever relu (write_dym covers this path); dyT
+    // from the preamble is unused here
     auto dymc = dym.contiguous();
     wgrad_glds(x, dymc, KH, KW, pad, dw2d, Cout);
     at::Tensor db;

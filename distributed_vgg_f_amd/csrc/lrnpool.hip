@@ -245,8 +245,10 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kernel,
                                     int64_t stride, int64_t pl, int64_t pr) {
   check_bf16(x, "x");
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  TORCH_CHECK(H == W, "maxpool_fwd assumes square inputs (H==W), got ",
+              H, "x", W);
   int OH = (H + int(pl) + int(pr) - int(kernel)) / int(stride) + 1;
-  int OW = OH;  // square inputs throughout; generalize if needed
+  int OW = OH;  // square inputs (checked above)
   auto y = at::empty({N, OH, OW, C}, x.options());
   auto idx = at::empty({N, OH, OW, C}, x.options().dtype(at::kByte));
   TORCH_CHECK(C % 8 == 0, "maxpool needs C % 8 == 0");

@@ -31,7 +31,9 @@ static void check_bf16(const at::Tensor& t, const char* name) {
 
 static inline int grid_for(i64 chunks) {
   i64 g = (chunks + 255) / 256;
-  return int(std::min<i64>(g, 2048));
+  // clamp to >=1: numel<8 would otherwise launch 0 blocks and skip the
+  // scalar tail (handled by blockIdx.x==0), leaving the output garbage
+  return int(std::min<i64>(std::max<i64>(g, 1), 2048));
 }
 
 // ------------------------------------------------------------- relu bwd

@@ -131,27 +131,41 @@ class RecordPipeline:
         return t, l
 
     def _producer(self):
-        while not self._stop.is_set():
-            got = False
-            for img, lab in self._iter_epoch(self._epoch):
-                got = True
-                item = self._convert(img, lab)
-                while not self._stop.is_set():
-                    try:
-                        self._q.put(item, timeout=0.25)
-                        break
-                    except queue.Full:
-                        continue
-                if self._stop.is_set():
+        try:
+            while not self._stop.is_set():
+                got = False
+                for img, lab in self._iter_epoch(self._epoch):
+                    got = True
+                    item = self._convert(img, lab)
+                    while not self._stop.is_set():
+                        try:
+                            self._q.put(item, timeout=0.25)
+                            break
+                        except queue.Full:
+                            continue
+                    if self._stop.is_set():
+                        return
+                self._epoch += 1
+                if not got:
+                    raise RuntimeError(
+                        "record shards hold fewer samples than one batch; "
+                        "lower --batch-size")
+        except BaseException as e:
+            # surface errors to the consumer instead of dying silently in
+            # the daemon thread (the training loop would block forever)
+            while not self._stop.is_set():
+                try:
+                    self._q.put(e, timeout=0.25)
                     return
-            self._epoch += 1
-            if not got:   # shards smaller than one batch: loop the data
-                raise RuntimeError("record shards hold fewer samples than "
-                                   "one batch; lower --batch-size")
+                except queue.Full:
+                    continue
 
     # ---------------- consumer (training loop) ----------------
     def next(self) -> Tuple[torch.Tensor, torch.Tensor]:
-        himg, hlab = self._q.get()
+        item = self._q.get()
+        if isinstance(item, BaseException):
+            raise item
+        himg, hlab = item
         if not self._pin:
             return himg, hlab
         with torch.cuda.stream(self.copy_stream):

@@ -37,10 +37,21 @@ for fc in ("fc6", "fc7", "fc8"):
 _COMPAT_INV = {v: k for k, v in NAME_COMPAT.items()}
 
 
-def save(ckpt_dir: str, arena: ArenaOptimizer, step: int) -> str:
+def save(ckpt_dir: str, arena: ArenaOptimizer, step: int,
+         model: Optional[torch.nn.Module] = None) -> str:
     os.makedirs(ckpt_dir, exist_ok=True)
     arrays = {"__step__": np.array(step, dtype=np.int64),
               "__momentum__": arena.m32.cpu().numpy()}
+    if model is not None:
+        # non-trainable state the arena doesn't hold: BN running stats
+        # (needed for resnet50 eval/resume) and dropout RNG call counters
+        # (so resume doesn't replay mask offsets)
+        for name, buf in model.named_buffers():
+            arrays["__buf__" + name] = buf.detach().cpu().numpy()
+        calls = [int(getattr(m, "_calls", 0)) for m in model.modules()
+                 if type(m).__name__ == "Dropout"]
+        if calls:
+            arrays["__dropout_calls__"] = np.array(calls, dtype=np.int64)
     for name, w in arena.named_master():
         key = NAME_COMPAT.get(name, name)
         arrays[key] = w.detach().cpu().numpy()
@@ -65,13 +76,26 @@ def latest(ckpt_dir: str) -> Optional[str]:
     return path if os.path.exists(path) else None
 
 
-def load(path: str, arena: ArenaOptimizer) -> int:
+def load(path: str, arena: ArenaOptimizer,
+         model: Optional[torch.nn.Module] = None) -> int:
     data = np.load(path)
+    bufs = dict(model.named_buffers()) if model is not None else {}
     for key in data.files:
-        if key == "__step__" or key == "__momentum__":
+        if key in ("__step__", "__momentum__", "__dropout_calls__"):
+            continue
+        if key.startswith("__buf__"):
+            buf = bufs.get(key[len("__buf__"):])
+            if buf is not None:
+                buf.copy_(torch.from_numpy(data[key]).to(buf.device,
+                                                         buf.dtype))
             continue
         name = _COMPAT_INV.get(key, key)
         arena.load_master(name, torch.from_numpy(data[key]))
+    if model is not None and "__dropout_calls__" in data.files:
+        drops = [m for m in model.modules()
+                 if type(m).__name__ == "Dropout"]
+        for m, c in zip(drops, data["__dropout_calls__"]):
+            m._calls = int(c)
     if "__momentum__" in data.files:
         m = torch.from_numpy(data["__momentum__"])
         if m.numel() == arena.m32.numel():
@@ -80,8 +104,9 @@ def load(path: str, arena: ArenaOptimizer) -> int:
     return int(data["__step__"]) if "__step__" in data.files else 0
 
 
-def restore_latest(ckpt_dir: str, arena: ArenaOptimizer) -> int:
+def restore_latest(ckpt_dir: str, arena: ArenaOptimizer,
+                   model: Optional[torch.nn.Module] = None) -> int:
     path = latest(ckpt_dir)
     if path is None:
         return 0
-    return load(path, arena)
+    return load(path, arena, model)

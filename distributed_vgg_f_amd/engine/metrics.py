@@ -43,10 +43,13 @@ class PhaseTimers:
     """hipEvent-based per-phase timers (h2d/fwd/bwd/ar/opt), GPU only.
 
     Cheap enough to keep on; events are recorded per window, elapsed read
-    at the window boundary after a sync.
+    at the window boundary after a sync. `ar` measures EXPOSED all-reduce:
+    the compute-stream stall waiting on the bucket reductions after
+    backward ends (RCCL runs on its own streams; with good overlap this
+    is ~0). `h2d` is the compute-stream wait on the copy-stream event.
     """
 
-    PHASES = ("fwd", "bwd", "opt")
+    PHASES = ("h2d", "fwd", "bwd", "ar", "opt")
 
     def __init__(self, device, enabled: bool = True):
         import torch

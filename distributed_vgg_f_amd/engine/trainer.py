@@ -44,6 +44,9 @@ class Trainer:
         self.model = build_model(cfg)        # broadcast still run for safety
         self.model.to(self.device)           # buffers (BN stats); params get
         self.model.train()                   # re-homed into arenas below
+        for m in self.model.modules():       # decorrelate dropout across DP
+            if type(m).__name__ == "Dropout":  # ranks (each rank sees its
+                m.seed += self.ctx.rank * 9973  # own shard of the batch)
 
         compute_dtype = torch.bfloat16 if (cfg.precision == "bf16" and
                                            self.device.type == "cuda") \
@@ -52,8 +55,10 @@ class Trainer:
                                         self.device.type == "cuda") \
             else torch.float32
         self.arena = ArenaOptimizer(self.model, self.device, compute_dtype,
-                                    grad_dtype, cfg.lr, cfg.momentum,
-                                    cfg.weight_decay, cfg.bucket_mb)
+                                    grad_dtype,
+                                    cfg.scaled_lr(self.ctx.world_size),
+                                    cfg.momentum, cfg.weight_decay,
+                                    cfg.bucket_mb)
         broadcast_init(self.arena, self.ctx)
         self.reducer = GradReducer(self.arena, self.ctx, cfg.overlap_comm)
         if cfg.data == "records":
@@ -73,7 +78,8 @@ class Trainer:
         self.global_batch = cfg.batch_size * self.ctx.world_size
 
         if cfg.resume and cfg.checkpoint_dir:
-            self.step_i = ckpt.restore_latest(cfg.checkpoint_dir, self.arena)
+            self.step_i = ckpt.restore_latest(cfg.checkpoint_dir, self.arena,
+                                              self.model)
             if self.ctx.is_chief and self.step_i:
                 self.metrics.log(event="resumed", step=self.step_i)
 
@@ -112,6 +118,11 @@ class Trainer:
             return False
         if not isinstance(self.pipeline, SyntheticPipeline):
             return False   # capture pins fixed host buffers (synthetic pool)
+        if any(getattr(m, "p", 0) > 0 for m in self.model.modules()
+               if type(m).__name__ == "Dropout"):
+            # capture bakes the (seed, offset) kernel args: every replay
+            # would reuse the identical dropout mask — stay eager
+            return False
         try:
             self._gimg_h = [self.pipeline.images[0], self.pipeline.images[1]]
             self._glab_h = [self.pipeline.labels[0], self.pipeline.labels[1]]
@@ -163,9 +174,11 @@ class Trainer:
             self._graphs[i].replay()
             self.step_i += 1
             return self._glosses[i]
-        images, labels = self.pipeline.next()
-        self.reducer.prepare_step()
         tm = self.timers
+        tm.start("h2d")
+        images, labels = self.pipeline.next()
+        tm.stop("h2d")
+        self.reducer.prepare_step()
         tm.start("fwd")
         logits = self.model(images)
         loss = ops.softmax_xent(logits, labels).sum()
@@ -176,8 +189,10 @@ class Trainer:
             # AccumulateGrad stores them back as .grad (no adds, no zeroing)
             self.arena.detach_grads()
         loss.backward()
-        self.reducer.finish()
         tm.stop("bwd")
+        tm.start("ar")   # exposed comm: stream stall on bucket reductions
+        self.reducer.finish()
+        tm.stop("ar")
         # loss summed per sample; SUM all-reduce over ranks => scale by 1/GB
         tm.start("opt")
         self.arena.step(grad_scale=1.0 / self.global_batch)
@@ -205,15 +220,16 @@ class Trainer:
                                  loss=float(last_loss) / cfg.batch_size,
                                  images_per_sec=ips,
                                  ms_per_step=1000.0 * dt / window_steps,
-                                 lr=cfg.lr, **self.timers.collect())
+                                 lr=self.arena.lr, **self.timers.collect())
                 window_t0 = time.time()
                 window_steps = 0
             if (cfg.checkpoint_every and cfg.checkpoint_dir and
                     self.step_i % cfg.checkpoint_every == 0 and
                     self.ctx.is_chief):
-                ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i)
+                ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i,
+                          self.model)
         if cfg.checkpoint_dir and self.ctx.is_chief:
-            ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i)
+            ckpt.save(cfg.checkpoint_dir, self.arena, self.step_i, self.model)
 
     # ------------------------------------------------------------------ #
     def _eval_pipeline(self):

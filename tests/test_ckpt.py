@@ -55,6 +55,45 @@ def test_latest_pointer_tracks_newest(tmp_path):
     assert ckpt.latest(str(tmp_path)).endswith("model.ckpt-9.npz")
 
 
+def test_bn_buffers_and_dropout_calls_roundtrip(tmp_path):
+    """BN running stats + dropout RNG counters must
+    survive save/restore (resnet50 eval after resume was wrong without)."""
+    cfg = Config(model="resnet50", image_size=64, seed=3)
+    torch.manual_seed(cfg.seed)
+    model = build_model(cfg)
+    arena = ArenaOptimizer(model, torch.device("cpu"))
+    # perturb the BN running stats away from init
+    for name, buf in model.named_buffers():
+        buf.add_(1.5)
+    bufs_before = {n: b.clone() for n, b in model.named_buffers()}
+    assert bufs_before, "resnet50 should have BN buffers"
+    ckpt.save(str(tmp_path), arena, step=5, model=model)
+    for _, buf in model.named_buffers():
+        buf.zero_()
+    step = ckpt.restore_latest(str(tmp_path), arena, model=model)
+    assert step == 5
+    for n, b in model.named_buffers():
+        assert torch.equal(b, bufs_before[n]), n
+
+    # dropout call counters (vggf has drop6/drop7)
+    cfg = Config(model="vggf", image_size=64, dropout=0.5, seed=3)
+    torch.manual_seed(cfg.seed)
+    model = build_model(cfg)
+    arena = ArenaOptimizer(model, torch.device("cpu"))
+    model.drop6._calls, model.drop7._calls = 17, 19
+    ckpt.save(str(tmp_path), arena, step=6, model=model)
+    model.drop6._calls = model.drop7._calls = 0
+    ckpt.restore_latest(str(tmp_path), arena, model=model)
+    assert (model.drop6._calls, model.drop7._calls) == (17, 19)
+
+
+def test_scaled_lr():
+    cfg = Config(lr=0.01)
+    assert cfg.scaled_lr(8) == 0.01          # default: no scaling
+    cfg = Config(lr=0.01, scale_lr=True)
+    assert abs(cfg.scaled_lr(8) - 0.08) < 1e-12
+
+
 def test_resume_path_via_trainer(tmp_path):
     from distributed_vgg_f_amd.engine.trainer import Trainer
     cfg = Config(model="vggf", image_size=64, batch_size=2, train_steps=2,

@@ -108,3 +108,15 @@ def test_eval_uses_val_split(tmp_path):
     assert t._eval_pipeline() is t._val_pipeline   # val shards were picked
     t.pipeline.close()
     t._val_pipeline.close()
+
+
+def test_producer_error_surfaces_in_next(tmp_path):
+    """a producer-thread error (shards smaller than one batch)
+    must raise in next(), not hang the training loop forever."""
+    import pytest
+    _mkdata(tmp_path, n=4, size=8, shard=4)       # 4 samples < batch 16
+    p = RecordPipeline(str(tmp_path), batch_size=16,
+                       device=torch.device("cpu"), dtype=torch.float32)
+    with pytest.raises(RuntimeError, match="fewer samples"):
+        p.next()
+    p.close()
