@@ -105,10 +105,9 @@ __global__ void transpose2d8_kernel(const u16* __restrict__ in,
   }
 }
 
-static at::Tensor transpose2d(const at::Tensor& in, int Rp = -1) {
+static void transpose2d_into(const at::Tensor& in, at::Tensor& out) {
   int R = in.size(0), C = in.numel() / in.size(0);
-  if (Rp < 0) Rp = R;
-  auto out = at::zeros({C, Rp}, in.options());
+  int Rp = out.size(1);
   if (C % 8 == 0 && Rp % 8 == 0) {
     dim3 grid(ceil_div(C, 64), ceil_div(Rp, 64));
     hipLaunchKernelGGL(transpose2d8_kernel, grid, dim3(256), 0, cur_stream(),
@@ -118,7 +117,56 @@ static at::Tensor transpose2d(const at::Tensor& in, int Rp = -1) {
     hipLaunchKernelGGL(transpose2d_kernel, grid, dim3(256), 0, cur_stream(),
                        bfp(in), bfp_mut(out), R, C, Rp);
   }
+}
+
+static at::Tensor transpose2d(const at::Tensor& in, int Rp = -1) {
+  int R = in.size(0), C = in.numel() / in.size(0);
+  if (Rp < 0) Rp = R;
+  auto out = at::zeros({C, Rp}, in.options());
+  transpose2d_into(in, out);
   return out;
+}
+
+// --------------- persistent weight-transform cache (round-2 item 4)
+// Weights mutate exactly once per step (sgd_step bumps the generation; the
+// engine bumps via bump_weight_gen after checkpoint-load/broadcast). The
+// per-forward transpose2d(w), the dgrad permute wD and the Cin<8 channel
+// pad w8 then refresh once per generation instead of once per call
+// (~8 transpose launches + their zero-fills per VGG-F step). Keyed by
+// data_ptr and shape; OPT-IN per call (cache_w=true only for arena-homed
+// weights, engine/arena.py) so ad-hoc op calls (numerics tests build fresh
+// weights constantly, recycling allocator pointers) can never see a stale
+// hit, and bypassed during hipGraph capture (a captured step must contain
+// the refresh kernels, since replays re-run sgd_step).
+i64 g_weight_gen = 0;
+void bump_weight_gen() { ++g_weight_gen; }
+
+static bool stream_capturing() {
+  hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+  (void)hipStreamIsCapturing(cur_stream(), &st);
+  return st != hipStreamCaptureStatusNone;
+}
+
+struct WTEntry { i64 gen; std::vector<i64> src_sizes; at::Tensor t; };
+
+// kind 0: wT [Cout][K]; kind 1: wD [Cin][KhKw*Cout]; kind 2: w8 pad
+template <class F>
+static at::Tensor cached_weight_form(int kind, bool on, const at::Tensor& w,
+                                     F&& make) {
+  if (!on || stream_capturing()) return make(c10::nullopt);
+  static std::unordered_map<const void*, WTEntry> cache[3];
+  auto& c = cache[kind];
+  std::vector<i64> sizes(w.sizes().begin(), w.sizes().end());
+  auto it = c.find(w.data_ptr());
+  if (it != c.end() && it->second.src_sizes == sizes) {
+    if (it->second.gen == g_weight_gen) return it->second.t;
+    it->second.t = make(c10::optional<at::Tensor>(it->second.t));
+    it->second.gen = g_weight_gen;
+    return it->second.t;
+  }
+  auto t = make(c10::nullopt);
+  c[w.data_ptr()] = WTEntry{g_weight_gen, std::move(sizes), t};
+  return t;
 }
 
 // --------------------------------------------- NHWC -> C,N,Hp,Wp (padded)
@@ -772,12 +820,16 @@ static int epi_of(bool has_bias, bool relu) {
 
 // ---------------------------------------------------------------- linear
 at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
-                      const c10::optional<at::Tensor>& bias, bool relu) {
+                      const c10::optional<at::Tensor>& bias, bool relu,
+                      bool cache_w) {
   check_bf16(x, "x"); check_bf16(w, "w");
   int M = x.size(0), K = x.size(1), N = w.size(1);
   TORCH_CHECK(w.size(0) == K && K % 8 == 0 && N % 8 == 0, "bad fc shape");
   auto y = at::empty({M, N}, x.options());
-  auto wT = transpose2d(w);  // [N][K]
+  auto wT = cached_weight_form(0, cache_w, w, [&](c10::optional<at::Tensor> prev) {
+    if (prev) { transpose2d_into(w, *prev); return *prev; }
+    return transpose2d(w);
+  });  // [N][K]
   PlainGather ga{bfp(x), M, K, K};
   PlainGather gb{bfp(wT), N, K, K};
   int tiles = ceil_div(M, BM) * ceil_div(N, BN);
@@ -910,7 +962,7 @@ static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
 
 at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                       const c10::optional<at::Tensor>& bias, int64_t stride_,
-                      int64_t pad_, bool relu) {
+                      int64_t pad_, bool relu, bool cache_w) {
   check_bf16(x, "x"); check_bf16(w, "w");
   int stride = int(stride_), pad = int(pad_);
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
@@ -921,7 +973,8 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int M = N * OH * OW, Kfull = KH * KW * Cin;
   if (KH == 1 && KW == 1 && stride == 1 && pad == 0 && Cin % 8 == 0) {
     // 1x1/s1 conv IS the FC GEMM
-    auto y2 = linear_fwd(x.view({M, Cin}), w.view({Cin, Cout}), bias, relu);
+    auto y2 = linear_fwd(x.view({M, Cin}), w.view({Cin, Cout}), bias,
+                         relu, cache_w);
     return y2.view({N, OH, OW, Cout});
   }
   auto y = at::empty({N, OH, OW, Cout}, x.options());
@@ -932,12 +985,19 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
     // first layers (conv1 Cin=3, ResNet stem): pad channels to 8 and take
     // the vector-gather glds path; zero channels contribute nothing
     auto x8 = pad_channels8(x);
-    auto w8 = at::zeros({KH, KW, 8, Cout}, w.options());
-    w8.narrow(2, 0, Cin).copy_(w);
-    return conv2d_fwd(x8, w8, bias, stride_, pad_, relu);
+    auto w8 = cached_weight_form(2, cache_w, w, [&](c10::optional<at::Tensor> prev) {
+      auto t = prev ? *prev : at::zeros({KH, KW, 8, Cout}, w.options());
+      t.narrow(2, 0, Cin).copy_(w);
+      return t;
+    });
+    return conv2d_fwd(x8, w8, bias, stride_, pad_, relu, cache_w);
   }
   TORCH_CHECK(Cin % 8 == 0, "conv fwd needs Cin % 8 == 0 (or Cin < 8)");
-  auto wT = transpose2d(w.view({Kfull, Cout}));  // [Cout][K]
+  auto w2d = w.view({Kfull, Cout});
+  auto wT = cached_weight_form(0, cache_w, w, [&](c10::optional<at::Tensor> prev) {
+    if (prev) { transpose2d_into(w2d, *prev); return *prev; }
+    return transpose2d(w2d);
+  });  // [Cout][K]
   ConvFwdGather ga{bfp(x), M, H, W, Cin, stride, pad,
                    MDiv(u32(OH * OW)), MDiv(u32(OW)), MDiv(u32(Cin)),
                    MDiv(u32(KW))};
@@ -956,7 +1016,8 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
 }
 
 at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
-                        int64_t W_, int64_t stride_, int64_t pad_) {
+                        int64_t W_, int64_t stride_, int64_t pad_,
+                        bool cache_w) {
   check_bf16(dy, "dy"); check_bf16(w, "w");
   int H = int(H_), W = int(W_), stride = int(stride_), pad = int(pad_);
   int N = dy.size(0), OH = dy.size(1), OW = dy.size(2), Cout = dy.size(3);
@@ -967,8 +1028,11 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   int M = N * H * W, K = KH * KW * Cout;
   auto dx = at::empty({N, H, W, Cin}, dy.options());
   // Bt[ci][(kh,kw,co)] = w[kh][kw][ci][co]: permute HWIO -> [Cin][KhKw*Cout]
-  auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
-      .view({Cin, KH * KW * Cout});
+  auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
+    auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
+    if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
+    return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+  });
   TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
   ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad,
                      MDiv(u32(H * W)), MDiv(u32(W)), MDiv(u32(Cout)),
@@ -1051,7 +1115,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                                    int64_t stride_, int64_t pad_, bool relu,
                                    bool need_dx, bool need_db,
                                    const c10::optional<at::Tensor>& dw_out,
-                                   const c10::optional<at::Tensor>& db_out) {
+                                   const c10::optional<at::Tensor>& db_out,
+                                   bool cache_w) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   int stride = int(stride_), pad = int(pad_);
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
@@ -1125,8 +1190,11 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     TORCH_CHECK(Cout % 8 == 0 && Cin % 8 == 0, "dgrad channel alignment");
     int K = KH * KW * Cout;
     dx = at::empty({N, H, W, Cin}, dy.options());
-    auto wD = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2}).contiguous()
-        .view({Cin, KH * KW * Cout});
+    auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
+      auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
+      if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
+      return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+    });
     TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
     int Mx = N * H * W;
     ConvDgradGather ga{bfp(dym), Mx, H, W, Cout, OH, OW, stride, pad,

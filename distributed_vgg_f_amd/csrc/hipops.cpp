@@ -7,22 +7,23 @@ namespace dvgg {
 
 // gemm_ops.hip
 at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&,
-                      const c10::optional<at::Tensor>&, bool);
+                      const c10::optional<at::Tensor>&, bool, bool);
 at::Tensor linear_dgrad(const at::Tensor&, const at::Tensor&);
 at::Tensor linear_wgrad(const at::Tensor&, const at::Tensor&,
                         const c10::optional<at::Tensor>&);
 at::Tensor conv2d_fwd(const at::Tensor&, const at::Tensor&,
                       const c10::optional<at::Tensor>&, int64_t, int64_t,
-                      bool);
+                      bool, bool);
 at::Tensor conv2d_dgrad(const at::Tensor&, const at::Tensor&, int64_t,
-                        int64_t, int64_t, int64_t);
+                        int64_t, int64_t, int64_t, bool);
 at::Tensor conv2d_wgrad(const at::Tensor&, const at::Tensor&, int64_t,
                         int64_t, int64_t, int64_t);
 std::vector<at::Tensor> conv2d_bwd(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
                                    int64_t, int64_t, bool, bool, bool,
                                    const c10::optional<at::Tensor>&,
-                                   const c10::optional<at::Tensor>&);
+                                   const c10::optional<at::Tensor>&, bool);
+void bump_weight_gen();
 
 // mem.hip
 at::Tensor relu_bwd(const at::Tensor&, const at::Tensor&);
@@ -60,22 +61,29 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for distributed_vgg_f_amd";
-  m.def("linear_fwd", &dvgg::linear_fwd);
+  m.def("linear_fwd", &dvgg::linear_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias"), py::arg("relu"), py::arg("cache_w") = false);
   m.def("linear_dgrad", &dvgg::linear_dgrad);
   m.def("linear_wgrad", &dvgg::linear_wgrad, py::arg("x"), py::arg("dy"),
         py::arg("dw_out") = c10::nullopt);
-  m.def("conv2d_fwd", &dvgg::conv2d_fwd);
-  m.def("conv2d_dgrad", &dvgg::conv2d_dgrad);
+  m.def("conv2d_fwd", &dvgg::conv2d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias"), py::arg("stride"), py::arg("pad"), py::arg("relu"),
+        py::arg("cache_w") = false);
+  m.def("conv2d_dgrad", &dvgg::conv2d_dgrad, py::arg("dy"), py::arg("w"),
+        py::arg("H"), py::arg("W"), py::arg("stride"), py::arg("pad"),
+        py::arg("cache_w") = false);
   m.def("conv2d_wgrad", &dvgg::conv2d_wgrad);
   m.def("conv2d_bwd", &dvgg::conv2d_bwd, py::arg("dy"), py::arg("y"),
         py::arg("x"), py::arg("w"), py::arg("stride"), py::arg("pad"),
         py::arg("relu"), py::arg("need_dx"), py::arg("need_db"),
-        py::arg("dw_out") = c10::nullopt, py::arg("db_out") = c10::nullopt);
+        py::arg("dw_out") = c10::nullopt, py::arg("db_out") = c10::nullopt,
+        py::arg("cache_w") = false);
   m.def("relu_bwd", &dvgg::relu_bwd);
   m.def("add_relu_fwd", &dvgg::add_relu_fwd);
   m.def("bias_grad", &dvgg::bias_grad, py::arg("dy"),
         py::arg("db_out") = c10::nullopt);
   m.def("sgd_step", &dvgg::sgd_step);
+  m.def("bump_weight_gen", &dvgg::bump_weight_gen);
   m.def("dropout_fwd", &dvgg::dropout_fwd);
   m.def("dropout_bwd", &dvgg::dropout_bwd);
   m.def("lrn_fwd", &dvgg::lrn_fwd);

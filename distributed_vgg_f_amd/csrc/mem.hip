@@ -166,6 +166,8 @@ __global__ void sgd_kernel(float* __restrict__ w32, u16* __restrict__ wc,
   }
 }
 
+void bump_weight_gen();  // gemm_ops.hip: weight-transform cache generation
+
 void sgd_step(at::Tensor& w32, at::Tensor& wc, const at::Tensor& g,
               at::Tensor& m32, double lr, double mu, double wd, double scale) {
   TORCH_CHECK(w32.is_cuda() && w32.scalar_type() == at::kFloat);
@@ -176,6 +178,7 @@ void sgd_step(at::Tensor& w32, at::Tensor& wc, const at::Tensor& g,
                      cur_stream(), w32.data_ptr<float>(), bfp_mut(wc), bfp(g),
                      m32.data_ptr<float>(), n, float(lr), float(mu),
                      float(wd), float(scale));
+  bump_weight_gen();  // wc changed: cached wT/wD/w8 refresh on next forward
 }
 
 // ----------------------------------------------------------- dropout K12

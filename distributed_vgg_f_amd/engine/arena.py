@@ -106,6 +106,15 @@ class ArenaOptimizer:
             ofs += _aligned(n)
         if compute_dtype != torch.float32:
             self.wc.copy_(self.w32.to(compute_dtype))
+        if device.type == "cuda":
+            # persistent wT/wD cache: arena-homed weights mutate only in
+            # sgd_step (which bumps the generation itself) and via
+            # sync_compute_weights below; mark them cacheable so the GPU
+            # ops reuse their transposed/permuted forms across steps
+            from ..ops import native
+            for s in self.slots:
+                s.param._dvgg_wcache = True
+            native.ext().bump_weight_gen()
 
         # buckets: groups of whole params up to bucket_mb (grad-dtype MB)
         self.buckets: List[Bucket] = []
@@ -161,3 +170,6 @@ class ArenaOptimizer:
     def sync_compute_weights(self) -> None:
         if self.wc.data_ptr() != self.w32.data_ptr():
             self.wc.copy_(self.w32.to(self.compute_dtype))
+        if self.device.type == "cuda":
+            from ..ops import native
+            native.ext().bump_weight_gen()

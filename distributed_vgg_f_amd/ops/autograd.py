@@ -31,10 +31,12 @@ class ConvFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, bias, stride: int, pad: int, relu: bool):
-        y = _ext().conv2d_fwd(x, w, bias, stride, pad, relu)
+        cw = getattr(w, "_dvgg_wcache", False)  # arena-homed: persistent wT
+        y = _ext().conv2d_fwd(x, w, bias, stride, pad, relu, cw)
         ctx.save_for_backward(x, w, y if relu else torch.empty(0, device=x.device))
         ctx.stride, ctx.pad, ctx.relu = stride, pad, relu
         ctx.has_bias = bias is not None
+        ctx.cw = cw
         # direct-grad: arena views attached by ArenaOptimizer (engine/arena.py)
         ctx.gw = getattr(w, "_dvgg_gbuf", None)
         ctx.gb = getattr(bias, "_dvgg_gbuf", None) if bias is not None else None
@@ -52,7 +54,7 @@ class ConvFunction(torch.autograd.Function):
         # the params carry _dvgg_gbuf views (skips AccumulateGrad adds).
         dx, dw, db = _ext().conv2d_bwd(dy.contiguous(), y, x, w, ctx.stride,
                                        ctx.pad, ctx.relu, need_dx, need_db,
-                                       ctx.gw, ctx.gb)
+                                       ctx.gw, ctx.gb, ctx.cw)
         return (dx if need_dx else None, dw, db if need_db else None,
                 None, None, None)
 
@@ -62,7 +64,8 @@ class LinearFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, bias, relu: bool):
-        y = _ext().linear_fwd(x, w, bias, relu)
+        y = _ext().linear_fwd(x, w, bias, relu,
+                              getattr(w, "_dvgg_wcache", False))
         ctx.save_for_backward(x, w, y if relu else torch.empty(0, device=x.device))
         ctx.relu = relu
         ctx.has_bias = bias is not None
