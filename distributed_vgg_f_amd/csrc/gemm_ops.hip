@@ -9,6 +9,7 @@
 #include "common.h"
 #include "gemm_tile.h"
 #include "gemm8.h"
+#include "wgrad_fused.h"
 
 namespace dvgg {
 
@@ -914,6 +915,56 @@ static at::Tensor pad_channels8(const at::Tensor& x) {
   return out;
 }
 
+// ---------------- tap-fused 3x3 wgrad (wgrad_fused.h) -----------------
+static bool wgrad_fused_ok(int KH, int KW, int stride, int Cin, int Cout) {
+  return KH == 3 && KW == 3 && stride == 1 && Cin % 64 == 0 &&
+         Cout % 64 == 0 && env_int("DVGG_WGRAD_FUSED", 1);
+}
+
+// OW -> the fused kernel's padded column extent: multiple of 8 with
+// OWp2 % 32 == 0 or 32 % OWp2 == 0 (so a 32-k tile is whole rows or a
+// whole 32-column segment). 56 -> 64, 112 -> 128 (14% zero-k waste).
+static int owp2_of(int OW) {
+  int p = ceil8(OW);
+  if (p % 32 == 0 || 32 % p == 0) return p;
+  return (p + 31) & ~31;
+}
+
+static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
+                        at::Tensor& c32, int Cin, int Cout, int N, int Hp,
+                        int Wp, int OH, int OWp2) {
+  int CT = std::min(OWp2, 32), R = 32 / CT;
+  int SEG = CT == 32 ? OWp2 / 32 : 1;
+  int OHK = (OH + R - 1) / R * R;
+  int TPI = CT == 32 ? OH * SEG : OHK / R;
+  i64 tiles_k = (i64)N * TPI;
+  int wavesn = (Cout % 128 == 0) ? 4 : 2;
+  int tiles = (Cin / 64) * (Cout / (32 * wavesn));
+  int target = env_int("DVGG_WGF_BLOCKS", 768) * sk_scale() / 100;
+  int sk = std::max(1, std::min<int>(int(tiles_k),
+                                     target / std::max(1, tiles)));
+  i64 span = (tiles_k + sk - 1) / sk;
+  sk = int((tiles_k + span - 1) / span);   // non-empty slices only
+  const u16* zp = zero_page(xT.options());
+  dim3 grid(tiles, sk);
+  auto s = cur_stream();
+#define WGF_LAUNCH(CTV, WNV)                                                 \
+  hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WNV>), grid,                  \
+                     dim3(2 * WNV * 64), 0, s, bfp(xT), bfp(dyT),            \
+                     c32.data_ptr<float>(), zp, Cin, Cout, N, Hp, Wp, OH,    \
+                     OWp2, int(tiles_k), sk)
+  if (wavesn == 4) {
+    if (CT == 32) WGF_LAUNCH(32, 4);
+    else if (CT == 16) WGF_LAUNCH(16, 4);
+    else WGF_LAUNCH(8, 4);
+  } else {
+    if (CT == 32) WGF_LAUNCH(32, 2);
+    else if (CT == 16) WGF_LAUNCH(16, 2);
+    else WGF_LAUNCH(8, 2);
+  }
+#undef WGF_LAUNCH
+}
+
 // glds wgrad (stride==1, Cin %% 128 == 0): substituted-reduction implicit
 // GEMM with both operands staged by global_load_lds (see gemm_tile.h).
 static bool wgrad_glds_ok(int Cin, int stride, int KW, int pad) {
@@ -1068,9 +1119,19 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   // Padding Cin<8 to 8 was tried and measured 3.4x SLOWER for wgrad
   // (zero-channel waste x scalar stride-4 staging); the direct NHWC
   // gather below stays for Cin < 8.
-  int OWp = ceil8(OW);
+  bool wfused = wgrad_fused_ok(KH, KW, stride, Cin, Cout);
+  int OWp = wfused ? owp2_of(OW) : ceil8(OW);
   int Kred = N * OH * OWp;
   auto dyT = nhwc2cnhw_pad(dy, OH, OWp, 0, 0, 'd');  // [Cout][N][OH][OWp]
+  if (wfused) {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
+    auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+    wgrad_fused(xT, dyT, c32, Cin, Cout, N, Hp, Wp, OH, OWp);
+    run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+    return dw;
+  }
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
@@ -1122,7 +1183,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
   int KH = w.size(0), KW = w.size(1), Cout = w.size(3);
   int OH = dy.size(1), OW = dy.size(2);
-  int OWp = ceil8(OW);
+  bool wfused = wgrad_fused_ok(KH, KW, stride, Cin, Cout);
+  int OWp = wfused ? owp2_of(OW) : ceil8(OW);
   int M = N * OH * OW, Kfull = KH * KW * Cin;
   auto s = cur_stream();
 
@@ -1229,11 +1291,26 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  if (wfused) {
+    int Hp = H + 2 * pad;
+    int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
+    auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
+    wgrad_fused(xT, dyT, c32, Cin, Cout, N, Hp, Wp, OH, OWp);
+    run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+    at::Tensor db;
+    if (need_db) {
+      db = db_out ? *db_out : at::empty({Cout}, dy.options());
+      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                         dim3(256), 0, s, db32.data_ptr<float>(),
+                         bfp_mut(db), Cout);
+    }
+    return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
+  }
   int Kred = N * OH * OWp;
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
                   ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);

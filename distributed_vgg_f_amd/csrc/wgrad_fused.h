@@ -1,0 +1,254 @@
+// Tap-fused 3x3 stride-1 wgrad (SURVEY.md K3, round-2 structural fix).
+//
+// The round-1 wgrad lowered dw[(kh,kw,ci)][co] as a GEMM with
+// M = KH*KW*Cin, so the dy B-panel was re-read M/BM times (7-9x; PMC:
+// 21% MFMA / 63% wait, 23% of the VGG-F step). Here the GEMM M-dim is
+// ci ONLY:
+//   * kh enters as an LDS ROW OFFSET into a staged x slab that covers the
+//     k-tile's output rows plus KH-1 halo rows — all three kh taps read
+//     the same slab;
+//   * kw enters as a REGISTER WINDOW: per (ci-frag, kh) the kernel reads
+//     12 contiguous bf16 (one ds_read_b128 + one ds_read_b64) and derives
+//     the kw=0/2 fragments by dword re-selection (free) and kw=1 by four
+//     v_alignbit (VALU, hidden under MFMA);
+//   * each wave holds 9 accumulator sets (one per tap), 32ci x 32co each.
+// dy is then read Cin/64 times per co-tile instead of KH*KW*Cin/256 —
+// conv3: 245 MB -> 109 MB of B traffic; per-k-tile arithmetic intensity
+// 87 -> ~230 FLOP/B.
+//
+// Geometry: 2(m) x WAVES_N waves, 64 ci x (32*WAVES_N) co block, BK=32
+// k-positions per tile where k = (n, oh, ow) over the OWp2-padded output
+// (OWp2 = OW padded to a multiple of 8 with OWp2 % 32 == 0 or
+// 32 % OWp2 == 0, so a k-tile is R = 32/CT full rows (CT = OWp2 < 32) or
+// a 32-column segment of one row (CT = 32)).
+//
+// LDS (one __shared__ arena, glds-staged, double-buffered):
+//   A slab [rr = R+2][ci 64][CTW = CT+8]: CTW is an odd multiple of 4
+//   dwords, so the 16-lane ds_read_b128 fragment groups (ci varies per
+//   lane) land on 16 distinct bank quads — conflict-free without an XOR
+//   swizzle, while the glds destination stays lane-linear.
+//   B tile [kofs 4][co][8]: fragment k-elems contiguous per co; 4-dword
+//   co stride -> conflict-free b128 reads.
+#pragma once
+
+#include "common.h"
+#include "gemm_tile.h"   // zero8, as_bf16x8, xcd_swizzle
+
+namespace dvgg {
+
+// CT = k-tile column extent; BMC = 64 ci rows; BNC = 32*WAVES_N co cols.
+template <int CT, int WAVES_N>
+__global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
+    const u16* __restrict__ xT,    // [Cin][N][Hp][Wp], Wp = OWp2 + 8
+    const u16* __restrict__ dyT,   // [Cout][N][OH][OWp2], zero col pad
+    float* __restrict__ c32,       // [9*Cin][Cout] pre-zeroed
+    const u16* __restrict__ zpage, int Cin, int Cout, int N, int Hp, int Wp,
+    int OH, int OWp2, int tiles_k, int slices) {
+  constexpr int KH = 3, KW = 3;
+  constexpr int NW = 2 * WAVES_N;          // waves per block
+  constexpr int THREADS_W = NW * 64;
+  constexpr int BMC = 64, BNC = 32 * WAVES_N;
+  constexpr int R = 32 / CT;               // output rows per k-tile
+  constexpr int RH = R + KH - 1;           // staged slab rows
+  constexpr int CTW = CT + 8;              // slab cols (halo + align slack)
+  constexpr int CPR = CTW / 8;             // 16B chunks per slab row
+  constexpr int ACH_RAW = RH * BMC * CPR;  // real A chunks
+  constexpr int ACH = (ACH_RAW + THREADS_W - 1) / THREADS_W * THREADS_W;
+  constexpr int ASLAB = ACH * 8;           // elems (incl. slack)
+  constexpr int BCH = 4 * BNC;             // B chunks (one per k-frag x co)
+  constexpr int BCHP = (BCH + THREADS_W - 1) / THREADS_W * THREADS_W;
+  constexpr int BSLAB = BCHP * 8;
+  // 3-buffer ring + counted vmcnt (cf. gemm_glds3_kernel): this kernel
+  // runs 1 block/CU (195-197 VGPR), so a per-tile vmcnt(0) drain has no
+  // occupancy to hide behind — stages are issued two tiles ahead and the
+  // per-tile wait leaves the newest stage in flight.
+  __shared__ u16 smem[3][ASLAB + BSLAB];
+
+  const int tiles_m = Cin / BMC, tiles_n = Cout / BNC;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int ci0 = int(wg) / tiles_n * BMC;
+  const int co0 = int(wg) % tiles_n * BNC;
+  // split-K slice over whole k-tiles
+  const int span = (tiles_k + slices - 1) / slices;
+  const int t_lo = blockIdx.y * span;
+  const int t_hi = min(tiles_k, t_lo + span);
+  if (t_lo >= t_hi) return;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave / WAVES_N) * 32;    // ci sub-tile base
+  const int wn = (wave % WAVES_N) * 32;    // co sub-tile base
+  const int l15 = lane & 15;
+  const int kfrag = (lane >> 4) * 8;       // k offset of this lane's frags
+
+  // tile decomposition over the padded per-image k-space
+  const int SEG = (CT == 32) ? (OWp2 / 32) : 1;     // col segments per row
+  const int OHK = (OH + R - 1) / R * R;
+  const int TPI = (CT == 32) ? OH * SEG : OHK / R;  // k-tiles per image
+
+  // --- per-slot staging descriptors. glds dest must be WAVE-UNIFORM
+  // (hardware adds lane*16 itself); the per-lane part lives entirely in
+  // the SOURCE address. chunk c = slot base (wave-uniform) + lane.
+  constexpr int AGI = ACH / THREADS_W;     // A glds per thread
+  constexpr int BGI = BCHP / THREADS_W;
+  int a_rr[AGI]; i64 a_cibase[AGI]; int a_col[AGI]; bool a_ok[AGI];
+  u16* a_dst[AGI];
+#pragma unroll
+  for (int g = 0; g < AGI; ++g) {
+    int c = g * THREADS_W + t;             // this lane's chunk id
+    int rr = c / (BMC * CPR), rem = c % (BMC * CPR);
+    int ci = rem / CPR, kof = rem % CPR;
+    a_rr[g] = rr;
+    a_cibase[g] = ((i64)(ci0 + ci) * N) * Hp * (i64)Wp;
+    a_col[g] = kof * 8;
+    a_ok[g] = c < ACH_RAW;
+    a_dst[g] = &smem[0][(i64)(g * THREADS_W + wave * 64) * 8];
+  }
+  int b_kofs[BGI]; i64 b_cobase[BGI]; bool b_ok[BGI]; u16* b_dst[BGI];
+#pragma unroll
+  for (int g = 0; g < BGI; ++g) {
+    int c = g * THREADS_W + t;
+    int kofs = c / BNC, co = c % BNC;
+    b_kofs[g] = kofs;
+    b_cobase[g] = ((i64)(co0 + co) * N) * OH * (i64)OWp2;
+    b_ok[g] = c < BCH;
+    b_dst[g] = &smem[0][ASLAB + (i64)(g * THREADS_W + wave * 64) * 8];
+  }
+
+#define WGF_STAGE(pbuf, tile_id)                                             \
+  do {                                                                       \
+    int n_ = (tile_id) / TPI, r_ = (tile_id) % TPI;                          \
+    int oh0_ = (CT == 32) ? r_ / SEG : r_ * R;                               \
+    int c0_ = (CT == 32) ? (r_ % SEG) * 32 : 0;                              \
+    _Pragma("unroll")                                                        \
+    for (int g = 0; g < AGI; ++g) {                                          \
+      int row_ = oh0_ + a_rr[g];                                             \
+      const u16* src_ = (a_ok[g] && row_ < Hp)                               \
+          ? xT + a_cibase[g] + ((i64)n_ * Hp + row_) * Wp + c0_ + a_col[g]   \
+          : zpage;                                                           \
+      __builtin_amdgcn_global_load_lds(                                      \
+          (const __attribute__((address_space(1))) unsigned int*)src_,       \
+          (__attribute__((address_space(3))) unsigned int*)                  \
+              (a_dst[g] + (pbuf) * (ASLAB + BSLAB)),                         \
+          16, 0, 0);                                                         \
+    }                                                                        \
+    _Pragma("unroll")                                                        \
+    for (int g = 0; g < BGI; ++g) {                                          \
+      int k8_ = b_kofs[g] * 8;                                               \
+      int oh_ = oh0_ + k8_ / CT;                                             \
+      int colk_ = c0_ + k8_ % CT;                                            \
+      const u16* src_ = (b_ok[g] && oh_ < OH)                                \
+          ? dyT + b_cobase[g] + ((i64)n_ * OH + oh_) * OWp2 + colk_          \
+          : zpage;                                                           \
+      __builtin_amdgcn_global_load_lds(                                      \
+          (const __attribute__((address_space(1))) unsigned int*)src_,       \
+          (__attribute__((address_space(3))) unsigned int*)                  \
+              (b_dst[g] + (pbuf) * (ASLAB + BSLAB)),                         \
+          16, 0, 0);                                                         \
+    }                                                                        \
+  } while (0)
+
+  // per-wave output: 32 ci x 32 co per tap -> i in {0,1}, j in {0,1}
+  static_assert(WAVES_N == 2 || WAVES_N == 4, "WAVES_N in {2,4}");
+  constexpr int NJ = 2;  // 32 co per wave = 2 x 16-col fragments
+  f32x4 acc[KH * KW][2][NJ];
+#pragma unroll
+  for (int tp = 0; tp < KH * KW; ++tp)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < NJ; ++j)
+        acc[tp][i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = t_hi - t_lo;
+  constexpr int SPT = AGI + BGI;           // glds per thread per stage
+  WGF_STAGE(0, t_lo);
+  if (nt > 1) WGF_STAGE(1, t_lo + 1);
+
+  int p = 0;
+  for (int tile = 0; tile < nt; ++tile) {
+    // keep exactly the newest stage in flight across the wait; the ring
+    // write target (tile+2)%3 was last read at tile-1, one barrier back
+    if (tile + 1 < nt)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(SPT) : "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const u16* As_ = &smem[p][0];
+    const u16* Bs_ = As_ + ASLAB;
+    // B fragments: [kofs][co][8], co = wn + j*16 + l15
+    bf16x8 bfr[NJ];
+#pragma unroll
+    for (int j = 0; j < NJ; ++j)
+      bfr[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          Bs_ + ((i64)(kfrag >> 3) * BNC + wn + j * 16 + l15) * 8));
+    const int rA = kfrag / CT;             // slab row of this k-frag
+    const int cA = kfrag % CT;             // col within slab row
+#pragma unroll
+    for (int kh = 0; kh < KH; ++kh) {
+      u32 d[2][6];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const u16* base = As_ +
+            ((i64)(rA + kh) * BMC + wm + i * 16 + l15) * CTW + cA;
+        *reinterpret_cast<s16x8*>(&d[i][0]) =
+            *reinterpret_cast<const s16x8*>(base);        // ds_read_b128
+        *reinterpret_cast<s16x4*>(&d[i][4]) =
+            *reinterpret_cast<const s16x4*>(base + 8);    // ds_read_b64
+      }
+      if (kh == 0) {
+        // issue tile+2's stage under the MFMA stream (ring slot is free)
+        asm volatile("" ::: "memory");
+        if (tile + 2 < nt) WGF_STAGE((tile + 2) % 3, t_lo + tile + 2);
+      }
+#pragma unroll
+      for (int kw = 0; kw < KW; ++kw) {
+        bf16x8 afr[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          u32 w[4];
+          if (kw == 0) {
+            w[0] = d[i][0]; w[1] = d[i][1]; w[2] = d[i][2]; w[3] = d[i][3];
+          } else if (kw == 2) {
+            w[0] = d[i][1]; w[1] = d[i][2]; w[2] = d[i][3]; w[3] = d[i][4];
+          } else {
+            w[0] = __builtin_amdgcn_alignbit(d[i][1], d[i][0], 16);
+            w[1] = __builtin_amdgcn_alignbit(d[i][2], d[i][1], 16);
+            w[2] = __builtin_amdgcn_alignbit(d[i][3], d[i][2], 16);
+            w[3] = __builtin_amdgcn_alignbit(d[i][4], d[i][3], 16);
+          }
+          afr[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(&w[0]));
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < NJ; ++j)
+            acc[kh * KW + kw][i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[i], bfr[j], acc[kh * KW + kw][i][j], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+    p = p + 1 == 3 ? 0 : p + 1;
+  }
+#undef WGF_STAGE
+
+  // epilogue: fp32 atomics into c32[(kh*KW+kw)*Cin + ci][co]
+#pragma unroll
+  for (int tp = 0; tp < KH * KW; ++tp) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row_base = tp * Cin + ci0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+      for (int j = 0; j < NJ; ++j) {
+        int col = co0 + wn + j * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          atomicAdd(&c32[(i64)(row_base + r) * Cout + col],
+                    acc[tp][i][j][r]);
+      }
+    }
+  }
+}
+
+}  // namespace dvgg

@@ -79,6 +79,12 @@ CONV_CASES = [
     (2, 16, 16, 64, 128, 1, 1, 0, "conv1x1"),
     (2, 16, 16, 64, 128, 3, 2, 1, "conv3x3s2"),
     (2, 16, 16, 64, 128, 1, 2, 0, "conv1x1s2"),
+    # tap-fused 3x3 wgrad coverage (wgrad_fused.h): CT=32 whole-segment,
+    # OWp2 over-pad (33->64), CT=8 with padded k-rows, CT=16 pad=0 + co64
+    (2, 28, 28, 64, 64, 3, 1, 1, "conv3x3-ow28"),
+    (2, 33, 33, 64, 128, 3, 1, 1, "conv3x3-ow33pad64"),
+    (2, 7, 7, 128, 128, 3, 1, 1, "conv3x3-ow7"),
+    (2, 14, 14, 128, 64, 3, 1, 0, "conv3x3-p0-co64"),
 ]
 
 
@@ -351,3 +357,31 @@ def test_conv_fwd_dgrad_variants(env, monkeypatch):
     yr = cpu_ref.conv2d(xr, w.float(), None, s, p)
     yr.backward(dy.float())
     check(dx, xr.grad, 0.05, "variant_dgrad")
+
+
+@pytest.mark.parametrize("shape", [
+    (2, 13, 13, 256, 256, 1),   # CT=16, WAVES_N=4, pad row (OH=13 -> 14)
+    (2, 28, 28, 64, 64, 1),     # CT=32 single segment, WAVES_N=2
+    (2, 33, 33, 64, 128, 1),    # OWp2 over-pad 40 -> 64, 2 segments
+    (2, 7, 7, 128, 128, 1),     # CT=8, R=4, padded k-rows
+    (2, 14, 14, 128, 64, 0),    # CT=16, pad=0, Cout=64
+], ids=["ct16", "ct32", "ct32pad", "ct8", "ct16p0"])
+def test_wgrad_fused_vs_classic(shape, monkeypatch):
+    """The tap-fused 3x3 wgrad (wgrad_fused.h) must agree with both the
+    classic implicit-GEMM path and the CPU reference on every CT regime."""
+    N, H, W, Ci, Co, p = shape
+    k, s = 3, 1
+    torch.manual_seed(81)
+    x = rnd(N, H, W, Ci, scale=0.5)
+    dy_sp = [N, H, W, Co] if p == 1 else [N, H - 2, W - 2, Co]
+    dy = rnd(*dy_sp, scale=0.5)
+    w32 = torch.zeros(k, k, Ci, Co, requires_grad=True)
+    cpu_ref.conv2d(x.float(), w32, None, s, p).backward(dy.float())
+    xg, dyg = x.to(DEV), dy.to(DEV)
+    monkeypatch.setenv("DVGG_WGRAD_FUSED", "1")
+    dw_f = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
+    monkeypatch.setenv("DVGG_WGRAD_FUSED", "0")
+    dw_c = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
+    check(dw_f, w32.grad, 0.06, "wgrad_fused_vs_cpu")
+    check(dw_c, w32.grad, 0.06, "wgrad_classic_vs_cpu")
+    check(dw_f, dw_c.float().cpu(), 0.02, "wgrad_fused_vs_classic")
