@@ -915,10 +915,12 @@ static at::Tensor pad_channels8(const at::Tensor& x) {
   return out;
 }
 
-// ---------------- tap-fused 3x3 wgrad (wgrad_fused.h) -----------------
+// ---------------- tap-fused 3x3 / 5x5 wgrad (wgrad_fused.h) -----------
 static bool wgrad_fused_ok(int KH, int KW, int stride, int Cin, int Cout) {
-  return KH == 3 && KW == 3 && stride == 1 && Cin % 64 == 0 &&
-         Cout % 64 == 0 && env_int("DVGG_WGRAD_FUSED", 1);
+  if (stride != 1 || Cin % 64 != 0 || Cout % 64 != 0) return false;
+  if (KH == 3 && KW == 3) return env_int("DVGG_WGRAD_FUSED", 1);
+  if (KH == 5 && KW == 5) return env_int("DVGG_WGRAD_FUSED5", 1);
+  return false;
 }
 
 // OW -> the fused kernel's padded column extent: multiple of 8 with
@@ -931,14 +933,15 @@ static int owp2_of(int OW) {
 }
 
 static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
-                        at::Tensor& c32, int Cin, int Cout, int N, int Hp,
-                        int Wp, int OH, int OWp2) {
+                        at::Tensor& c32, int KH, int KW, int Cin, int Cout,
+                        int N, int Hp, int Wp, int OH, int OWp2) {
   int CT = std::min(OWp2, 32), R = 32 / CT;
   int SEG = CT == 32 ? OWp2 / 32 : 1;
   int OHK = (OH + R - 1) / R * R;
   int TPI = CT == 32 ? OH * SEG : OHK / R;
   i64 tiles_k = (i64)N * TPI;
-  int wavesn = (Cout % 128 == 0) ? 4 : 2;
+  bool k5 = KH == 5;
+  int wavesn = (!k5 && Cout % 128 == 0) ? 4 : 2;
   int tiles = (Cin / 64) * (Cout / (32 * wavesn));
   int target = env_int("DVGG_WGF_BLOCKS", 768) * sk_scale() / 100;
   int sk = std::max(1, std::min<int>(int(tiles_k),
@@ -948,19 +951,23 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   const u16* zp = zero_page(xT.options());
   dim3 grid(tiles, sk);
   auto s = cur_stream();
-#define WGF_LAUNCH(CTV, WNV)                                                 \
-  hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WNV>), grid,                  \
-                     dim3(2 * WNV * 64), 0, s, bfp(xT), bfp(dyT),            \
+#define WGF_LAUNCH(CTV, WMV, WNV, KHV, KWV)                                  \
+  hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WMV, WNV, KHV, KWV>), grid,   \
+                     dim3(WMV * WNV * 64), 0, s, bfp(xT), bfp(dyT),          \
                      c32.data_ptr<float>(), zp, Cin, Cout, N, Hp, Wp, OH,    \
                      OWp2, int(tiles_k), sk)
-  if (wavesn == 4) {
-    if (CT == 32) WGF_LAUNCH(32, 4);
-    else if (CT == 16) WGF_LAUNCH(16, 4);
-    else WGF_LAUNCH(8, 4);
+  if (k5) {          // 5x5: 16ci x 32co per wave, 25 accumulator sets
+    if (CT == 32) WGF_LAUNCH(32, 4, 2, 5, 5);
+    else if (CT == 16) WGF_LAUNCH(16, 4, 2, 5, 5);
+    else WGF_LAUNCH(8, 4, 2, 5, 5);
+  } else if (wavesn == 4) {
+    if (CT == 32) WGF_LAUNCH(32, 2, 4, 3, 3);
+    else if (CT == 16) WGF_LAUNCH(16, 2, 4, 3, 3);
+    else WGF_LAUNCH(8, 2, 4, 3, 3);
   } else {
-    if (CT == 32) WGF_LAUNCH(32, 2);
-    else if (CT == 16) WGF_LAUNCH(16, 2);
-    else WGF_LAUNCH(8, 2);
+    if (CT == 32) WGF_LAUNCH(32, 2, 2, 3, 3);
+    else if (CT == 16) WGF_LAUNCH(16, 2, 2, 3, 3);
+    else WGF_LAUNCH(8, 2, 2, 3, 3);
   }
 #undef WGF_LAUNCH
 }
@@ -1128,7 +1135,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
     auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-    wgrad_fused(xT, dyT, c32, Cin, Cout, N, Hp, Wp, OH, OWp);
+    wgrad_fused(xT, dyT, c32, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
     run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
     return dw;
   }
@@ -1296,7 +1303,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     int Hp = H + 2 * pad;
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
-    wgrad_fused(xT, dyT, c32, Cin, Cout, N, Hp, Wp, OH, OWp);
+    wgrad_fused(xT, dyT, c32, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
     run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
     at::Tensor db;
     if (need_db) {

@@ -37,17 +37,21 @@
 namespace dvgg {
 
 // CT = k-tile column extent; BMC = 64 ci rows; BNC = 32*WAVES_N co cols.
-template <int CT, int WAVES_N>
-__global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
+// WAVES_M=2 -> 32 ci per wave (2 A-frags; 3x3 fits 9x2x2 accumulators),
+// WAVES_M=4 -> 16 ci per wave (1 A-frag; 5x5 fits 25x1x2).
+template <int CT, int WAVES_M, int WAVES_N, int KH, int KW>
+__global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_fused3_kernel(
     const u16* __restrict__ xT,    // [Cin][N][Hp][Wp], Wp = OWp2 + 8
     const u16* __restrict__ dyT,   // [Cout][N][OH][OWp2], zero col pad
     float* __restrict__ c32,       // [9*Cin][Cout] pre-zeroed
     const u16* __restrict__ zpage, int Cin, int Cout, int N, int Hp, int Wp,
     int OH, int OWp2, int tiles_k, int slices) {
-  constexpr int KH = 3, KW = 3;
-  constexpr int NW = 2 * WAVES_N;          // waves per block
+  constexpr int NW = WAVES_M * WAVES_N;    // waves per block
   constexpr int THREADS_W = NW * 64;
-  constexpr int BMC = 64, BNC = 32 * WAVES_N;
+  constexpr int NI = WAVES_M == 2 ? 2 : 1; // A fragments per wave
+  constexpr int BMC = WAVES_M * 16 * NI;   // = 64 ci rows per block
+  constexpr int BNC = 32 * WAVES_N;
+  static_assert(KW <= 5, "kw window covers 12 staged elements");
   constexpr int R = 32 / CT;               // output rows per k-tile
   constexpr int RH = R + KH - 1;           // staged slab rows
   constexpr int CTW = CT + 8;              // slab cols (halo + align slack)
@@ -76,7 +80,7 @@ __global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
 
   const int t = threadIdx.x;
   const int lane = t & 63, wave = t >> 6;
-  const int wm = (wave / WAVES_N) * 32;    // ci sub-tile base
+  const int wm = (wave / WAVES_N) * (16 * NI);  // ci sub-tile base
   const int wn = (wave % WAVES_N) * 32;    // co sub-tile base
   const int l15 = lane & 15;
   const int kfrag = (lane >> 4) * 8;       // k offset of this lane's frags
@@ -148,14 +152,14 @@ __global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
     }                                                                        \
   } while (0)
 
-  // per-wave output: 32 ci x 32 co per tap -> i in {0,1}, j in {0,1}
+  // per-wave output: (16*NI) ci x 32 co per tap
   static_assert(WAVES_N == 2 || WAVES_N == 4, "WAVES_N in {2,4}");
   constexpr int NJ = 2;  // 32 co per wave = 2 x 16-col fragments
-  f32x4 acc[KH * KW][2][NJ];
+  f32x4 acc[KH * KW][NI][NJ];
 #pragma unroll
   for (int tp = 0; tp < KH * KW; ++tp)
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < NI; ++i)
 #pragma unroll
       for (int j = 0; j < NJ; ++j)
         acc[tp][i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -186,9 +190,9 @@ __global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
     const int cA = kfrag % CT;             // col within slab row
 #pragma unroll
     for (int kh = 0; kh < KH; ++kh) {
-      u32 d[2][6];
+      u32 d[NI][6];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < NI; ++i) {
         const u16* base = As_ +
             ((i64)(rA + kh) * BMC + wm + i * 16 + l15) * CTW + cA;
         *reinterpret_cast<s16x8*>(&d[i][0]) =
@@ -203,25 +207,24 @@ __global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
       }
 #pragma unroll
       for (int kw = 0; kw < KW; ++kw) {
-        bf16x8 afr[2];
+        bf16x8 afr[NI];
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < NI; ++i) {
           u32 w[4];
-          if (kw == 0) {
-            w[0] = d[i][0]; w[1] = d[i][1]; w[2] = d[i][2]; w[3] = d[i][3];
-          } else if (kw == 2) {
-            w[0] = d[i][1]; w[1] = d[i][2]; w[2] = d[i][3]; w[3] = d[i][4];
-          } else {
-            w[0] = __builtin_amdgcn_alignbit(d[i][1], d[i][0], 16);
-            w[1] = __builtin_amdgcn_alignbit(d[i][2], d[i][1], 16);
-            w[2] = __builtin_amdgcn_alignbit(d[i][3], d[i][2], 16);
-            w[3] = __builtin_amdgcn_alignbit(d[i][4], d[i][3], 16);
+          if (kw % 2 == 0) {       // even shift: dword re-select, free
+#pragma unroll
+            for (int q = 0; q < 4; ++q) w[q] = d[i][kw / 2 + q];
+          } else {                 // odd shift: 16-bit funnel
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+              w[q] = __builtin_amdgcn_alignbit(d[i][kw / 2 + 1 + q],
+                                               d[i][kw / 2 + q], 16);
           }
           afr[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(&w[0]));
         }
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int i = 0; i < 2; ++i)
+        for (int i = 0; i < NI; ++i)
 #pragma unroll
           for (int j = 0; j < NJ; ++j)
             acc[kh * KW + kw][i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -237,7 +240,7 @@ __global__ __launch_bounds__((2 * WAVES_N) * 64) void wgrad_fused3_kernel(
 #pragma unroll
   for (int tp = 0; tp < KH * KW; ++tp) {
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < NI; ++i) {
       int row_base = tp * Cin + ci0 + wm + i * 16 + (lane >> 4) * 4;
 #pragma unroll
       for (int j = 0; j < NJ; ++j) {

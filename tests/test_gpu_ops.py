@@ -371,16 +371,33 @@ def test_wgrad_fused_vs_classic(shape, monkeypatch):
     classic implicit-GEMM path and the CPU reference on every CT regime."""
     N, H, W, Ci, Co, p = shape
     k, s = 3, 1
+    _wgrad_fused_check(N, H, W, Ci, Co, p, k, s, monkeypatch,
+                       "DVGG_WGRAD_FUSED")
+
+
+@pytest.mark.parametrize("shape", [
+    (2, 27, 27, 64, 256, 2),    # VGG-F conv2 geometry (CT=32)
+    (2, 13, 13, 64, 64, 2),     # CT=16 5x5
+], ids=["k5ct32", "k5ct16"])
+def test_wgrad_fused5_vs_classic(shape, monkeypatch):
+    """5x5 instantiation (WAVES_M=4, 25 accumulator sets)."""
+    N, H, W, Ci, Co, p = shape
+    k, s = 5, 1
+    _wgrad_fused_check(N, H, W, Ci, Co, p, k, s, monkeypatch,
+                       "DVGG_WGRAD_FUSED5")
+
+
+def _wgrad_fused_check(N, H, W, Ci, Co, p, k, s, monkeypatch, env_name):
     torch.manual_seed(81)
     x = rnd(N, H, W, Ci, scale=0.5)
-    dy_sp = [N, H, W, Co] if p == 1 else [N, H - 2, W - 2, Co]
-    dy = rnd(*dy_sp, scale=0.5)
+    oh = (H + 2 * p - k) // s + 1
+    dy = rnd(N, oh, oh, Co, scale=0.5)
     w32 = torch.zeros(k, k, Ci, Co, requires_grad=True)
     cpu_ref.conv2d(x.float(), w32, None, s, p).backward(dy.float())
     xg, dyg = x.to(DEV), dy.to(DEV)
-    monkeypatch.setenv("DVGG_WGRAD_FUSED", "1")
+    monkeypatch.setenv(env_name, "1")
     dw_f = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
-    monkeypatch.setenv("DVGG_WGRAD_FUSED", "0")
+    monkeypatch.setenv(env_name, "0")
     dw_c = ext.conv2d_wgrad(dyg, xg, k, k, s, p)
     check(dw_f, w32.grad, 0.06, "wgrad_fused_vs_cpu")
     check(dw_c, w32.grad, 0.06, "wgrad_classic_vs_cpu")
