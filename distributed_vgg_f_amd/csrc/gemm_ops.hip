@@ -943,9 +943,14 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   bool k5 = KH == 5;
   int wavesn = (!k5 && Cout % 128 == 0) ? 4 : 2;
   int tiles = (Cin / 64) * (Cout / (32 * wavesn));
-  int target = env_int("DVGG_WGF_BLOCKS", 768) * sk_scale() / 100;
+  int target = env_int("DVGG_WGF_BLOCKS", 256) * sk_scale() / 100;
   int sk = std::max(1, std::min<int>(int(tiles_k),
                                      target / std::max(1, tiles)));
+  // atomic budget: every slice atomically re-adds the whole dw output
+  // (KH*KW*Cin*Cout fp32); cap total atomic ops (~1/cycle/CU processing)
+  i64 outs = (i64)KH * KW * Cin * Cout;
+  i64 amax = (i64)env_int("DVGG_WGF_ATOMIC_MB", 24) * (1 << 20);
+  sk = std::max(std::min(sk, 8), std::min<int>(sk, int(amax / outs)));
   i64 span = (tiles_k + sk - 1) / sk;
   sk = int((tiles_k + span - 1) / span);   // non-empty slices only
   const u16* zp = zero_page(xT.options());
@@ -955,7 +960,7 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WMV, WNV, KHV, KWV>), grid,   \
                      dim3(WMV * WNV * 64), 0, s, bfp(xT), bfp(dyT),          \
                      c32.data_ptr<float>(), zp, Cin, Cout, N, Hp, Wp, OH,    \
-                     OWp2, int(tiles_k), sk)
+                     OWp2, int(tiles_k), sk, env_int("DVGG_WGF_EPI", 1))
   if (k5) {          // 5x5: 16ci x 32co per wave, 25 accumulator sets
     if (CT == 32) WGF_LAUNCH(32, 4, 2, 5, 5);
     else if (CT == 16) WGF_LAUNCH(16, 4, 2, 5, 5);
