@@ -536,12 +536,41 @@ __global__ void finalize_kernel(const float* __restrict__ c32,
   }
 }
 
+// 1-elem/thread finalize for SMALL outputs x MANY slices (the 8-wide
+// kernel would launch total/8/256 blocks — 18 blocks for a 37 MB
+// 256-slice slab read on VGG-16 conv1_2): full-width parallelism, each
+// wave's lanes read consecutive elems of each slice (coalesced).
+__global__ void finalize_narrow_kernel(const float* __restrict__ c32,
+                                       u16* __restrict__ out, i64 total,
+                                       int slices, i64 slice_stride) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (id >= total) return;
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  const float* p = c32 + id;
+  int z = 0;
+  for (; z + 4 <= slices; z += 4) {
+    a0 += p[(i64)z * slice_stride];
+    a1 += p[(i64)(z + 1) * slice_stride];
+    a2 += p[(i64)(z + 2) * slice_stride];
+    a3 += p[(i64)(z + 3) * slice_stride];
+  }
+  for (; z < slices; ++z) a0 += p[(i64)z * slice_stride];
+  out[id] = f2bf((a0 + a1) + (a2 + a3));
+}
+
 static void run_finalize(const at::Tensor& c32, at::Tensor& out,
                          const at::Tensor* bias, int epi, int R, int C,
                          int slices = 1) {
   i64 total = (i64)R * C;
   i64 sstride = (i64)R * C;
   int blocks = int((total / 8 + 256) / 256);
+  if (epi == 0 && slices > 8 && blocks < 512) {
+    hipLaunchKernelGGL(finalize_narrow_kernel,
+                       dim3(int((total + 255) / 256)), dim3(256), 0,
+                       cur_stream(), c32.data_ptr<float>(), bfp_mut(out),
+                       total, slices, sstride);
+    return;
+  }
   const u16* bp = bias ? bfp(*bias) : nullptr;
   auto s = cur_stream();
   switch (epi) {
@@ -933,7 +962,7 @@ static int owp2_of(int OW) {
 }
 
 static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
-                        at::Tensor& c32, int KH, int KW, int Cin, int Cout,
+                        at::Tensor& dw2d, int KH, int KW, int Cin, int Cout,
                         int N, int Hp, int Wp, int OH, int OWp2) {
   int CT = std::min(OWp2, 32), R = 32 / CT;
   int SEG = CT == 32 ? OWp2 / 32 : 1;
@@ -946,13 +975,16 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   int target = env_int("DVGG_WGF_BLOCKS", 256) * sk_scale() / 100;
   int sk = std::max(1, std::min<int>(int(tiles_k),
                                      target / std::max(1, tiles)));
-  // atomic budget: every slice atomically re-adds the whole dw output
-  // (KH*KW*Cin*Cout fp32); cap total atomic ops (~1/cycle/CU processing)
+  // slab budget: every slice writes a full |dw| fp32 slab (read back by
+  // the summing finalize) — cap the total slab footprint
   i64 outs = (i64)KH * KW * Cin * Cout;
-  i64 amax = (i64)env_int("DVGG_WGF_ATOMIC_MB", 24) * (1 << 20);
-  sk = std::max(std::min(sk, 8), std::min<int>(sk, int(amax / outs)));
+  i64 emax = (i64)env_int("DVGG_WGF_SLAB_MB", 256) * (1 << 20) / 4;
+  sk = std::max(std::min(sk, 8), std::min<int>(sk, int(emax / outs)));
   i64 span = (tiles_k + sk - 1) / sk;
   sk = int((tiles_k + span - 1) / span);   // non-empty slices only
+  // per-slice slabs, fully written by construction: no zero-fill
+  auto c32 = at::empty({sk, (i64)KH * KW * Cin, Cout},
+                       xT.options().dtype(at::kFloat));
   const u16* zp = zero_page(xT.options());
   dim3 grid(tiles, sk);
   auto s = cur_stream();
@@ -960,7 +992,7 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WMV, WNV, KHV, KWV>), grid,   \
                      dim3(WMV * WNV * 64), 0, s, bfp(xT), bfp(dyT),          \
                      c32.data_ptr<float>(), zp, Cin, Cout, N, Hp, Wp, OH,    \
-                     OWp2, int(tiles_k), sk, env_int("DVGG_WGF_EPI", 1))
+                     OWp2, int(tiles_k), sk)
   if (k5) {          // 5x5: 16ci x 32co per wave, 25 accumulator sets
     if (CT == 32) WGF_LAUNCH(32, 4, 2, 5, 5);
     else if (CT == 16) WGF_LAUNCH(16, 4, 2, 5, 5);
@@ -970,11 +1002,15 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
     else if (CT == 16) WGF_LAUNCH(16, 2, 4, 3, 3);
     else WGF_LAUNCH(8, 2, 4, 3, 3);
   } else {
-    if (CT == 32) WGF_LAUNCH(32, 2, 2, 3, 3);
-    else if (CT == 16) WGF_LAUNCH(16, 2, 2, 3, 3);
-    else WGF_LAUNCH(8, 2, 2, 3, 3);
+    // Cout == 64: keep 8 waves (a 4-wave block at ~195 VGPR is 1
+    // wave/SIMD — measured catastrophic on VGG-16 conv1_x) by going
+    // 16ci x 32co per wave (NI=1, 72 acc regs)
+    if (CT == 32) WGF_LAUNCH(32, 4, 2, 3, 3);
+    else if (CT == 16) WGF_LAUNCH(16, 4, 2, 3, 3);
+    else WGF_LAUNCH(8, 4, 2, 3, 3);
   }
 #undef WGF_LAUNCH
+  run_finalize(c32, dw2d, nullptr, 0, KH * KW * Cin, Cout, sk);
 }
 
 // glds wgrad (stride==1, Cin %% 128 == 0): substituted-reduction implicit
@@ -1139,9 +1175,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     int Hp = H + 2 * pad;
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
-    auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
-    wgrad_fused(xT, dyT, c32, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
-    run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+    wgrad_fused(xT, dyT, dw2d, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
     return dw;
   }
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
@@ -1303,13 +1337,11 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   if (wfused) {
     int Hp = H + 2 * pad;
     int Wp = ceil8(std::max(W + 2 * pad, (OWp - 1) * stride + KW));
     auto xT = nhwc2cnhw_pad(x, Hp, Wp, pad, pad, 'x');
-    wgrad_fused(xT, dyT, c32, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
-    run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+    wgrad_fused(xT, dyT, dw2d, KH, KW, Cin, Cout, N, Hp, Wp, OH, OWp);
     at::Tensor db;
     if (need_db) {
       db = db_out ? *db_out : at::empty({Cout}, dy.options());
@@ -1323,6 +1355,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
+  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
                   ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);
