@@ -1130,7 +1130,9 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
     auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
     if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
-    return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+    auto t = p.contiguous();
+    if (t.data_ptr() == w.data_ptr()) t = t.clone();  // see conv2d_bwd note
+    return t.view({Cin, (i64)KH * KW * Cout});
   });
   TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
   ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad,
@@ -1301,7 +1303,11 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
       auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
       if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
-      return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+      auto t = p.contiguous();
+      // KH*KW==1: contiguous() returns a VIEW of w — clone so the cache
+      // entry owns its storage (the refresh copy_ must not alias w)
+      if (t.data_ptr() == w.data_ptr()) t = t.clone();
+      return t.view({Cin, (i64)KH * KW * Cout});
     });
     TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
     int Mx = N * H * W;

@@ -402,3 +402,25 @@ def _wgrad_fused_check(N, H, W, Ci, Co, p, k, s, monkeypatch, env_name):
     check(dw_f, w32.grad, 0.06, "wgrad_fused_vs_cpu")
     check(dw_c, w32.grad, 0.06, "wgrad_classic_vs_cpu")
     check(dw_f, dw_c.float().cpu(), 0.02, "wgrad_fused_vs_classic")
+
+
+def test_cached_wd_refresh_1x1s2():
+    """ResNet downsample (1x1 stride-2) with the weight-form cache on:
+    contiguous() of the [Cin,1,Cout] permute aliases w, so the cache must
+    clone at creation and refresh across generations without aliasing."""
+    torch.manual_seed(91)
+    N, H, W, Ci, Co, k, s, p = 2, 16, 16, 64, 128, 1, 2, 0
+    x = rnd(N, H, W, Ci, scale=0.5).to(DEV)
+    w = rnd(k, k, Ci, Co, scale=0.1).to(DEV)
+    oh = (H + 2 * p - k) // s + 1
+    dy = rnd(N, oh, oh, Co, scale=0.5).to(DEV)
+    y = ext.conv2d_fwd(x, w, None, s, p, False, True)
+    dx1, dw1, _ = ext.conv2d_bwd(dy, y, x, w, s, p, False, True, False,
+                                 None, None, True)
+    ext.bump_weight_gen()
+    w.copy_(rnd(k, k, Ci, Co, scale=0.1))   # new generation's weights
+    dx2, dw2, _ = ext.conv2d_bwd(dy, y, x, w, s, p, False, True, False,
+                                 None, None, True)
+    # dgrad must reflect the NEW weights (stale cache would reuse old wD)
+    dx_ref = ext.conv2d_dgrad(dy, w, H, W, s, p)
+    check(dx2, dx_ref.float().cpu(), 1e-3, "cached_wd_refresh")
