@@ -973,21 +973,49 @@ static void wgrad_fused(const at::Tensor& xT, const at::Tensor& dyT,
   int wavesn = (!k5 && Cout % 128 == 0) ? 4 : 2;
   int tiles = (Cin / 64) * (Cout / (32 * wavesn));
   int target = env_int("DVGG_WGF_BLOCKS", 256) * sk_scale() / 100;
-  int sk = std::max(1, std::min<int>(int(tiles_k),
+  // rolling-window ring variant: slices are whole (n, c0) column runs
+  int LT = CT == 32 ? OH : OHK / R;        // tiles per run
+  int runs_total = N * SEG;
+  bool ring = env_int("DVGG_WGF_RING", 1) && LT >= 3;
+  i64 sk_units = ring ? runs_total : tiles_k;
+  int sk = std::max(1, std::min<int>(int(sk_units),
                                      target / std::max(1, tiles)));
   // slab budget: every slice writes a full |dw| fp32 slab (read back by
   // the summing finalize) — cap the total slab footprint
   i64 outs = (i64)KH * KW * Cin * Cout;
   i64 emax = (i64)env_int("DVGG_WGF_SLAB_MB", 256) * (1 << 20) / 4;
   sk = std::max(std::min(sk, 8), std::min<int>(sk, int(emax / outs)));
-  i64 span = (tiles_k + sk - 1) / sk;
-  sk = int((tiles_k + span - 1) / span);   // non-empty slices only
+  i64 span = (sk_units + sk - 1) / sk;
+  sk = int((sk_units + span - 1) / span);  // non-empty slices only
   // per-slice slabs, fully written by construction: no zero-fill
   auto c32 = at::empty({sk, (i64)KH * KW * Cin, Cout},
                        xT.options().dtype(at::kFloat));
   const u16* zp = zero_page(xT.options());
   dim3 grid(tiles, sk);
   auto s = cur_stream();
+  if (ring) {
+#define WGR_LAUNCH(CTV, WMV, WNV, KHV, KWV)                                  \
+    hipLaunchKernelGGL((wgrad_ring_kernel<CTV, WMV, WNV, KHV, KWV>), grid,   \
+                       dim3(WMV * WNV * 64), 0, s, bfp(xT), bfp(dyT),        \
+                       c32.data_ptr<float>(), zp, Cin, Cout, N, Hp, Wp, OH,  \
+                       OWp2, runs_total, LT, sk)
+    if (k5) {
+      if (CT == 32) WGR_LAUNCH(32, 4, 2, 5, 5);
+      else if (CT == 16) WGR_LAUNCH(16, 4, 2, 5, 5);
+      else WGR_LAUNCH(8, 4, 2, 5, 5);
+    } else if (wavesn == 4) {
+      if (CT == 32) WGR_LAUNCH(32, 2, 4, 3, 3);
+      else if (CT == 16) WGR_LAUNCH(16, 2, 4, 3, 3);
+      else WGR_LAUNCH(8, 2, 4, 3, 3);
+    } else {
+      if (CT == 32) WGR_LAUNCH(32, 4, 2, 3, 3);
+      else if (CT == 16) WGR_LAUNCH(16, 4, 2, 3, 3);
+      else WGR_LAUNCH(8, 4, 2, 3, 3);
+    }
+#undef WGR_LAUNCH
+    run_finalize(c32, dw2d, nullptr, 0, KH * KW * Cin, Cout, sk);
+    return;
+  }
 #define WGF_LAUNCH(CTV, WMV, WNV, KHV, KWV)                                  \
   hipLaunchKernelGGL((wgrad_fused3_kernel<CTV, WMV, WNV, KHV, KWV>), grid,   \
                      dim3(WMV * WNV * 64), 0, s, bfp(xT), bfp(dyT),          \

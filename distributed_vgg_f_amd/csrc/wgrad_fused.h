@@ -258,4 +258,255 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_fused3_kernel(
   }
 }
 
+
+
+// ------------------------------------------------------------------ ring
+// Rolling-window variant: within one (n, c0) COLUMN RUN the x rows of
+// consecutive k-tiles overlap by KH-1 rows; the plain kernel re-stages
+// the whole RH-row slab every tile (3x A traffic for CT=32, 2x for
+// CT=16). Here tiles walk oh within a run and each tile stages exactly
+// ONE new R-row "set" into an NSS-slot LDS ring; the KH-1 halo rows stay
+// resident. Run transitions batch-stage the next run's S0+1 sets during
+// the second-to-last tile, keeping the counted-vmcnt discipline at two
+// static values (steady vmcnt(2); vmcnt(S0+3) across the seam).
+// Slices are whole runs; host guarantees LT >= 3 tiles per run.
+template <int CT, int WAVES_M, int WAVES_N, int KH, int KW>
+__global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
+    const u16* __restrict__ xT,    // [Cin][N][Hp][Wp], Wp = OWp2 + 8
+    const u16* __restrict__ dyT,   // [Cout][N][OH][OWp2], zero col pad
+    float* __restrict__ c32,       // [slices][KH*KW*Cin][Cout] slabs
+    const u16* __restrict__ zpage, int Cin, int Cout, int N, int Hp, int Wp,
+    int OH, int OWp2, int runs_total, int LT, int slices) {
+  constexpr int NW = WAVES_M * WAVES_N;
+  constexpr int THREADS_W = NW * 64;
+  static_assert(THREADS_W == 512, "ring kernel assumes 512 threads");
+  constexpr int NI = WAVES_M == 2 ? 2 : 1;
+  constexpr int BMC = WAVES_M * 16 * NI;   // 64
+  constexpr int BNC = 32 * WAVES_N;
+  constexpr int R = 32 / CT;               // rows per tile / per set
+  constexpr int RH = R + KH - 1;           // live window rows
+  constexpr int S0 = (RH + R - 1) / R;     // window sets
+  constexpr int NSS = 2 * S0 + 2;          // ring slots (seam batch safe)
+  constexpr int CTW = CT + 8;
+  constexpr int CPR = CTW / 8;
+  constexpr int SETC = R * BMC * CPR;      // real chunks per set (<=512)
+  static_assert(SETC <= 512, "one glds per thread per set");
+  constexpr int SETE = 512 * 8;            // padded set elems (8 KiB)
+  constexpr int BTC = 4 * BNC;             // real B chunks per tile
+  constexpr int BTE = 512 * 8;             // padded B tile elems (8 KiB)
+  constexpr int NB = 4;                    // B ring slots
+  __shared__ u16 smem[NSS * SETE + NB * BTE];
+
+  const int tiles_m = Cin / BMC, tiles_n = Cout / BNC;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int ci0 = int(wg) / tiles_n * BMC;
+  const int co0 = int(wg) % tiles_n * BNC;
+  const int SEG = (CT == 32) ? (OWp2 / 32) : 1;
+  // slice = whole runs
+  const int rspan = (runs_total + slices - 1) / slices;
+  const int r_lo = blockIdx.y * rspan;
+  const int r_hi = min(runs_total, r_lo + rspan);
+  if (r_lo >= r_hi) return;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave / WAVES_N) * (16 * NI);
+  const int wn = (wave % WAVES_N) * 32;
+  const int l15 = lane & 15;
+  const int kfrag = (lane >> 4) * 8;
+
+  // ---- staging descriptors (one glds per thread per A set / B tile)
+  // A chunk c = (r, ci, kof) within a set; dest wave-uniform
+  const int a_r = t / (BMC * CPR);
+  const int a_ci = (t % (BMC * CPR)) / CPR;
+  const int a_kof = (t % CPR) * 8;
+  const bool a_ok = t < SETC;
+  const i64 a_cib = ((i64)(ci0 + a_ci) * N) * Hp * (i64)Wp;
+  u16* const a_dst0 = &smem[(i64)(wave * 64) * 8];
+  const int b_kofs = t / BNC;
+  const int b_co = t % BNC;
+  const bool b_ok = t < BTC;               // BNC=64: half the threads idle
+  const i64 b_cob = ((i64)(co0 + b_co) * N) * OH * (i64)OWp2;
+  u16* const b_dst0 = &smem[NSS * SETE + (i64)(wave * 64) * 8];
+
+  // stage A set s (run-local) of run (n_, c0_) into ring slot 'slot'
+#define WGR_STAGE_A(slot, n_, c0_, s_)                                       \
+  do {                                                                       \
+    int row_ = (s_) * R + a_r;                                               \
+    const u16* src_ = (a_ok && row_ < Hp)                                    \
+        ? xT + a_cib + ((i64)(n_) * Hp + row_) * Wp + (c0_) + a_kof          \
+        : zpage;                                                             \
+    __builtin_amdgcn_global_load_lds(                                        \
+        (const __attribute__((address_space(1))) unsigned int*)src_,         \
+        (__attribute__((address_space(3))) unsigned int*)                    \
+            (a_dst0 + (i64)(slot) * SETE),                                   \
+        16, 0, 0);                                                           \
+  } while (0)
+#define WGR_STAGE_B(bslot, n_, c0_, tile_)                                   \
+  do {                                                                       \
+    int k8_ = b_kofs * 8;                                                    \
+    int oh_ = (tile_) * R + k8_ / CT;                                        \
+    const u16* src_ = (b_ok && oh_ < OH)                                     \
+        ? dyT + b_cob + ((i64)(n_) * OH + oh_) * OWp2 + (c0_) + k8_ % CT     \
+        : zpage;                                                             \
+    __builtin_amdgcn_global_load_lds(                                        \
+        (const __attribute__((address_space(1))) unsigned int*)src_,         \
+        (__attribute__((address_space(3))) unsigned int*)                    \
+            (b_dst0 + (i64)(bslot) * BTE),                                   \
+        16, 0, 0);                                                           \
+  } while (0)
+
+  f32x4 acc[KH * KW][NI][2];
+#pragma unroll
+  for (int tp = 0; tp < KH * KW; ++tp)
+#pragma unroll
+    for (int i = 0; i < NI; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[tp][i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // run-local state; slot counters advance monotonically mod ring size
+  int run = r_lo;
+  int n_r = run / SEG, c0_r = (run % SEG) * 32;
+  int sbase = 0;       // ring slot of the current tile's FIRST window set
+  int bbase = 0;
+  // prologue of the first run. ISSUE ORDER MATTERS: vmcnt counts in
+  // issue order, so tile0's stages go first and tile1's (A top set + B)
+  // last — the steady vmcnt(2) then leaves exactly tile1's pair in
+  // flight while guaranteeing tile0's window.
+#pragma unroll
+  for (int s = 0; s < S0; ++s) WGR_STAGE_A(s % NSS, n_r, c0_r, s);
+  WGR_STAGE_B(0, n_r, c0_r, 0);
+  WGR_STAGE_A(S0 % NSS, n_r, c0_r, S0);
+  WGR_STAGE_B(1, n_r, c0_r, 1);
+  int sfree = (S0 + 1) % NSS;  // next free A slot
+  int bfree = 2;
+
+  for (; run < r_hi; ++run) {
+    if (run > r_lo) { n_r = run / SEG; c0_r = (run % SEG) * 32; }
+    const bool last_run = run + 1 == r_hi;
+    const int nn = (run + 1 < r_hi) ? (run + 1) / SEG : 0;
+    const int nc0 = (run + 1 < r_hi) ? ((run + 1) % SEG) * 32 : 0;
+    for (int tile = 0; tile < LT; ++tile) {
+      // ---- wait + barrier: current tile's window and B tile are in LDS
+      const bool seam = (tile == LT - 1) && !last_run;
+      if (tile == LT - 1 && last_run)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      else if (seam)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(S0 + 3) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+
+      const u16* Bs_ = &smem[NSS * SETE + (i64)bbase * BTE];
+      bf16x8 bfr[2];
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        bfr[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+            Bs_ + ((i64)(kfrag >> 3) * BNC + wn + j * 16 + l15) * 8));
+      const int rA = kfrag / CT;
+      const int cA = kfrag % CT;
+#pragma unroll
+      for (int kh = 0; kh < KH; ++kh) {
+        u32 d[NI][6];
+#pragma unroll
+        for (int i = 0; i < NI; ++i) {
+          int q = rA + kh;                       // window row 0..RH-1
+          int slot = sbase + q / R;
+          if (slot >= NSS) slot -= NSS;
+          const u16* base = &smem[0] + (i64)slot * SETE +
+              ((i64)(q % R) * BMC + wm + i * 16 + l15) * CTW + cA;
+          *reinterpret_cast<s16x8*>(&d[i][0]) =
+              *reinterpret_cast<const s16x8*>(base);
+          *reinterpret_cast<s16x4*>(&d[i][4]) =
+              *reinterpret_cast<const s16x4*>(base + 8);
+        }
+        if (kh == 0) {
+          asm volatile("" ::: "memory");
+          // ---- issue next stages under the MFMAs
+          if (tile + 2 < LT) {                  // regular: tile+2's new set
+            WGR_STAGE_A(sfree, n_r, c0_r, tile + 2 + S0 - 1);
+            sfree = sfree + 1 == NSS ? 0 : sfree + 1;
+            WGR_STAGE_B(bfree, n_r, c0_r, tile + 2);
+            bfree = bfree + 1 == NB ? 0 : bfree + 1;
+          } else if (tile == LT - 2 && !last_run) {
+            // seam batch: next run's prologue in the prologue's issue
+            // order (tile0' window + B0', then tile1' top + B1')
+#pragma unroll
+            for (int s = 0; s < S0; ++s) {
+              int sl = sfree + s;
+              while (sl >= NSS) sl -= NSS;
+              WGR_STAGE_A(sl, nn, nc0, s);
+            }
+            WGR_STAGE_B(bfree, nn, nc0, 0);
+            bfree = bfree + 1 == NB ? 0 : bfree + 1;
+            {
+              int sl = sfree + S0;
+              while (sl >= NSS) sl -= NSS;
+              WGR_STAGE_A(sl, nn, nc0, S0);
+            }
+            sfree = sfree + S0 + 1;
+            while (sfree >= NSS) sfree -= NSS;
+            WGR_STAGE_B(bfree, nn, nc0, 1);
+            bfree = bfree + 1 == NB ? 0 : bfree + 1;
+          }
+        }
+#pragma unroll
+        for (int kw = 0; kw < KW; ++kw) {
+          bf16x8 afr[NI];
+#pragma unroll
+          for (int i = 0; i < NI; ++i) {
+            u32 w[4];
+            if (kw % 2 == 0) {
+#pragma unroll
+              for (int q = 0; q < 4; ++q) w[q] = d[i][kw / 2 + q];
+            } else {
+#pragma unroll
+              for (int q = 0; q < 4; ++q)
+                w[q] = __builtin_amdgcn_alignbit(d[i][kw / 2 + 1 + q],
+                                                 d[i][kw / 2 + q], 16);
+            }
+            afr[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(&w[0]));
+          }
+          __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+          for (int i = 0; i < NI; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+              acc[kh * KW + kw][i][j] =
+                  __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                      afr[i], bfr[j], acc[kh * KW + kw][i][j], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+        }
+      }
+      sbase = sbase + 1 == NSS ? 0 : sbase + 1;   // window advances 1 set
+      bbase = bbase + 1 == NB ? 0 : bbase + 1;
+    }
+    // window of the next run starts at its prologue's first slot: the
+    // in-run advance moved sbase by LT-1... recompute from sfree:
+    // next run's sets occupy [sfree - (S0+1), sfree) -> first = that
+    if (!last_run) {
+      sbase = sfree - (S0 + 1);
+      while (sbase < 0) sbase += NSS;
+    }
+  }
+#undef WGR_STAGE_A
+#undef WGR_STAGE_B
+
+  float* slab = c32 + (i64)blockIdx.y * ((i64)KH * KW * Cin) * Cout;
+#pragma unroll
+  for (int tp = 0; tp < KH * KW; ++tp) {
+#pragma unroll
+    for (int i = 0; i < NI; ++i) {
+      int row_base = tp * Cin + ci0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int col = co0 + wn + j * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          slab[(i64)(row_base + r) * Cout + col] = acc[tp][i][j][r];
+      }
+    }
+  }
+}
+
 }  // namespace dvgg
