@@ -50,7 +50,8 @@ std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor&, const at::Tensor&);
 std::vector<at::Tensor> bn_fwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, double, bool,
                                const c10::optional<at::Tensor>&,
-                               const c10::optional<at::Tensor>&, double);
+                               const c10::optional<at::Tensor>&, double,
+                               const c10::optional<at::Tensor>&);
 std::vector<at::Tensor> bn_bwd(const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&, bool,
@@ -91,7 +92,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_fwd", &dvgg::maxpool_fwd);
   m.def("maxpool_bwd", &dvgg::maxpool_bwd);
   m.def("softmax_xent_fwd", &dvgg::softmax_xent_fwd);
-  m.def("bn_fwd", &dvgg::bn_fwd);
+  m.def("bn_fwd", &dvgg::bn_fwd, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("eps"), py::arg("relu"),
+        py::arg("run_mean") = c10::nullopt, py::arg("run_var") = c10::nullopt,
+        py::arg("momentum") = 0.9, py::arg("res") = c10::nullopt);
   m.def("bn_bwd", &dvgg::bn_bwd, py::arg("dy"), py::arg("x"),
         py::arg("gamma"), py::arg("mean"), py::arg("rstd"), py::arg("yr"),
         py::arg("relu"), py::arg("dg_out") = c10::nullopt,

@@ -143,8 +143,9 @@ __global__ void bn_stats_kernel(const float* __restrict__ sum,
   }
 }
 
-template <bool RELU>
+template <bool RELU, bool RES>
 __global__ void bn_apply_kernel(const u16* __restrict__ x,
+                                const u16* __restrict__ res,
                                 const float4* __restrict__ scale4,
                                 const float4* __restrict__ shift4,
                                 u16* __restrict__ y, i64 total8, int C) {
@@ -154,6 +155,8 @@ __global__ void bn_apply_kernel(const u16* __restrict__ x,
   int c8 = int(id % nch) * 8;
   i64 o = (id / nch) * C + c8;
   s16x8 xv = *reinterpret_cast<const s16x8*>(x + o);
+  s16x8 rv{};
+  if (RES) rv = *reinterpret_cast<const s16x8*>(res + o);
   float4 sA = scale4[c8 / 4], sB = scale4[c8 / 4 + 1];
   float4 hA = shift4[c8 / 4], hB = shift4[c8 / 4 + 1];
   float sc[8] = {sA.x, sA.y, sA.z, sA.w, sB.x, sB.y, sB.z, sB.w};
@@ -162,6 +165,7 @@ __global__ void bn_apply_kernel(const u16* __restrict__ x,
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float v = bf2f(u16(xv[j])) * sc[j] + sh[j];
+    if (RES) v += bf2f(u16(rv[j]));     // K18 fused: the residual add
     if (RELU) v = v > 0.f ? v : 0.f;
     out[j] = short(f2bf(v));
   }
@@ -172,7 +176,8 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
                                const at::Tensor& beta, double eps, bool relu,
                                const c10::optional<at::Tensor>& run_mean,
                                const c10::optional<at::Tensor>& run_var,
-                               double momentum) {
+                               double momentum,
+                               const c10::optional<at::Tensor>& res) {
   check_bf16(x, "x"); check_bf16(gamma, "gamma"); check_bf16(beta, "beta");
   int C = x.size(-1);
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
@@ -207,14 +212,20 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   i64 total8 = x.numel() / 8;
   auto sc4 = reinterpret_cast<const float4*>(scale.data_ptr<float>());
   auto sh4 = reinterpret_cast<const float4*>(shift.data_ptr<float>());
-  if (relu)
-    hipLaunchKernelGGL(bn_apply_kernel<true>, dim3(int((total8 + 255) / 256)),
-                       dim3(256), 0, s, bfp(x), sc4, sh4, bfp_mut(y), total8,
-                       C);
+  dim3 agrid(int((total8 + 255) / 256));
+  const u16* rp = res ? bfp(*res) : nullptr;
+  if (res && relu)
+    hipLaunchKernelGGL((bn_apply_kernel<true, true>), agrid, dim3(256), 0, s,
+                       bfp(x), rp, sc4, sh4, bfp_mut(y), total8, C);
+  else if (res)
+    hipLaunchKernelGGL((bn_apply_kernel<false, true>), agrid, dim3(256), 0,
+                       s, bfp(x), rp, sc4, sh4, bfp_mut(y), total8, C);
+  else if (relu)
+    hipLaunchKernelGGL((bn_apply_kernel<true, false>), agrid, dim3(256), 0,
+                       s, bfp(x), nullptr, sc4, sh4, bfp_mut(y), total8, C);
   else
-    hipLaunchKernelGGL(bn_apply_kernel<false>,
-                       dim3(int((total8 + 255) / 256)), dim3(256), 0, s,
-                       bfp(x), sc4, sh4, bfp_mut(y), total8, C);
+    hipLaunchKernelGGL((bn_apply_kernel<false, false>), agrid, dim3(256), 0,
+                       s, bfp(x), nullptr, sc4, sh4, bfp_mut(y), total8, C);
   return {y, mean, rstd};
 }
 

@@ -99,11 +99,13 @@ class BatchNorm(nn.Module):
         self.register_buffer("running_mean", torch.zeros(c))
         self.register_buffer("running_var", torch.ones(c))
 
-    def forward(self, x):
+    def forward(self, x, res=None):
         if self.training:
             return ops.batchnorm(x, self.weight, self.bias, self.eps,
                                  self.relu, self.running_mean,
-                                 self.running_var, self.momentum)
+                                 self.running_var, self.momentum, res=res)
         rstd = torch.rsqrt(self.running_var.to(x.dtype) + self.eps)
         y = (x - self.running_mean.to(x.dtype)) * rstd * self.weight + self.bias
+        if res is not None:
+            return torch.relu(y + res)
         return torch.relu(y) if self.relu else y

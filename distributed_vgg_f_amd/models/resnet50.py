@@ -37,12 +37,12 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         idt = x
-        y = self.bn1(self.conv1(x))
-        y = self.bn2(self.conv2(y))
-        y = self.bn3(self.conv3(y))
         if self.ds_conv is not None:
             idt = self.ds_bn(self.ds_conv(x))
-        return ops.add_relu(y, idt)
+        y = self.bn1(self.conv1(x))
+        y = self.bn2(self.conv2(y))
+        # K17+K18 fused: bn3's apply pass adds the residual and ReLUs
+        return self.bn3(self.conv3(y), res=idt)
 
 
 class ResNet50(nn.Module):

@@ -66,12 +66,18 @@ def softmax_xent(logits, labels):
 
 
 def batchnorm(x, gamma, beta, eps: float = 1e-5, relu: bool = False,
-              run_mean=None, run_var=None, momentum: float = 0.9):
+              run_mean=None, run_var=None, momentum: float = 0.9, res=None):
     if x.is_cuda:
+        if res is not None:
+            from .autograd import BNAddReluFunction
+            return BNAddReluFunction.apply(x, gamma, beta, res, eps,
+                                           run_mean, run_var, momentum)
         from .autograd import BatchNormFunction
         return BatchNormFunction.apply(x, gamma, beta, eps, relu,
                                        run_mean, run_var, momentum)
-    y = cpu_ref.batchnorm(x, gamma, beta, eps, relu)
+    y = cpu_ref.batchnorm(x, gamma, beta, eps, relu and res is None)
+    if res is not None:
+        y = cpu_ref.add_relu(y, res)
     if run_mean is not None:
         with torch.no_grad():
             m = x.float().mean(dim=(0, 1, 2))

@@ -185,6 +185,31 @@ class BatchNormFunction(torch.autograd.Function):
         return dx, dgamma, dbeta, None, None, None, None, None
 
 
+class BNAddReluFunction(torch.autograd.Function):
+    """K17+K18 fused: z = relu(bn(x) + res) in one apply pass (the
+    separate add_relu kernel read+wrote the whole block output)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, res, eps: float,
+                run_mean=None, run_var=None, momentum: float = 0.9):
+        z, mean, rstd = _ext().bn_fwd(x, gamma, beta, eps, True,
+                                      run_mean, run_var, momentum, res)
+        ctx.save_for_backward(x, gamma, mean, rstd, z)
+        ctx.gg = getattr(gamma, "_dvgg_gbuf", None)
+        ctx.gb = getattr(beta, "_dvgg_gbuf", None)
+        return z
+
+    @staticmethod
+    def backward(ctx, dz):
+        x, gamma, mean, rstd, z = ctx.saved_tensors
+        # da feeds BOTH the bn input grad chain and the residual branch
+        da = _ext().relu_bwd(dz.contiguous(), z)
+        dx, dgamma, dbeta = _ext().bn_bwd(da, x, gamma, mean, rstd,
+                                          torch.empty(0, device=x.device),
+                                          False, ctx.gg, ctx.gb)
+        return dx, dgamma, dbeta, da, None, None, None, None
+
+
 class AddReluFunction(torch.autograd.Function):
     """K18 (ResNet-50): y = relu(a + b)."""
 

@@ -424,3 +424,35 @@ def test_cached_wd_refresh_1x1s2():
     # dgrad must reflect the NEW weights (stale cache would reuse old wD)
     dx_ref = ext.conv2d_dgrad(dy, w, H, W, s, p)
     check(dx2, dx_ref.float().cpu(), 1e-3, "cached_wd_refresh")
+
+
+def test_bn_add_relu_fused():
+    """K17+K18 fusion: z = relu(bn(x) + res) in one apply pass must match
+    the composed CPU reference in forward AND all gradients."""
+    from distributed_vgg_f_amd import ops
+    torch.manual_seed(93)
+    N, H, W, C = 4, 8, 8, 64
+    x = rnd(N, H, W, C, scale=0.7)
+    res = rnd(N, H, W, C, scale=0.7)
+    g = rnd(C, scale=0.3) + 1.0
+    b = rnd(C, scale=0.3)
+
+    xg = x.to(DEV).requires_grad_(True)
+    rg = res.to(DEV).requires_grad_(True)
+    gg = g.to(DEV).requires_grad_(True)
+    bg = b.to(DEV).requires_grad_(True)
+    z = ops.batchnorm(xg, gg, bg, 1e-5, False, res=rg)
+    z.float().square().sum().backward()
+
+    xc = x.float().requires_grad_(True)
+    rc = res.float().requires_grad_(True)
+    gc = g.float().requires_grad_(True)
+    bc = b.float().requires_grad_(True)
+    zc = ops.batchnorm(xc, gc, bc, 1e-5, False, res=rc)
+    zc.square().sum().backward()
+
+    check(z, zc.detach(), 0.05, "bnaddrelu_fwd")
+    check(xg.grad, xc.grad, 0.08, "bnaddrelu_dx")
+    check(rg.grad, rc.grad, 0.05, "bnaddrelu_dres")
+    check(gg.grad, gc.grad, 0.05, "bnaddrelu_dgamma")
+    check(bg.grad, bc.grad, 0.05, "bnaddrelu_dbeta")
