@@ -42,7 +42,8 @@ def main():
                  precision="bf16" if use_gpu else "fp32",
                  log_every=0, bucket_mb=args.bucket_mb,
                  overlap_comm=bool(args.overlap_comm),
-                 use_graph=bool(args.use_graph))
+                 use_graph=bool(args.use_graph),
+                 phase_timers=not bool(args.use_graph))
     ctx = init_distributed(cfg.backend)
     world = ctx.world_size
     if world > 1 and args.gpus != world:
@@ -114,6 +115,10 @@ def main():
             "ms_per_step_min": round(min(chunk_ms), 3) if chunk_ms else None,
             "ms_per_step_mean": round(sum(chunk_ms) / len(chunk_ms), 3)
             if chunk_ms else None,
+            # last-step hipEvent phase times; ms_ar is EXPOSED all-reduce
+            # (compute-stream stall on the bucket reductions): with the
+            # backward overlap working it stays near 0 at any world size
+            "phases": {k: round(v, 3) for k, v in t.timers.collect().items()},
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
