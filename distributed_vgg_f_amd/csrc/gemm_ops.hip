@@ -779,6 +779,33 @@ static void launch_gemm_tn_t(const PlainGather& ga, const PlainGather& gb,
 static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
                            at::Tensor* outB, at::Tensor* c32, int M, int N,
                            int K, int splitk) {
+  if (env_int("DVGG_TNPACK", 1)) {
+    // packed-b64 transpose staging (gemm_tn_kernel): 4x fewer LDS write
+    // instructions than the scalar TN mode of gemm_kernel
+    auto s = cur_stream();
+    int tiles4 = ceil_div(M, BM) * ceil_div(N, BN);
+    int tiles2 = ceil_div(M, BM) * ceil_div(N, 64);
+    if (splitk > 1) {
+      if (N <= nfrag_threshold())
+        hipLaunchKernelGGL((gemm_tn_kernel<true, 2>), dim3(tiles2, splitk),
+                           dim3(THREADS), 0, s, ga, gb, nullptr,
+                           c32->data_ptr<float>(), M, N, K, 0, K);
+      else
+        hipLaunchKernelGGL((gemm_tn_kernel<true, 4>), dim3(tiles4, splitk),
+                           dim3(THREADS), 0, s, ga, gb, nullptr,
+                           c32->data_ptr<float>(), M, N, K, 0, K);
+    } else {
+      if (N <= nfrag_threshold())
+        hipLaunchKernelGGL((gemm_tn_kernel<false, 2>), dim3(tiles2),
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, M, N, K, 0, K);
+      else
+        hipLaunchKernelGGL((gemm_tn_kernel<false, 4>), dim3(tiles4),
+                           dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
+                           nullptr, M, N, K, 0, K);
+    }
+    return;
+  }
   if (N <= nfrag_threshold())
     launch_gemm_tn_t<2>(ga, gb, outB, c32, M, N, K, splitk);
   else

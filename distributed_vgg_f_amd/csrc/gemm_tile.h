@@ -828,4 +828,133 @@ __global__ __launch_bounds__(THREADS) void gemm_glds3_kernel(
   }
 }
 
+
+// ------------------------------------------------- packed-TN GEMM (K9)
+// Both operands in [reduction][col] layout (fc and 1x1-conv wgrad:
+// dw[M][N] = sum_k A[k][m] * B[k][n]). The generic gemm_kernel's TN mode
+// transposes during staging with EIGHT scalar ds_writes per 8-elem chunk
+// and measured LDS-WRITE-BOUND (41 TF on ResNet layer1 1x1 wgrads, 5x off
+// the bandwidth roofline). Here each staging thread owns FOUR consecutive
+// reduction rows of one 8-column group and writes the transpose as packed
+// ds_write_b64 (one per column, 4x fewer write instructions); the slot
+// map keeps a 16-lane write group on 8 distinct bank pairs (2-way).
+// Geometry: 256 threads, 4 waves 2x2, BM=128 x BN_T=32*NFRAG, BK=32.
+template <bool SPLITK, int NFRAG>
+__global__ __launch_bounds__(THREADS) void gemm_tn_kernel(
+    PlainGather ga, PlainGather gb, u16* __restrict__ C,
+    float* __restrict__ c32, int M, int N, int K, int k_lo, int k_hi) {
+  constexpr int BN_T = 32 * NFRAG;
+  __shared__ u16 As[2][BM * LDS_STRIDE];
+  __shared__ u16 Bs[2][BN_T * LDS_STRIDE];
+
+  const int tiles_n = (N + BN_T - 1) / BN_T;
+  const int tiles_m = (M + BM - 1) / BM;
+  u32 wg = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int tm = wg / tiles_n, tn = wg % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN_T;
+  if (SPLITK) {
+    int nchunk = gridDim.y;
+    int kt = (K + BK - 1) / BK;
+    int span = (kt + nchunk - 1) / nchunk * BK;
+    k_lo = blockIdx.y * span;
+    k_hi = min(K, k_lo + span);
+    if (k_lo >= k_hi) return;
+  }
+  const int t = threadIdx.x;
+  const int lane = t & 63, wave = t >> 6;
+  const int wm = (wave >> 1) * 64, wn = (wave & 1) * (BN_T / 2);
+
+  // staging slots: (cols/8) x (BK/4) k-quads; A on t<128, B above
+  constexpr int KQ = BK / 4;               // 8
+  constexpr int ASL = (BM / 8) * KQ;       // 128
+  constexpr int BSL = (BN_T / 8) * KQ;     // 128 / 64
+  const bool is_a = t < ASL;
+  const int slot = is_a ? t : t - ASL;
+  const bool active = is_a || slot < BSL;
+  const int kq = slot % KQ, col8 = slot / KQ;
+  const int scol = (is_a ? m0 : n0) + col8 * 8;
+  const int sk0 = kq * 4;
+
+  f32x4 acc[4][NFRAG];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = (k_hi - k_lo + BK - 1) / BK;
+  s16x8 r[4];
+  if (nt > 0 && active) {
+#pragma unroll
+    for (int q = 0; q < 4; ++q)
+      r[q] = is_a ? ga.load8(ga.prep(k_lo + sk0 + q), scol)
+                  : gb.load8(gb.prep(k_lo + sk0 + q), scol);
+  }
+
+  int p = 0;
+  for (int tile = 0; tile < nt; ++tile) {
+    if (active) {
+      u16* base = is_a ? &As[p][(i64)(col8 * 8) * LDS_STRIDE + sk0]
+                       : &Bs[p][(i64)(col8 * 8) * LDS_STRIDE + sk0];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        union { unsigned long long v; u16 e[4]; } pk;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) pk.e[q] = u16(r[q][j]);
+        *reinterpret_cast<unsigned long long*>(base + (i64)j * LDS_STRIDE) =
+            pk.v;
+      }
+    }
+    __syncthreads();
+    if (tile + 1 < nt && active) {   // next tile's loads under the MFMAs
+      int kb = k_lo + (tile + 1) * BK + sk0;
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        r[q] = is_a ? ga.load8(ga.prep(kb + q), scol)
+                    : gb.load8(gb.prep(kb + q), scol);
+    }
+    const int kfrag = (lane >> 4) * 8;
+    bf16x8 af[4], bf[NFRAG];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wm + i * 16 + (lane & 15);
+      af[i] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          &As[p][row * LDS_STRIDE + kfrag]));
+    }
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = wn + j * 16 + (lane & 15);
+      bf[j] = as_bf16x8(*reinterpret_cast<const s16x8*>(
+          &Bs[p][col * LDS_STRIDE + kfrag]));
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < NFRAG; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf[j], acc[i][j], 0, 0, 0);
+    // ONE barrier per tile: tile t+2's writes to this buffer are fenced
+    // from this tile's readers by tile t+1's barrier (cf. gemm_kernel)
+    p ^= 1;
+  }
+
+  // epilogue (plain or split-K atomics, EPI=0 only — wgrad has no bias)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row_base = m0 + wm + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < NFRAG; ++j) {
+      int col = n0 + wn + j * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int row = row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr];
+        if (SPLITK) atomicAdd(&c32[(i64)row * N + col], v);
+        else C[(i64)row * N + col] = f2bf(v);
+      }
+    }
+  }
+}
+
 }  // namespace dvgg
