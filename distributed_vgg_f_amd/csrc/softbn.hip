@@ -89,9 +89,6 @@ std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor& logits,
 // Pass 2: normalize (+affine, optional fused ReLU); saves mean/rstd fp32.
 // partials land in 256 slots per channel (slot = blockIdx.y & 255):
 // direct atomics onto C floats from thousands of blocks serialize.
-// each blockIdx.y owns its own partial slot: plain stores, and the
-// partial buffers need no zero-fill (they were at::zeros + atomics
-// before — ~150 Fill launches/step on ResNet-50)
 __global__ void bn_reduce_kernel(const u16* __restrict__ x,
                                  float* __restrict__ sum,
                                  float* __restrict__ sumsq, i64 R, int C,
@@ -107,8 +104,9 @@ __global__ void bn_reduce_kernel(const u16* __restrict__ x,
     s0 += v0; q0 += v0 * v0; s1 += v1; q1 += v1 * v1;
   }
   for (; r < r1; ++r) { float v = bf2f(x[r * C + c]); s0 += v; q0 += v * v; }
-  sum[(i64)blockIdx.y * C + c] = s0 + s1;
-  sumsq[(i64)blockIdx.y * C + c] = q0 + q1;
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sum[(i64)slot * C + c], s0 + s1);
+  atomicAdd(&sumsq[(i64)slot * C + c], q0 + q1);
 }
 
 __global__ void bn_stats_kernel(const float* __restrict__ sum,
@@ -121,12 +119,12 @@ __global__ void bn_stats_kernel(const float* __restrict__ sum,
                                 float* __restrict__ shift,
                                 float* __restrict__ run_mean,
                                 float* __restrict__ run_var, i64 R, int C,
-                                int nslots, float eps, float momentum) {
+                                float eps, float momentum) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float sa = 0.f, qa = 0.f;
 #pragma unroll 8
-  for (int t = 0; t < nslots; ++t) {
+  for (int t = 0; t < 256; ++t) {
     sa += sum[(i64)t * C + c];
     qa += sumsq[(i64)t * C + c];
   }
@@ -185,6 +183,8 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = x.numel() / C;
   auto f32 = x.options().dtype(at::kFloat);
+  auto part = at::zeros({2, 256, C}, f32);   // one fill for both partials
+  auto sum = part[0], sumsq = part[1];
   auto stats = at::empty({4, C}, f32);       // mean/rstd/scale/shift
   auto mean = stats[0], rstd = stats[1], scale = stats[2], shift = stats[3];
   auto y = at::empty_like(x);
@@ -196,12 +196,8 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   // (coalesced 2 B/lane across a 256-lane channel window) and the wide
   // version only multiplied end-of-thread atomics. Kept scalar.
   int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
-  rpb = std::max(rpb, int(ceil_div(R, (i64)256)));  // <= 256 slots
-  int ny = int(ceil_div(R, (i64)rpb));
-  auto part = at::empty({2, ny, C}, f32);   // slot-per-block: no zero-fill
-  auto sum = part[0], sumsq = part[1];
   auto s = cur_stream();
-  dim3 rgrid(ceil_div(C, 256), ny);
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, (i64)rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(), R, C,
                      rpb);
@@ -212,7 +208,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
                      bfp(gamma), bfp(beta), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(), rmp, rvp,
-                     R, C, ny, float(eps), float(momentum));
+                     R, C, float(eps), float(momentum));
   i64 total8 = x.numel() / 8;
   auto sc4 = reinterpret_cast<const float4*>(scale.data_ptr<float>());
   auto sh4 = reinterpret_cast<const float4*>(shift.data_ptr<float>());
@@ -268,8 +264,9 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
     float h = (bf2f(x[r * C + c]) - m) * rs;
     a0 += d; b0 += d * h;
   }
-  sdy[(i64)blockIdx.y * C + c] = a0 + a1;
-  sdyx[(i64)blockIdx.y * C + c] = b0 + b1;
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sdy[(i64)slot * C + c], a0 + a1);
+  atomicAdd(&sdyx[(i64)slot * C + c], b0 + b1);
 }
 
 // One pass over the 256-way partials: emits dgamma/dbeta and the three
@@ -286,12 +283,12 @@ __global__ void bn_bwd_stats_kernel(const float* __restrict__ sdyp,
                                     float* __restrict__ cD,
                                     u16* __restrict__ dgamma,
                                     u16* __restrict__ dbeta, int C,
-                                    int nslots, float invR) {
+                                    float invR) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float a = 0.f, b = 0.f;
 #pragma unroll 8
-  for (int t = 0; t < nslots; ++t) {
+  for (int t = 0; t < 256; ++t) {
     a += sdyp[(i64)t * C + c];
     b += sdyxp[(i64)t * C + c];
   }
@@ -349,18 +346,16 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = dy.numel() / C;
   auto f32 = dy.options().dtype(at::kFloat);
+  auto bpart = at::zeros({2, 256, C}, f32);
+  auto sdyp = bpart[0], sdyxp = bpart[1];
   auto coef = at::empty({3, C}, f32);
   auto cA = coef[0], cB = coef[1], cD = coef[2];
   auto dx = at::empty_like(dy);
   auto dgamma = dg_out ? *dg_out : at::empty({C}, dy.options());
   auto dbeta = db_out ? *db_out : at::empty({C}, dy.options());
   int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
-  rpb = std::max(rpb, int(ceil_div(R, (i64)256)));  // <= 256 slots
-  int ny = int(ceil_div(R, (i64)rpb));
-  auto bpart = at::empty({2, ny, C}, f32);  // slot-per-block: no zero-fill
-  auto sdyp = bpart[0], sdyxp = bpart[1];
   auto s = cur_stream();
-  dim3 rgrid(ceil_div(C, 256), ny);
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, (i64)rpb));
   const u16* yp = relu ? bfp(yr) : nullptr;
   if (relu)
     hipLaunchKernelGGL(bn_bwd_reduce_kernel<true>, rgrid, dim3(256), 0, s,
@@ -377,7 +372,7 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
                      bfp(gamma), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(), cA.data_ptr<float>(),
                      cB.data_ptr<float>(), cD.data_ptr<float>(),
-                     bfp_mut(dgamma), bfp_mut(dbeta), C, ny,
+                     bfp_mut(dgamma), bfp_mut(dbeta), C,
                      float(1.0 / double(R)));
   i64 total8 = dy.numel() / 8;
   auto a4 = reinterpret_cast<const float4*>(cA.data_ptr<float>());
