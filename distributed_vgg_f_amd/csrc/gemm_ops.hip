@@ -776,10 +776,14 @@ static void launch_gemm_tn_t(const PlainGather& ga, const PlainGather& gb,
   }
 }
 
-static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
+template <class GA, class GB>
+static void launch_gemm_tn(const GA& ga, const GB& gb,
                            at::Tensor* outB, at::Tensor* c32, int M, int N,
                            int K, int splitk) {
-  if (env_int("DVGG_TNPACK", 1)) {
+  constexpr bool PLAIN =
+      std::is_same<GA, PlainGather>::value &&
+      std::is_same<GB, PlainGather>::value;
+  if (!PLAIN || env_int("DVGG_TNPACK", 1)) {
     // packed-b64 transpose staging (gemm_tn_kernel): 4x fewer LDS write
     // instructions than the scalar TN mode of gemm_kernel
     auto s = cur_stream();
@@ -787,29 +791,31 @@ static void launch_gemm_tn(const PlainGather& ga, const PlainGather& gb,
     int tiles2 = ceil_div(M, BM) * ceil_div(N, 64);
     if (splitk > 1) {
       if (N <= nfrag_threshold())
-        hipLaunchKernelGGL((gemm_tn_kernel<true, 2>), dim3(tiles2, splitk),
-                           dim3(THREADS), 0, s, ga, gb, nullptr,
-                           c32->data_ptr<float>(), M, N, K, 0, K);
+        hipLaunchKernelGGL((gemm_tn_kernel<true, 2, GA, GB>),
+                           dim3(tiles2, splitk), dim3(THREADS), 0, s, ga, gb,
+                           nullptr, c32->data_ptr<float>(), M, N, K, 0, K);
       else
-        hipLaunchKernelGGL((gemm_tn_kernel<true, 4>), dim3(tiles4, splitk),
-                           dim3(THREADS), 0, s, ga, gb, nullptr,
-                           c32->data_ptr<float>(), M, N, K, 0, K);
+        hipLaunchKernelGGL((gemm_tn_kernel<true, 4, GA, GB>),
+                           dim3(tiles4, splitk), dim3(THREADS), 0, s, ga, gb,
+                           nullptr, c32->data_ptr<float>(), M, N, K, 0, K);
     } else {
       if (N <= nfrag_threshold())
-        hipLaunchKernelGGL((gemm_tn_kernel<false, 2>), dim3(tiles2),
+        hipLaunchKernelGGL((gemm_tn_kernel<false, 2, GA, GB>), dim3(tiles2),
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, M, N, K, 0, K);
       else
-        hipLaunchKernelGGL((gemm_tn_kernel<false, 4>), dim3(tiles4),
+        hipLaunchKernelGGL((gemm_tn_kernel<false, 4, GA, GB>), dim3(tiles4),
                            dim3(THREADS), 0, s, ga, gb, bfp_mut(*outB),
                            nullptr, M, N, K, 0, K);
     }
     return;
   }
-  if (N <= nfrag_threshold())
-    launch_gemm_tn_t<2>(ga, gb, outB, c32, M, N, K, splitk);
-  else
-    launch_gemm_tn_t<4>(ga, gb, outB, c32, M, N, K, splitk);
+  if constexpr (PLAIN) {
+    if (N <= nfrag_threshold())
+      launch_gemm_tn_t<2>(ga, gb, outB, c32, M, N, K, splitk);
+    else
+      launch_gemm_tn_t<4>(ga, gb, outB, c32, M, N, K, splitk);
+  }
 }
 
 // wgrad-dedicated launcher: BM=256/512-thread tile when Cout >= 128
@@ -1221,6 +1227,27 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     wgrad_glds(x, dyc, KH, KW, pad, dw2d, Cout);
     return dw;
   }
+  if (KH == 1 && KW == 1 && pad == 0 && Cin % 8 == 0 && stride > 1) {
+    // strided 1x1 (ResNet downsample): TN GEMM over x sampled at
+    // (s*oh, s*ow) — see the conv2d_bwd fast path
+    int M1 = N * OH * OW;
+    int Mred = ((M1 + BK - 1) / BK) * BK;
+    auto dy2 = dy.contiguous().view({M1, Cout});
+    StridedRowGather ga{bfp(x), M1, Cin, H, W, stride,
+                        MDiv(u32(OH * OW)), MDiv(u32(OW))};
+    PlainGather gb{bfp(dy2), M1, Cout, Cout};
+    int tiles1 = ceil_div(Cin, BM) * ceil_div(Cout, BN);
+    int sk1 = pick_splitk(tiles1, Mred);
+    auto dwt = dw2d.view({Cin, Cout});
+    if (sk1 > 1) {
+      auto c32w = at::zeros({Cin, Cout}, x.options().dtype(at::kFloat));
+      launch_gemm_tn(ga, gb, nullptr, &c32w, Cin, Cout, Mred, sk1);
+      run_finalize(c32w, dwt, nullptr, 0, Cin, Cout);
+    } else {
+      launch_gemm_tn(ga, gb, &dwt, nullptr, Cin, Cout, Mred, 1);
+    }
+    return dw;
+  }
   // Padding Cin<8 to 8 was tried and measured 3.4x SLOWER for wgrad
   // (zero-channel waste x scalar stride-4 staging); the direct NHWC
   // gather below stays for Cin < 8.
@@ -1291,16 +1318,44 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   int M = N * OH * OW, Kfull = KH * KW * Cin;
   auto s = cur_stream();
 
-  if (KH == 1 && KW == 1 && stride == 1 && pad == 0 && Cin % 8 == 0) {
-    // 1x1/s1 conv backward == FC backward on flattened spatial dims:
-    // no transposed dy staging, no implicit-GEMM gathers
+  if (KH == 1 && KW == 1 && pad == 0 && Cin % 8 == 0) {
+    // 1x1 conv backward == FC backward on flattened spatial dims (no
+    // transposed dy staging). stride>1 (ResNet downsample): the wgrad
+    // reduction samples x at (s*oh, s*ow) via StridedRowGather — the
+    // implicit-GEMM wgrad path measured ~5x slower on these; dgrad
+    // keeps the implicit path (dx scatter is strided).
     auto dy2 = dy.view({M, Cout});
     if (relu) dy2 = relu_bwd(dy2, y.view({M, Cout}));
     at::Tensor dx1;
-    if (need_dx)
-      dx1 = linear_dgrad(dy2, w.view({Cin, Cout})).view({N, H, W, Cin});
-    auto dw1 = linear_wgrad(x.view({M, Cin}), dy2, dw_out)
-        .view({1, 1, Cin, Cout});
+    if (need_dx) {
+      if (stride == 1)
+        dx1 = linear_dgrad(dy2, w.view({Cin, Cout})).view({N, H, W, Cin});
+      else
+        dx1 = conv2d_dgrad(dy2.view({N, OH, OW, Cout}), w, H, W, stride,
+                           pad, cache_w);
+    }
+    at::Tensor dw1;
+    if (stride == 1) {
+      dw1 = linear_wgrad(x.view({M, Cin}), dy2, dw_out)
+          .view({1, 1, Cin, Cout});
+    } else {
+      auto dwt = dw_out ? dw_out->view({Cin, Cout})
+                        : at::empty({Cin, Cout}, x.options());
+      int Mred = ((M + BK - 1) / BK) * BK;
+      StridedRowGather ga{bfp(x), M, Cin, H, W, stride,
+                          MDiv(u32(OH * OW)), MDiv(u32(OW))};
+      PlainGather gb{bfp(dy2), M, Cout, Cout};
+      int tiles = ceil_div(Cin, BM) * ceil_div(Cout, BN);
+      int sk = pick_splitk(tiles, Mred);
+      if (sk > 1) {
+        auto c32w = at::zeros({Cin, Cout}, x.options().dtype(at::kFloat));
+        launch_gemm_tn(ga, gb, nullptr, &c32w, Cin, Cout, Mred, sk);
+        run_finalize(c32w, dwt, nullptr, 0, Cin, Cout);
+      } else {
+        launch_gemm_tn(ga, gb, &dwt, nullptr, Cin, Cout, Mred, 1);
+      }
+      dw1 = dwt.view({1, 1, Cin, Cout});
+    }
     at::Tensor db1;
     if (need_db) db1 = bias_grad(dy2, db_out);
     return {dx1, dw1, db1};

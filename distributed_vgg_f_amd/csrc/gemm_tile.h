@@ -829,6 +829,28 @@ __global__ __launch_bounds__(THREADS) void gemm_glds3_kernel(
 }
 
 
+// TN A-side gather for 1x1 STRIDED conv wgrad: reduction row m =
+// (n, oh, ow) reads x at (n, oh*s, ow*s) — a strided sampling of the
+// input. Used with gemm_tn_kernel (cols = ci are contiguous as in x).
+struct StridedRowGather {
+  const u16* x;
+  int Mred, Cin, H, W, stride;
+  MDiv dOHOW, dOW;
+  struct Row { const u16* rp; bool valid; };
+  DEV Row prep(int m) const {
+    u32 n = dOHOW.div(m), r = dOHOW.mod(m, n);
+    u32 oh = dOW.div(r), ow = dOW.mod(r, oh);
+    return {x + (((i64)n * H + (i64)oh * stride) * W + (i64)ow * stride) *
+                    Cin,
+            m < Mred};
+  }
+  DEV s16x8 load8(const Row& r, int k) const {
+    if (r.valid && k < Cin)
+      return *reinterpret_cast<const s16x8*>(r.rp + k);
+    return zero8();
+  }
+};
+
 // ------------------------------------------------- packed-TN GEMM (K9)
 // Both operands in [reduction][col] layout (fc and 1x1-conv wgrad:
 // dw[M][N] = sum_k A[k][m] * B[k][n]). The generic gemm_kernel's TN mode
@@ -839,9 +861,10 @@ __global__ __launch_bounds__(THREADS) void gemm_glds3_kernel(
 // ds_write_b64 (one per column, 4x fewer write instructions); the slot
 // map keeps a 16-lane write group on 8 distinct bank pairs (2-way).
 // Geometry: 256 threads, 4 waves 2x2, BM=128 x BN_T=32*NFRAG, BK=32.
-template <bool SPLITK, int NFRAG>
+template <bool SPLITK, int NFRAG, class GA = PlainGather,
+          class GB = PlainGather>
 __global__ __launch_bounds__(THREADS) void gemm_tn_kernel(
-    PlainGather ga, PlainGather gb, u16* __restrict__ C,
+    GA ga, GB gb, u16* __restrict__ C,
     float* __restrict__ c32, int M, int N, int K, int k_lo, int k_hi) {
   constexpr int BN_T = 32 * NFRAG;
   __shared__ u16 As[2][BM * LDS_STRIDE];
