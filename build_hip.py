@@ -21,7 +21,8 @@ OUT = os.path.join(ROOT, "distributed_vgg_f_amd", "_hipops.so")
 
 SOURCES = ["gemm_ops.hip", "mem.hip", "lrnpool.hip", "softbn.hip",
            "hipops.cpp"]
-HEADERS = ["common.h", "gemm_tile.h", "gemm8.h"]
+HEADERS = ["common.h", "gemm_tile.h", "gemm8.h", "wgrad_fused.h",
+           "conv_ring.h"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

@@ -10,6 +10,7 @@
 #include "gemm_tile.h"
 #include "gemm8.h"
 #include "wgrad_fused.h"
+#include "conv_ring.h"
 
 namespace dvgg {
 
@@ -155,7 +156,7 @@ template <class F>
 static at::Tensor cached_weight_form(int kind, bool on, const at::Tensor& w,
                                      F&& make) {
   if (!on || stream_capturing()) return make(c10::nullopt);
-  static std::unordered_map<const void*, WTEntry> cache[3];
+  static std::unordered_map<const void*, WTEntry> cache[4];
   auto& c = cache[kind];
   std::vector<i64> sizes(w.sizes().begin(), w.sizes().end());
   auto it = c.find(w.data_ptr());
@@ -1113,6 +1114,55 @@ static void wgrad_glds(const at::Tensor& x, const at::Tensor& dym, int KH,
   run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
 }
 
+// --------------- ring direct conv (conv_ring.h) -----------------------
+static bool conv_ring_ok(int KH, int KW, int stride, int Ci, int Co,
+                         int OH) {
+  return stride == 1 && KH == KW && (KH == 3 || KH == 5) && Ci == 64 &&
+         Co % 64 == 0 && OH >= 8 && env_int("DVGG_CONV_RING", 1);
+}
+
+static void conv_ring(const at::Tensor& x, const at::Tensor& wT,
+                      const at::Tensor* bias, at::Tensor& y, int KH, int KW,
+                      int pad, int epi) {
+  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int OH = y.size(1), OW = y.size(2), Cout = y.size(3);
+  int OWp2 = (OW + 31) / 32 * 32;
+  int SEG = OWp2 / 32;
+  int LT = (OH + 3) / 4;                 // TR = 4
+  int runs = N * SEG;
+  int wavesn = Cout % 128 == 0 ? 4 : 2;
+  int blocks = runs * (Cout / (32 * wavesn));
+  const u16* zp = zero_page(x.options());
+  const u16* bp = bias ? bfp(*bias) : nullptr;
+  auto s = cur_stream();
+#define CRL(WN, KHV, KWV, EPIV)                                                hipLaunchKernelGGL((conv_ring_kernel<WN, KHV, KWV, EPIV>), dim3(blocks),                        dim3(512), 0, s, bfp(x), bfp(wT), bp, bfp_mut(y), zp,                        N, H, W, Cin, Cout, OH, OW, OWp2, pad, runs, LT)
+#define CRL_EPI(WN, KHV, KWV)                                                  do {                                                                           if (epi == 0) CRL(WN, KHV, KWV, 0);                                          else if (epi == 1) CRL(WN, KHV, KWV, 1);                                     else if (epi == 2) CRL(WN, KHV, KWV, 2);                                     else CRL(WN, KHV, KWV, 3);                                                 } while (0)
+  if (wavesn == 4) {
+    if (KH == 3) CRL_EPI(4, 3, 3); else CRL_EPI(4, 5, 5);
+  } else {
+    if (KH == 3) CRL_EPI(2, 3, 3); else CRL_EPI(2, 5, 5);
+  }
+#undef CRL_EPI
+#undef CRL
+}
+
+
+// flipped+transposed weight for the ring dgrad: wFT[ci][(kh,kw,co)] =
+// w[KH-1-kh][KW-1-kw][ci][co] (dx of a stride-1 conv IS a conv of dy with
+// this kernel at pad' = KH-1-pad). Cached per optimizer step (kind 3).
+static at::Tensor cached_wft(bool cache_w, const at::Tensor& w, int KH,
+                             int KW, int Cin, int Cout) {
+  return cached_weight_form(3, cache_w, w,
+                            [&](c10::optional<at::Tensor> prev) {
+    auto f = w.flip({0, 1}).permute({2, 0, 1, 3});   // [Cin][KH][KW][Cout]
+    if (prev) {
+      prev->view({Cin, KH, KW, Cout}).copy_(f);
+      return *prev;
+    }
+    return f.contiguous().view({Cin, (i64)KH * KW * Cout});
+  });
+}
+
 // ------------------------------------------------------------------ conv
 static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
                            int& OH, int& OW) {
@@ -1158,6 +1208,10 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
     if (prev) { transpose2d_into(w2d, *prev); return *prev; }
     return transpose2d(w2d);
   });  // [Cout][K]
+  if (conv_ring_ok(KH, KW, stride, Cin, Cout, OH)) {
+    conv_ring(x, wT, bias ? &*bias : nullptr, y, KH, KW, pad, epi);
+    return y;
+  }
   ConvFwdGather ga{bfp(x), M, H, W, Cin, stride, pad,
                    MDiv(u32(OH * OW)), MDiv(u32(OW)), MDiv(u32(Cin)),
                    MDiv(u32(KW))};
@@ -1187,6 +1241,11 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
               "dgrad for Cin<8 first layers is unused by design");
   int M = N * H * W, K = KH * KW * Cout;
   auto dx = at::empty({N, H, W, Cin}, dy.options());
+  if (conv_ring_ok(KH, KW, stride, Cout, Cin, H)) {
+    auto wFT = cached_wft(cache_w, w, KH, KW, Cin, Cout);
+    conv_ring(dy, wFT, nullptr, dx, KH, KW, KH - 1 - pad, 0);
+    return dx;
+  }
   // Bt[ci][(kh,kw,co)] = w[kh][kw][ci][co]: permute HWIO -> [Cin][KhKw*Cout]
   auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
     auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
@@ -1410,6 +1469,12 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     TORCH_CHECK(Cout % 8 == 0 && Cin % 8 == 0, "dgrad channel alignment");
     int K = KH * KW * Cout;
     dx = at::empty({N, H, W, Cin}, dy.options());
+    if (conv_ring_ok(KH, KW, stride, Cout, Cin, H)) {
+      auto wFT = cached_wft(cache_w, w, KH, KW, Cin, Cout);
+      conv_ring(dym.contiguous(), wFT, nullptr, dx, KH, KW, KH - 1 - pad, 0);
+      goto dgrad_done;
+    }
+    {
     auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
       auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
       if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
@@ -1434,6 +1499,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     } else {
       launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, Mx, Cin, K, 1);
     }
+    }
+    dgrad_done:;
   }
 
   // ---- wgrad
