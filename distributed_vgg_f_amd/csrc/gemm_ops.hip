@@ -1117,50 +1117,32 @@ static void wgrad_glds(const at::Tensor& x, const at::Tensor& dym, int KH,
 // --------------- ring direct conv (conv_ring.h) -----------------------
 static bool conv_ring_ok(int KH, int KW, int stride, int Ci, int Co,
                          int OH) {
-  return stride == 1 && KH == KW && (KH == 3 || KH == 5) && Ci == 64 &&
-         Co % 64 == 0 && OH >= 8 && env_int("DVGG_CONV_RING", 1);
+  return stride == 1 && KH == 3 && KW == 3 && Ci == 64 && Co % 64 == 0 &&
+         OH >= 8 && env_int("DVGG_CONV_RING", 1);
 }
 
 static void conv_ring(const at::Tensor& x, const at::Tensor& wT,
                       const at::Tensor* bias, at::Tensor& y, int KH, int KW,
                       int pad, int epi) {
-  int N = x.size(0), H = x.size(1), W = x.size(2), Cin = x.size(3);
+  int N = x.size(0), H = x.size(1), W = x.size(2);
   int OH = y.size(1), OW = y.size(2), Cout = y.size(3);
   int OWp2 = (OW + 31) / 32 * 32;
   int SEG = OWp2 / 32;
   int LT = (OH + 3) / 4;                 // TR = 4
   int runs = N * SEG;
-  int wavesn = Cout % 128 == 0 ? 4 : 2;
-  int blocks = runs * (Cout / (32 * wavesn));
+  int blocks = runs * (Cout / 64);
   const u16* zp = zero_page(x.options());
   const u16* bp = bias ? bfp(*bias) : nullptr;
   auto s = cur_stream();
-#define CRL(WN, KHV, KWV, EPIV)                                                hipLaunchKernelGGL((conv_ring_kernel<WN, KHV, KWV, EPIV>), dim3(blocks),                        dim3(512), 0, s, bfp(x), bfp(wT), bp, bfp_mut(y), zp,                        N, H, W, Cin, Cout, OH, OW, OWp2, pad, runs, LT)
-#define CRL_EPI(WN, KHV, KWV)                                                  do {                                                                           if (epi == 0) CRL(WN, KHV, KWV, 0);                                          else if (epi == 1) CRL(WN, KHV, KWV, 1);                                     else if (epi == 2) CRL(WN, KHV, KWV, 2);                                     else CRL(WN, KHV, KWV, 3);                                                 } while (0)
-  if (wavesn == 4) {
-    if (KH == 3) CRL_EPI(4, 3, 3); else CRL_EPI(4, 5, 5);
-  } else {
-    if (KH == 3) CRL_EPI(2, 3, 3); else CRL_EPI(2, 5, 5);
-  }
-#undef CRL_EPI
+#define CRL(EPIV)                                                            \
+  hipLaunchKernelGGL((conv_ring_kernel<EPIV>), dim3(blocks), dim3(512), 0,   \
+                     s, bfp(x), bfp(wT), bp, bfp_mut(y), zp, N, H, W, Cout,  \
+                     OH, OW, OWp2, pad, runs, LT)
+  if (epi == 0) CRL(0);
+  else if (epi == 1) CRL(1);
+  else if (epi == 2) CRL(2);
+  else CRL(3);
 #undef CRL
-}
-
-
-// flipped+transposed weight for the ring dgrad: wFT[ci][(kh,kw,co)] =
-// w[KH-1-kh][KW-1-kw][ci][co] (dx of a stride-1 conv IS a conv of dy with
-// this kernel at pad' = KH-1-pad). Cached per optimizer step (kind 3).
-static at::Tensor cached_wft(bool cache_w, const at::Tensor& w, int KH,
-                             int KW, int Cin, int Cout) {
-  return cached_weight_form(3, cache_w, w,
-                            [&](c10::optional<at::Tensor> prev) {
-    auto f = w.flip({0, 1}).permute({2, 0, 1, 3});   // [Cin][KH][KW][Cout]
-    if (prev) {
-      prev->view({Cin, KH, KW, Cout}).copy_(f);
-      return *prev;
-    }
-    return f.contiguous().view({Cin, (i64)KH * KW * Cout});
-  });
 }
 
 // ------------------------------------------------------------------ conv
