@@ -1145,6 +1145,22 @@ static void conv_ring(const at::Tensor& x, const at::Tensor& wT,
 #undef CRL
 }
 
+// flipped+transposed weight for the ring dgrad: wFT[ci][(kh,kw,co)] =
+// w[KH-1-kh][KW-1-kw][ci][co] (dx of a stride-1 conv IS a conv of dy with
+// this kernel at pad' = KH-1-pad). Cached per optimizer step (kind 3).
+static at::Tensor cached_wft(bool cache_w, const at::Tensor& w, int KH,
+                             int KW, int Cin, int Cout) {
+  return cached_weight_form(3, cache_w, w,
+                            [&](c10::optional<at::Tensor> prev) {
+    auto f = w.flip({0, 1}).permute({2, 0, 1, 3});   // [Cin][KH][KW][Cout]
+    if (prev) {
+      prev->view({Cin, KH, KW, Cout}).copy_(f);
+      return *prev;
+    }
+    return f.contiguous().view({Cin, (i64)KH * KW * Cout});
+  });
+}
+
 // ------------------------------------------------------------------ conv
 static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
                            int& OH, int& OW) {
