@@ -44,7 +44,7 @@ __global__ __launch_bounds__(512) void conv_ring_kernel(
   constexpr int TR = 4;                  // output rows per tile (128 pos)
   constexpr int RH = TR + KH - 1;
   constexpr int S0 = (RH + TR - 1) / TR; // 2 window sets
-  constexpr int NSS = S0 + 2;            // x ring slots (3 in use + 1)
+  constexpr int NSS = S0 + 1;            // x ring slots (prefetch depth 1)
   constexpr int WWP = CT + 8;
   constexpr int BNC = 64;
   constexpr int KFULL = KH * KW * CIN;   // 576
@@ -121,20 +121,20 @@ __global__ __launch_bounds__(512) void conv_ring_kernel(
   }
 #pragma unroll
   for (int s = 0; s < S0; ++s) CR_STAGE_X(s, s);
-  CR_STAGE_X(S0, S0);                    // tile 1's new set
-  int sfree = S0 + 1;
+  int sfree = S0;
   int sbase = 0;
 
   f32x4 acc[NI][2];
   const u16* const Bs_ = &smem[NSS * SETE];
   for (int tile = 0; tile < LT; ++tile) {
     const int oh0 = tile * TR;
-    // tile's window + the B panel are in LDS; tile+1's set may fly
-    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(ASI) : "memory");
+    // tile's window + the B panel are in LDS (the set staged last tile
+    // had a whole k-loop to land — the drain is cheap)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();        // also fences x slot reuse
-    // stage tile+2's new x set under this tile's compute (sets past the
-    // run read the zero page — keeps the wait count uniform)
-    CR_STAGE_X(sfree, tile + 2 + S0 - 1);
+    // stage tile+1's new x set (= set tile+2) under this tile's compute
+    // (sets past the run read the zero page)
+    CR_STAGE_X(sfree, tile + S0);
     sfree = sfree + 1 == NSS ? 0 : sfree + 1;
 
 #pragma unroll
