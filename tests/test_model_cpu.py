@@ -130,3 +130,27 @@ def test_vggf_dropout_train_vs_eval():
     m.train()
     y_tr = m(x)
     assert not torch.allclose(y_tr, y_eval1)  # train: masked path differs
+
+
+def test_batchnorm_residual_fusion_cpu():
+    """ops.batchnorm(..., res=) must equal relu(bn(x) + res) composed —
+    the CPU reference path for the fused GPU kernel (K17+K18)."""
+    from distributed_vgg_f_amd import ops
+    from distributed_vgg_f_amd.ops import cpu_ref
+    torch.manual_seed(5)
+    x = torch.randn(2, 4, 4, 8)
+    res = torch.randn(2, 4, 4, 8)
+    g = torch.rand(8) + 0.5
+    b = torch.randn(8)
+    got = ops.batchnorm(x, g, b, 1e-5, False, res=res)
+    want = cpu_ref.add_relu(cpu_ref.batchnorm(x, g, b, 1e-5, False), res)
+    assert torch.allclose(got, want, atol=1e-5)
+
+
+def test_resnet50_uses_fused_bn_residual():
+    """The bottleneck's bn3 call must carry the residual (fusion seam)."""
+    from distributed_vgg_f_amd.models.resnet50 import Bottleneck
+    blk = Bottleneck(64, 16, 1, downsample=True)
+    y = blk(torch.randn(1, 8, 8, 64))
+    assert y.shape == (1, 8, 8, 64)
+    assert float(y.min()) >= 0.0   # fused ReLU applied after the add
