@@ -231,9 +231,12 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   // scalar thread-per-channel reduce is already near bandwidth-bound
   // (coalesced 2 B/lane across a 256-lane channel window) and the wide
   // version only multiplied end-of-thread atomics. Kept scalar.
-  int rpb = std::max(256, int(R * ceil_div(C, 2048) / 1024));
+  int gridx = ceil_div(C, 2048);
+  int rpi = 256 / std::min(C / 8, 256);
+  i64 want_y = std::max<i64>(1, 2048 / gridx);
+  int rpb = int(std::max<i64>(rpi, ceil_div(R, want_y)));
   auto s = cur_stream();
-  dim3 rgrid(ceil_div(C, 2048), ceil_div(R, (i64)rpb));
+  dim3 rgrid(gridx, ceil_div(R, (i64)rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(), R, C,
                      rpb);
@@ -413,9 +416,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dx = at::empty_like(dy);
   auto dgamma = dg_out ? *dg_out : at::empty({C}, dy.options());
   auto dbeta = db_out ? *db_out : at::empty({C}, dy.options());
-  int rpb = std::max(256, int(R * ceil_div(C, 2048) / 1024));
+  int gridx = ceil_div(C, 2048);
+  int rpi = 256 / std::min(C / 8, 256);
+  i64 want_y = std::max<i64>(1, 2048 / gridx);
+  int rpb = int(std::max<i64>(rpi, ceil_div(R, want_y)));
   auto s = cur_stream();
-  dim3 rgrid(ceil_div(C, 2048), ceil_div(R, (i64)rpb));
+  dim3 rgrid(gridx, ceil_div(R, (i64)rpb));
   const u16* yp = relu ? bfp(yr) : nullptr;
   if (relu)
     hipLaunchKernelGGL(bn_bwd_reduce_kernel<true>, rgrid, dim3(256), 0, s,
