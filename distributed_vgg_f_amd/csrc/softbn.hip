@@ -89,60 +89,24 @@ std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor& logits,
 // Pass 2: normalize (+affine, optional fused ReLU); saves mean/rstd fp32.
 // partials land in 256 slots per channel (slot = blockIdx.y & 255):
 // direct atomics onto C floats from thousands of blocks serialize.
-// Vectorized: lane = (row_local, 8-channel chunk) so a wave's loads are
-// 1 KiB contiguous b128 (the round-1 scalar thread-per-channel form was
-// ~6x off roofline: 2 B/lane and, for C=64, 192 of 256 lanes idle).
-// Per-thread fp32 accumulators over a row stripe, then an LDS tree over
-// the rows sharing a chunk, one atomic per channel per block (into the
-// 256-way slotted partials as before).
 __global__ void bn_reduce_kernel(const u16* __restrict__ x,
                                  float* __restrict__ sum,
                                  float* __restrict__ sumsq, i64 R, int C,
                                  int rows_per_block) {
-  __shared__ float red[2][256 * 8];
-  int nch = C / 8;                       // 8-channel chunks per row
-  int cpb = nch < 256 ? nch : 256;       // chunks this block covers
-  int rpi = 256 / cpb;                   // rows read per iteration
-  int cid = threadIdx.x % cpb;           // chunk within block's span
-  int rloc = threadIdx.x / cpb;
-  int c8 = (blockIdx.x * 256 + cid) * 8;
-  bool act = threadIdx.x < cpb * rpi && c8 < C;
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
-  float sa[8], qa[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) sa[j] = qa[j] = 0.f;
-  if (act)
-    for (i64 r = r0 + rloc; r < r1; r += rpi) {
-      s16x8 v = *reinterpret_cast<const s16x8*>(x + r * C + c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = bf2f(u16(v[j]));
-        sa[j] += f; qa[j] += f * f;
-      }
-    }
-  // LDS tree over rloc groups (rpi is a power of two for C % 8 == 0...
-  // not guaranteed: do a simple strided loop reduce by thread 0 of each
-  // chunk group instead when rpi is small; rpi in {1,2,4,...,256/..}
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    red[0][threadIdx.x * 8 + j] = sa[j];
-    red[1][threadIdx.x * 8 + j] = qa[j];
+  float s0 = 0.f, s1 = 0.f, q0 = 0.f, q1 = 0.f;
+  i64 r = r0;
+  for (; r + 2 <= r1; r += 2) {
+    float v0 = bf2f(x[r * C + c]), v1 = bf2f(x[(r + 1) * C + c]);
+    s0 += v0; q0 += v0 * v0; s1 += v1; q1 += v1 * v1;
   }
-  __syncthreads();
-  if (rloc == 0 && act) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float a = 0.f, q = 0.f;
-      for (int rr = 0; rr < rpi; ++rr) {
-        a += red[0][(rr * cpb + cid) * 8 + j];
-        q += red[1][(rr * cpb + cid) * 8 + j];
-      }
-      int slot = blockIdx.y & 255;
-      atomicAdd(&sum[(i64)slot * C + c8 + j], a);
-      atomicAdd(&sumsq[(i64)slot * C + c8 + j], q);
-    }
-  }
+  for (; r < r1; ++r) { float v = bf2f(x[r * C + c]); s0 += v; q0 += v * v; }
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sum[(i64)slot * C + c], s0 + s1);
+  atomicAdd(&sumsq[(i64)slot * C + c], q0 + q1);
 }
 
 __global__ void bn_stats_kernel(const float* __restrict__ sum,
@@ -231,12 +195,9 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   // scalar thread-per-channel reduce is already near bandwidth-bound
   // (coalesced 2 B/lane across a 256-lane channel window) and the wide
   // version only multiplied end-of-thread atomics. Kept scalar.
-  int gridx = ceil_div(C, 2048);
-  int rpi = 256 / std::min(C / 8, 256);
-  i64 want_y = std::max<i64>(1, 2048 / gridx);
-  int rpb = int(std::max<i64>(rpi, ceil_div(R, want_y)));
+  int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
   auto s = cur_stream();
-  dim3 rgrid(gridx, ceil_div(R, (i64)rpb));
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, (i64)rpb));
   hipLaunchKernelGGL(bn_reduce_kernel, rgrid, dim3(256), 0, s, bfp(x),
                      sum.data_ptr<float>(), sumsq.data_ptr<float>(), R, C,
                      rpb);
@@ -280,56 +241,32 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
                                      float* __restrict__ sdy,
                                      float* __restrict__ sdyx, i64 R, int C,
                                      int rows_per_block) {
-  __shared__ float red[2][256 * 8];
-  int nch = C / 8;
-  int cpb = nch < 256 ? nch : 256;
-  int rpi = 256 / cpb;
-  int cid = threadIdx.x % cpb;
-  int rloc = threadIdx.x / cpb;
-  int c8 = (blockIdx.x * 256 + cid) * 8;
-  bool act = threadIdx.x < cpb * rpi && c8 < C;
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
-  float m[8], rs[8], aa[8], bb[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    aa[j] = bb[j] = 0.f;
-    m[j] = act ? mean[c8 + j] : 0.f;
-    rs[j] = act ? rstd[c8 + j] : 0.f;
-  }
-  if (act)
-    for (i64 r = r0 + rloc; r < r1; r += rpi) {
-      s16x8 dv = *reinterpret_cast<const s16x8*>(dy + r * C + c8);
-      s16x8 xv = *reinterpret_cast<const s16x8*>(x + r * C + c8);
-      s16x8 yv{};
-      if (RELU) yv = *reinterpret_cast<const s16x8*>(yr + r * C + c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float d = bf2f(u16(dv[j]));
-        if (RELU) d = bf2f(u16(yv[j])) > 0.f ? d : 0.f;
-        float h = (bf2f(u16(xv[j])) - m[j]) * rs[j];
-        aa[j] += d; bb[j] += d * h;
-      }
+  float a0 = 0.f, a1 = 0.f, b0 = 0.f, b1 = 0.f;
+  float m = mean[c], rs = rstd[c];
+  i64 r = r0;
+  for (; r + 2 <= r1; r += 2) {
+    float d0 = bf2f(dy[r * C + c]), d1 = bf2f(dy[(r + 1) * C + c]);
+    if (RELU) {
+      d0 = bf2f(yr[r * C + c]) > 0.f ? d0 : 0.f;
+      d1 = bf2f(yr[(r + 1) * C + c]) > 0.f ? d1 : 0.f;
     }
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    red[0][threadIdx.x * 8 + j] = aa[j];
-    red[1][threadIdx.x * 8 + j] = bb[j];
+    float h0 = (bf2f(x[r * C + c]) - m) * rs;
+    float h1 = (bf2f(x[(r + 1) * C + c]) - m) * rs;
+    a0 += d0; b0 += d0 * h0; a1 += d1; b1 += d1 * h1;
   }
-  __syncthreads();
-  if (rloc == 0 && act) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float a = 0.f, b = 0.f;
-      for (int rr = 0; rr < rpi; ++rr) {
-        a += red[0][(rr * cpb + cid) * 8 + j];
-        b += red[1][(rr * cpb + cid) * 8 + j];
-      }
-      int slot = blockIdx.y & 255;
-      atomicAdd(&sdy[(i64)slot * C + c8 + j], a);
-      atomicAdd(&sdyx[(i64)slot * C + c8 + j], b);
-    }
+  for (; r < r1; ++r) {
+    float d = bf2f(dy[r * C + c]);
+    if (RELU) d = bf2f(yr[r * C + c]) > 0.f ? d : 0.f;
+    float h = (bf2f(x[r * C + c]) - m) * rs;
+    a0 += d; b0 += d * h;
   }
+  int slot = blockIdx.y & 255;
+  atomicAdd(&sdy[(i64)slot * C + c], a0 + a1);
+  atomicAdd(&sdyx[(i64)slot * C + c], b0 + b1);
 }
 
 // One pass over the 256-way partials: emits dgamma/dbeta and the three
@@ -416,12 +353,9 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dx = at::empty_like(dy);
   auto dgamma = dg_out ? *dg_out : at::empty({C}, dy.options());
   auto dbeta = db_out ? *db_out : at::empty({C}, dy.options());
-  int gridx = ceil_div(C, 2048);
-  int rpi = 256 / std::min(C / 8, 256);
-  i64 want_y = std::max<i64>(1, 2048 / gridx);
-  int rpb = int(std::max<i64>(rpi, ceil_div(R, want_y)));
+  int rpb = std::max(32, int(R / std::max<i64>(1, 2048 / ceil_div(C, 256))));
   auto s = cur_stream();
-  dim3 rgrid(gridx, ceil_div(R, (i64)rpb));
+  dim3 rgrid(ceil_div(C, 256), ceil_div(R, (i64)rpb));
   const u16* yp = relu ? bfp(yr) : nullptr;
   if (relu)
     hipLaunchKernelGGL(bn_bwd_reduce_kernel<true>, rgrid, dim3(256), 0, s,
