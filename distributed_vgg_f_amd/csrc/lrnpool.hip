@@ -260,6 +260,57 @@ std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kernel,
   return {y, idx};
 }
 
+// k=3/s=2 specialization (every zoo pool): each input position is
+// covered by at most 2x2 windows at (ohs>>1 - a, ows>>1 - b). The
+// generic double loop diverges lane-by-lane (lanes of one wave span
+// neighbouring positions with different window counts); here all four
+// candidates run predicated with uniform control flow.
+__global__ void maxpool_bwd_k3s2_kernel(const u16* __restrict__ dy,
+                                        const unsigned char* __restrict__ idx,
+                                        u16* __restrict__ dx, int N, int H,
+                                        int W, int C, int OH, int OW,
+                                        int pl) {
+  i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  i64 total = (i64)N * H * W * (C / 8);
+  if (id >= total) return;
+  int c8 = int(id % (C / 8)) * 8;
+  i64 r = id / (C / 8);
+  int iw = int(r % W); r /= W;
+  int ih = int(r % H); r /= H;
+  int n = int(r);
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  int ohs = ih + pl, ows = iw + pl;
+#pragma unroll
+  for (int a = 0; a < 2; ++a) {
+    int oh = (ohs >> 1) - a;
+    int dh = ohs - 2 * oh;
+    bool okh = oh >= 0 && oh < OH && dh <= 2;
+#pragma unroll
+    for (int b = 0; b < 2; ++b) {
+      int ow = (ows >> 1) - b;
+      int dw = ows - 2 * ow;
+      bool ok = okh && ow >= 0 && ow < OW && dw <= 2;
+      i64 o = ok ? (((i64)n * OH + oh) * OW + ow) * C + c8 : 0;
+      unsigned want = ok ? unsigned(dh * 3 + dw) : 255u;
+      const unsigned char* ip = idx + o;
+      u32 i0 = *reinterpret_cast<const u32*>(ip);
+      u32 i1 = *reinterpret_cast<const u32*>(ip + 4);
+      s16x8 d = *reinterpret_cast<const s16x8*>(dy + o);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned iv = ((j < 4 ? i0 : i1) >> ((j & 3) * 8)) & 0xffu;
+        if (iv == want) acc[j] += bf2f(u16(d[j]));
+      }
+    }
+  }
+  s16x8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = short(f2bf(acc[j]));
+  *reinterpret_cast<s16x8*>(dx + (((i64)n * H + ih) * W + iw) * C + c8) = out;
+}
+
 // ----------------------------------------------------------- maxpool bwd
 // gather form (no atomics): each dx element scans its covering windows
 // (3x3 s2 windows overlap).
@@ -317,6 +368,14 @@ at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx,
   TORCH_CHECK(C % 8 == 0, "maxpool bwd needs C % 8 == 0");
   auto dx = at::empty({N, H, W, C}, dy.options());
   i64 total = (i64)N * H * W * (C / 8);
+  if (kernel == 3 && stride == 2) {
+    hipLaunchKernelGGL(maxpool_bwd_k3s2_kernel,
+                       dim3(int((total + 255) / 256)), dim3(256), 0,
+                       cur_stream(), bfp(dy),
+                       idx.data_ptr<unsigned char>(), bfp_mut(dx), N, H, W,
+                       C, OH, OW, int(pl));
+    return dx;
+  }
   hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(int((total + 255) / 256)),
                      dim3(256), 0, cur_stream(), bfp(dy),
                      idx.data_ptr<unsigned char>(), bfp_mut(dx), N, H, W, C,
