@@ -56,6 +56,10 @@ class Config:
     bucket_mb: int = 25                # gradient all-reduce bucket size (bf16 MB)
     overlap_comm: bool = True          # overlap all-reduce with backward
     backend: str = "auto"              # auto -> nccl(RCCL) on GPU, gloo on CPU
+    ring_fallback: bool = False        # hand-rolled p2p ring all-reduce
+    #                                    instead of the library collective
+    #                                    (SURVEY.md 5.8 fallback seam;
+    #                                     disables overlap)
 
     # io / checkpointing
     checkpoint_dir: Optional[str] = None

@@ -60,7 +60,8 @@ class Trainer:
                                     cfg.momentum, cfg.weight_decay,
                                     cfg.bucket_mb)
         broadcast_init(self.arena, self.ctx)
-        self.reducer = GradReducer(self.arena, self.ctx, cfg.overlap_comm)
+        self.reducer = GradReducer(self.arena, self.ctx, cfg.overlap_comm,
+                                   cfg.ring_fallback)
         if cfg.data == "records":
             self.pipeline = RecordPipeline(cfg.data_dir, cfg.batch_size,
                                            self.device, compute_dtype,

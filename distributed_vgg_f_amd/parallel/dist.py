@@ -24,12 +24,13 @@ from __future__ import annotations
 import datetime
 import os
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import TYPE_CHECKING, List, Optional
 
 import torch
 import torch.distributed as dist
 
-from ..engine.arena import ArenaOptimizer
+if TYPE_CHECKING:  # import cycle: engine imports parallel (annotations only)
+    from ..engine.arena import ArenaOptimizer
 
 
 @dataclass
@@ -69,7 +70,7 @@ def init_distributed(backend: str = "auto",
     return DistContext(rank, world, local_rank, backend)
 
 
-def broadcast_init(arena: ArenaOptimizer, ctx: DistContext) -> None:
+def broadcast_init(arena: "ArenaOptimizer", ctx: DistContext) -> None:
     """C4: rank-0's fp32 master arena becomes everyone's; compute copy synced."""
     if not ctx.is_distributed:
         return
@@ -77,14 +78,65 @@ def broadcast_init(arena: ArenaOptimizer, ctx: DistContext) -> None:
     arena.sync_compute_weights()
 
 
+def ring_all_reduce_(t: torch.Tensor, group=None) -> torch.Tensor:
+    """Hand-rolled ring all-reduce (SURVEY.md 5.8 fallback seam): a
+    reduce-scatter pass followed by an all-gather pass over P-1 steps of
+    point-to-point send/recv each. On ROCm the nccl backend's send/recv
+    are RCCL point-to-point over xGMI; on CPU/gloo this is the testable
+    reference. SUM semantics, in place, matching dist.all_reduce.
+
+    Kept behind the same interface as the RCCL collective so the engine
+    can swap it in (Config.ring_fallback) if the library collective
+    misbehaves on a given stack. The library path stays the default: a
+    single rcclAllReduce already builds ~7 parallel rings over the
+    fully-connected xGMI links, which this single ring does not."""
+    world = dist.get_world_size(group)
+    if world == 1:
+        return t
+    rank = dist.get_rank(group)
+    flat = t.view(-1)
+    # pad-free chunking: torch.chunk gives world pieces (last may be short)
+    n = flat.numel()
+    csz = (n + world - 1) // world
+    chunks = [flat[i * csz:min(n, (i + 1) * csz)] for i in range(world)]
+    right = (rank + 1) % world
+    left = (rank - 1) % world
+    buf = torch.empty(csz, dtype=t.dtype, device=t.device)
+    # reduce-scatter: after P-1 steps rank r owns the full sum of chunk
+    # (r+1) % world
+    for step in range(world - 1):
+        s_i = (rank - step) % world          # chunk we send
+        r_i = (rank - step - 1) % world      # chunk we receive+accumulate
+        rb = buf[:chunks[r_i].numel()]
+        ops = [dist.P2POp(dist.isend, chunks[s_i].contiguous(), right,
+                          group=group),
+               dist.P2POp(dist.irecv, rb, left, group=group)]
+        for w in dist.batch_isend_irecv(ops):
+            w.wait()
+        chunks[r_i].add_(rb)
+    # all-gather the owned chunks around the ring
+    for step in range(world - 1):
+        s_i = (rank + 1 - step) % world
+        r_i = (rank - step) % world
+        rb = buf[:chunks[r_i].numel()]
+        ops = [dist.P2POp(dist.isend, chunks[s_i].contiguous(), right,
+                          group=group),
+               dist.P2POp(dist.irecv, rb, left, group=group)]
+        for w in dist.batch_isend_irecv(ops):
+            w.wait()
+        chunks[r_i].copy_(rb)
+    return t
+
+
 class GradReducer:
     """Bucketed, overlapped gradient all-reduce over the arena (C1)."""
 
-    def __init__(self, arena: ArenaOptimizer, ctx: DistContext,
-                 overlap: bool = True):
+    def __init__(self, arena: "ArenaOptimizer", ctx: DistContext,
+                 overlap: bool = True, ring: bool = False):
         self.arena = arena
         self.ctx = ctx
-        self.overlap = overlap and ctx.is_distributed
+        self.ring = ring
+        self.overlap = overlap and ctx.is_distributed and not ring
         self._works: List[dist.Work] = []
         self._pending: dict = {}
         self._param_bucket = {}
@@ -126,5 +178,9 @@ class GradReducer:
             for w in self._works:
                 w.wait()
             self._works = []
+        elif self.ring:
+            # SURVEY.md 5.8 fallback: hand-rolled p2p ring (no overlap)
+            for b in self.arena.buckets:
+                ring_all_reduce_(self.arena.g[b.start:b.end])
         else:
             dist.all_reduce(self.arena.g)

@@ -93,6 +93,54 @@ def _run_trainer_2rank(rank, world, port, q):
             dist.destroy_process_group()
 
 
+def _run_ring_allreduce(rank, world, port, q):
+    try:
+        _init(rank, world, port)
+        from distributed_vgg_f_amd.parallel.dist import ring_all_reduce_
+
+        torch.manual_seed(50 + rank)
+        for n in (7, 64, 1 << 16, (1 << 16) + 13):   # odd sizes incl. tails
+            t = torch.randn(n)
+            ref = t.clone()
+            dist.all_reduce(ref)                     # library SUM
+            ring_all_reduce_(t)                      # hand-rolled ring
+            assert torch.allclose(t, ref, atol=1e-5), n
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+        q.put((rank, traceback.format_exc()))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def _run_trainer_ring_2rank(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        from distributed_vgg_f_amd.engine.trainer import Trainer
+        cfg = Config(model="vggf", image_size=64, batch_size=2, train_steps=2,
+                     log_every=0, backend="gloo", bucket_mb=8,
+                     ring_fallback=True)
+        t = Trainer(cfg)
+        t.train(2)
+        ws = [torch.zeros_like(t.arena.w32) for _ in range(world)]
+        dist.all_gather(ws, t.arena.w32)
+        assert torch.equal(ws[0], ws[1])
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+        q.put((rank, traceback.format_exc()))
+    finally:
+        for k in ("WORLD_SIZE", "RANK", "LOCAL_RANK"):
+            os.environ.pop(k, None)
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
 def _spawn(fn, port):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
@@ -115,3 +163,15 @@ def test_bucketed_allreduce_gloo():
 @pytest.mark.timeout(300)
 def test_trainer_replicas_identical_gloo():
     _spawn(_run_trainer_2rank, 29713)
+
+
+@pytest.mark.timeout(300)
+def test_ring_allreduce_matches_library():
+    """SURVEY.md 5.8 fallback: the hand-rolled ring reduce-scatter +
+    all-gather must equal the library all-reduce (SUM) at odd sizes."""
+    _spawn(_run_ring_allreduce, 29717)
+
+
+@pytest.mark.timeout(300)
+def test_trainer_ring_fallback_gloo():
+    _spawn(_run_trainer_ring_2rank, 29719)
