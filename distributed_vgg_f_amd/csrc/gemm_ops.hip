@@ -156,7 +156,7 @@ template <class F>
 static at::Tensor cached_weight_form(int kind, bool on, const at::Tensor& w,
                                      F&& make) {
   if (!on || stream_capturing()) return make(c10::nullopt);
-  static std::unordered_map<const void*, WTEntry> cache[4];
+  static std::unordered_map<const void*, WTEntry> cache[5];
   auto& c = cache[kind];
   std::vector<i64> sizes(w.sizes().begin(), w.sizes().end());
   auto it = c.find(w.data_ptr());
@@ -1161,6 +1161,57 @@ static at::Tensor cached_wft(bool cache_w, const at::Tensor& w, int KH,
   });
 }
 
+// Phase-interleaved NHWC transform for the small-Cin strided first
+// layer (conv1 11x11/s4/C3): xP[n][hq][wq][(ph*s+pw)*C + c] =
+// x[n][hq*s+ph-pad][wq*s+pw-pad][c]. The layer then lowers to a DENSE
+// stride-1 conv over s*s*C = 48 phase-channels with Kq = ceil(KH/s)
+// taps: K drops from 11*11*8 (channel-padded) = 968 to 3*3*48 = 432 and
+// every gather is 16B-aligned. One block per (n, hq): the s source rows
+// are staged in LDS, outputs written as b128.
+__global__ void nhwc2phase_nhwc_kernel(const u16* __restrict__ x,
+                                       u16* __restrict__ out, int N, int H,
+                                       int W, int C, int s, int Hq, int Wq,
+                                       int Wqp, int pad) {
+  extern __shared__ u16 rows[];            // s * W * C elems
+  int nh = blockIdx.x;
+  int n = nh / Hq, hq = nh % Hq;
+  int t = threadIdx.x;
+  int rowe = W * C;
+  for (int ph = 0; ph < s; ++ph) {
+    int ih = hq * s + ph - pad;
+    u16* dstrow = rows + ph * rowe;
+    if ((unsigned)ih < (unsigned)H) {
+      const u16* src = x + ((i64)n * H + ih) * (i64)rowe;
+      for (int e8 = t * 8; e8 < rowe; e8 += blockDim.x * 8) {
+        if (e8 + 8 <= rowe)
+          *reinterpret_cast<s16x8*>(dstrow + e8) =
+              *reinterpret_cast<const s16x8*>(src + e8);
+        else
+          for (int j = e8; j < rowe; ++j) dstrow[j] = src[j];
+      }
+    } else {
+      for (int e = t; e < rowe; e += blockDim.x) dstrow[e] = 0;
+    }
+  }
+  __syncthreads();
+  int PC = s * s * C;                      // phase-channels per position
+  int chunks = PC / 8;                     // host guarantees PC % 8 == 0
+  for (int job = t; job < Wq * chunks; job += blockDim.x) {
+    int wq = job / chunks, k8 = (job % chunks) * 8;
+    union { s16x8 v; u16 e[8]; } pk;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int pcc = k8 + j;
+      int c = pcc % C, pw = (pcc / C) % s, ph = pcc / (C * s);
+      int iw = wq * s + pw - pad;
+      pk.e[j] = ((unsigned)iw < (unsigned)W) ? rows[ph * W * C + iw * C + c]
+                                             : u16(0);
+    }
+    *reinterpret_cast<s16x8*>(
+        out + (((i64)n * Hq + hq) * Wqp + wq) * PC + k8) = pk.v;
+  }
+}
+
 // ------------------------------------------------------------------ conv
 static void conv_out_shape(int H, int W, int KH, int KW, int stride, int pad,
                            int& OH, int& OW) {
@@ -1190,8 +1241,40 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int epi = epi_of(bias.has_value(), relu);
 
   if (Cin < 8) {
-    // first layers (conv1 Cin=3, ResNet stem): pad channels to 8 and take
-    // the vector-gather glds path; zero channels contribute nothing
+    // conv1 (11x11/s4/C3): phase-plane lowering — a DENSE stride-1 conv
+    // over s*s*C = 48 phase-channels with ceil(KH/s) = 3 taps (K 968 ->
+    // 432 vs the channel-padded form, all gathers 16B-aligned)
+    int PC = stride * stride * Cin;
+    if (stride >= 2 && PC % 8 == 0 && KH == KW &&
+        env_int("DVGG_CONV1_PHASE", 1)) {
+      int Kq = (KH + stride - 1) / stride;
+      int Hq = OH + Kq - 1, Wq = OW + Kq - 1;
+      int Wqp = Wq;  // dense layout, no pad needed (pad folded into xP)
+      auto xP = at::empty({N, Hq, Wqp, PC}, x.options());
+      TORCH_CHECK(stride * W * Cin <= 16384, "phase rows exceed LDS");
+      hipLaunchKernelGGL(nhwc2phase_nhwc_kernel, dim3(N * Hq), dim3(256),
+                         stride * W * Cin * sizeof(u16), cur_stream(),
+                         bfp(x), bfp_mut(xP), N, H, W, Cin, stride, Hq, Wq,
+                         Wqp, pad);
+      // wP[q][r][(ph,pw,c)][co] = w[s q + ph][s r + pw][c][co], zero
+      // padded beyond KH; cached per optimizer step in its final layout
+      auto wP = cached_weight_form(4, cache_w, w,
+                                   [&](c10::optional<at::Tensor> prev) {
+        auto t6 = at::zeros({Kq * stride, Kq * stride, Cin, Cout},
+                            w.options());
+        t6.narrow(0, 0, KH).narrow(1, 0, KW).copy_(w);
+        auto perm = t6.view({Kq, stride, Kq, stride, Cin, Cout})
+            .permute({0, 2, 1, 3, 4, 5});
+        if (prev) {
+          prev->view({Kq, Kq, stride, stride, Cin, Cout}).copy_(perm);
+          return *prev;
+        }
+        return perm.contiguous().view({Kq, Kq, (i64)PC, Cout});
+      });
+      return conv2d_fwd(xP, wP, bias, 1, 0, relu, cache_w);
+    }
+    // ResNet-stem-style fallback: pad channels to 8 and take the
+    // vector-gather glds path; zero channels contribute nothing
     auto x8 = pad_channels8(x);
     auto w8 = cached_weight_form(2, cache_w, w, [&](c10::optional<at::Tensor> prev) {
       auto t = prev ? *prev : at::zeros({KH, KW, 8, Cout}, w.options());
