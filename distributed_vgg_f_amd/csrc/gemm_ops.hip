@@ -1194,8 +1194,9 @@ __global__ void nhwc2phase_nhwc_kernel(const u16* __restrict__ x,
     }
   }
   __syncthreads();
-  int PC = s * s * C;                      // phase-channels per position
-  int chunks = PC / 8;                     // host guarantees PC % 8 == 0
+  int PC = s * s * C;                      // real phase-channels
+  int PCp = (PC + 7) & ~7;                 // stored (zero-padded to 8)
+  int chunks = PCp / 8;
   for (int job = t; job < Wq * chunks; job += blockDim.x) {
     int wq = job / chunks, k8 = (job % chunks) * 8;
     union { s16x8 v; u16 e[8]; } pk;
@@ -1204,11 +1205,11 @@ __global__ void nhwc2phase_nhwc_kernel(const u16* __restrict__ x,
       int pcc = k8 + j;
       int c = pcc % C, pw = (pcc / C) % s, ph = pcc / (C * s);
       int iw = wq * s + pw - pad;
-      pk.e[j] = ((unsigned)iw < (unsigned)W) ? rows[ph * W * C + iw * C + c]
-                                             : u16(0);
+      pk.e[j] = (pcc < PC && (unsigned)iw < (unsigned)W)
+          ? rows[ph * W * C + iw * C + c] : u16(0);
     }
     *reinterpret_cast<s16x8*>(
-        out + (((i64)n * Hq + hq) * Wqp + wq) * PC + k8) = pk.v;
+        out + (((i64)n * Hq + hq) * Wqp + wq) * PCp + k8) = pk.v;
   }
 }
 
@@ -1245,12 +1246,12 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
     // over s*s*C = 48 phase-channels with ceil(KH/s) = 3 taps (K 968 ->
     // 432 vs the channel-padded form, all gathers 16B-aligned)
     int PC = stride * stride * Cin;
-    if (stride >= 2 && PC % 8 == 0 && KH == KW &&
-        env_int("DVGG_CONV1_PHASE", 1)) {
+    int PCp = ceil8(PC);
+    if (stride >= 2 && KH == KW && env_int("DVGG_CONV1_PHASE", 1)) {
       int Kq = (KH + stride - 1) / stride;
       int Hq = OH + Kq - 1, Wq = OW + Kq - 1;
       int Wqp = Wq;  // dense layout, no pad needed (pad folded into xP)
-      auto xP = at::empty({N, Hq, Wqp, PC}, x.options());
+      auto xP = at::empty({N, Hq, Wqp, PCp}, x.options());
       TORCH_CHECK(stride * W * Cin <= 16384, "phase rows exceed LDS");
       hipLaunchKernelGGL(nhwc2phase_nhwc_kernel, dim3(N * Hq), dim3(256),
                          stride * W * Cin * sizeof(u16), cur_stream(),
@@ -1266,10 +1267,13 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
         auto perm = t6.view({Kq, stride, Kq, stride, Cin, Cout})
             .permute({0, 2, 1, 3, 4, 5});
         if (prev) {
-          prev->view({Kq, Kq, stride, stride, Cin, Cout}).copy_(perm);
+          prev->view({Kq, Kq, (i64)PCp, Cout}).narrow(2, 0, PC)
+              .copy_(perm.reshape({Kq, Kq, (i64)PC, Cout}));
           return *prev;
         }
-        return perm.contiguous().view({Kq, Kq, (i64)PC, Cout});
+        auto t = at::zeros({Kq, Kq, (i64)PCp, Cout}, w.options());
+        t.narrow(2, 0, PC).copy_(perm.reshape({Kq, Kq, (i64)PC, Cout}));
+        return t;
       });
       return conv2d_fwd(xP, wP, bias, 1, 0, relu, cache_w);
     }
