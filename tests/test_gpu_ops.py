@@ -85,6 +85,8 @@ CONV_CASES = [
     (2, 33, 33, 64, 128, 3, 1, 1, "conv3x3-ow33pad64"),
     (2, 7, 7, 128, 128, 3, 1, 1, "conv3x3-ow7"),
     (2, 14, 14, 128, 64, 3, 1, 0, "conv3x3-p0-co64"),
+    # phase-plane first layer with PC % 8 != 0 (ResNet stem: PC 12 -> 16)
+    (2, 112, 112, 3, 64, 7, 2, 3, "stem-phase"),
 ]
 
 
