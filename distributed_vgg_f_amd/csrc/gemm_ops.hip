@@ -329,6 +329,18 @@ static at::Tensor cached_zero_buffer(const std::string& key, i64 numel,
   return it->second;
 }
 
+// Scratch fp32 split-K accumulators (atomic epilogue): cached per
+// (site, shape); the finalize kernel re-zeros the buffer after reading
+// it, so the per-call at::zeros Fill launch disappears (~10 small Fill
+// launches/step on VGG-F, each mostly launch latency).
+static at::Tensor scratch_c32(const char* site, i64 R, i64 C,
+                              const at::TensorOptions& opt) {
+  char key[64];
+  snprintf(key, sizeof key, "c32:%s:%lldx%lld", site, (long long)R,
+           (long long)C);
+  return cached_zero_buffer(key, R * C, opt.dtype(at::kFloat));
+}
+
 static at::Tensor nhwc2cnhw_pad(const at::Tensor& x, int Hp, int Wp, int ro,
                                 int co, char role) {
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
@@ -376,7 +388,7 @@ __global__ void cast_f32_bf16_2(const float* __restrict__ in,
 }
 
 // db[c] = bf16( sum over 256 slots of db32p[s][c] )
-__global__ void bias_slot_reduce_kernel(const float* __restrict__ db32p,
+__global__ void bias_slot_reduce_kernel(float* __restrict__ db32p,
                                         u16* __restrict__ db, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
@@ -384,6 +396,8 @@ __global__ void bias_slot_reduce_kernel(const float* __restrict__ db32p,
 #pragma unroll 8
   for (int s = 0; s < 256; ++s) a += db32p[(i64)s * C + c];
   db[c] = f2bf(a);
+#pragma unroll 8
+  for (int s = 0; s < 256; ++s) db32p[(i64)s * C + c] = 0.f;  // scratch_c32
 }
 
 // Vectorized preamble (see bwd_preamble_kernel below for semantics):
@@ -512,10 +526,11 @@ __global__ void bwd_preamble_kernel(const u16* __restrict__ dy,
 // out[id] = cast(sum over slices of c32[z][id] (+bias[c]) (relu));
 // c32 is [slices][R][C], 8 fp32 per thread per slice.
 template <int EPI>
-__global__ void finalize_kernel(const float* __restrict__ c32,
+__global__ void finalize_kernel(float* __restrict__ c32,
                                 u16* __restrict__ out,
                                 const u16* __restrict__ bias, i64 total,
-                                int C, int slices, i64 slice_stride) {
+                                int C, int slices, i64 slice_stride,
+                                int rezero) {
   i64 id8 = ((i64)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   if (id8 >= total) return;
   float acc[8];
@@ -527,6 +542,13 @@ __global__ void finalize_kernel(const float* __restrict__ c32,
     for (int j = 0; j < 8; ++j)
       if (id8 + j < total) acc[j] += p[j];
   }
+  if (rezero)  // scratch_c32 contract: leave the accumulator zeroed
+    for (int z = 0; z < slices; ++z) {
+      float* q = c32 + (i64)z * slice_stride + id8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (id8 + j < total) q[j] = 0.f;
+    }
   for (int j = 0; j < 8; ++j) {
     i64 id = id8 + j;
     if (id >= total) break;
@@ -561,7 +583,7 @@ __global__ void finalize_narrow_kernel(const float* __restrict__ c32,
 
 static void run_finalize(const at::Tensor& c32, at::Tensor& out,
                          const at::Tensor* bias, int epi, int R, int C,
-                         int slices = 1) {
+                         int slices = 1, bool rezero = false) {
   i64 total = (i64)R * C;
   i64 sstride = (i64)R * C;
   int blocks = int((total / 8 + 256) / 256);
@@ -577,16 +599,16 @@ static void run_finalize(const at::Tensor& c32, at::Tensor& out,
   switch (epi) {
     case 0: hipLaunchKernelGGL(finalize_kernel<0>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C, slices, sstride); break;
+                               total, C, slices, sstride, int(rezero)); break;
     case 1: hipLaunchKernelGGL(finalize_kernel<1>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C, slices, sstride); break;
+                               total, C, slices, sstride, int(rezero)); break;
     case 2: hipLaunchKernelGGL(finalize_kernel<2>, dim3(blocks), dim3(256), 0,
                                s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                               total, C, slices, sstride); break;
+                               total, C, slices, sstride, int(rezero)); break;
     default: hipLaunchKernelGGL(finalize_kernel<3>, dim3(blocks), dim3(256),
                                 0, s, c32.data_ptr<float>(), bfp_mut(out), bp,
-                                total, C, slices, sstride); break;
+                                total, C, slices, sstride, int(rezero)); break;
   }
 }
 
@@ -900,9 +922,9 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   int epi = epi_of(bias.has_value(), relu);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::zeros({M, N}, x.options().dtype(at::kFloat));
+    auto c32 = scratch_c32("fcf", M, N, x.options());
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, N, K, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, N, 1, true);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, N, K, 1);
   }
@@ -919,9 +941,9 @@ at::Tensor linear_dgrad(const at::Tensor& dy, const at::Tensor& w) {
   int tiles = ceil_div(M, BM) * ceil_div(K, BN);
   int sk = pick_splitk(tiles, N);
   if (sk > 1) {
-    auto c32 = at::zeros({M, K}, dy.options().dtype(at::kFloat));
+    auto c32 = scratch_c32("fcd", M, K, dy.options());
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, K, N, sk);
-    run_finalize(c32, dx, nullptr, 0, M, K);
+    run_finalize(c32, dx, nullptr, 0, M, K, 1, true);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, K, N, 1);
   }
@@ -945,9 +967,9 @@ at::Tensor linear_wgrad(const at::Tensor& x, const at::Tensor& dy,
   int tiles = ceil_div(K, BM) * ceil_div(N, BN);
   int sk = pick_splitk(tiles, Mred);
   if (sk > 1) {
-    auto c32 = at::zeros({K, N}, x.options().dtype(at::kFloat));
+    auto c32 = scratch_c32("fcw", K, N, x.options());
     launch_gemm_tn(ga, gb, nullptr, &c32, K, N, Mred, sk);
-    run_finalize(c32, dw, nullptr, 0, K, N);
+    run_finalize(c32, dw, nullptr, 0, K, N, 1, true);
   } else {
     launch_gemm_tn(ga, gb, &dw, nullptr, K, N, Mred, 1);
   }
@@ -1109,9 +1131,9 @@ static void wgrad_glds(const at::Tensor& x, const at::Tensor& dym, int KH,
   WgradDygGather gb{bfp(dyTS), Cout, N, OH, OWq, Kred, cstride, dOHWk, dWk};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = scratch_c32("wgp", Kfull, Cout, x.options());
   launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Kfull, Cout, Kred, sk);
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, 1, true);
 }
 
 // --------------- ring direct conv (conv_ring.h) -----------------------
@@ -1304,9 +1326,9 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   int tiles = ceil_div(M, BM) * ceil_div(Cout, BN);
   int sk = pick_splitk(tiles, Kfull);
   if (sk > 1) {
-    auto c32 = at::zeros({M, Cout}, x.options().dtype(at::kFloat));
+    auto c32 = scratch_c32("cvf", M, Cout, x.options());
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cout, Kfull, sk);
-    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout);
+    run_finalize(c32, y, bias ? &*bias : nullptr, epi, M, Cout, 1, true);
   } else {
     launch_gemm(ga, gb, &y, nullptr, bias ? &*bias : nullptr, epi, M, Cout,
                 Kfull, 1);
@@ -1347,9 +1369,9 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
   int tiles = ceil_div(M, BM) * ceil_div(Cin, BN);
   int sk = pick_splitk(tiles, K);
   if (sk > 1) {
-    auto c32 = at::zeros({M, Cin}, dy.options().dtype(at::kFloat));
+    auto c32 = scratch_c32("cvd", M, Cin, dy.options());
     launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, M, Cin, K, sk);
-    run_finalize(c32, dx, nullptr, 0, M, Cin);
+    run_finalize(c32, dx, nullptr, 0, M, Cin, 1, true);
   } else {
     launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, M, Cin, K, 1);
   }
@@ -1384,9 +1406,9 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
     int sk1 = pick_splitk(tiles1, Mred);
     auto dwt = dw2d.view({Cin, Cout});
     if (sk1 > 1) {
-      auto c32w = at::zeros({Cin, Cout}, x.options().dtype(at::kFloat));
+      auto c32w = scratch_c32("w11", Cin, Cout, x.options());
       launch_gemm_tn(ga, gb, nullptr, &c32w, Cin, Cout, Mred, sk1);
-      run_finalize(c32w, dwt, nullptr, 0, Cin, Cout);
+      run_finalize(c32w, dwt, nullptr, 0, Cin, Cout, 1, true);
     } else {
       launch_gemm_tn(ga, gb, &dwt, nullptr, Cin, Cout, Mred, 1);
     }
@@ -1409,7 +1431,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = scratch_c32("wgt", Kfull, Cout, x.options());
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
                   ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);
@@ -1439,7 +1461,7 @@ at::Tensor conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, int64_t KH_,
                     MDiv(u32(OWp)), 0};
     launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, 1, true);
   return dw;
 }
 
@@ -1492,9 +1514,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
       int tiles = ceil_div(Cin, BM) * ceil_div(Cout, BN);
       int sk = pick_splitk(tiles, Mred);
       if (sk > 1) {
-        auto c32w = at::zeros({Cin, Cout}, x.options().dtype(at::kFloat));
+        auto c32w = scratch_c32("w11b", Cin, Cout, x.options());
         launch_gemm_tn(ga, gb, nullptr, &c32w, Cin, Cout, Mred, sk);
-        run_finalize(c32w, dwt, nullptr, 0, Cin, Cout);
+        run_finalize(c32w, dwt, nullptr, 0, Cin, Cout, 1, true);
       } else {
         launch_gemm_tn(ga, gb, &dwt, nullptr, Cin, Cout, Mred, 1);
       }
@@ -1518,8 +1540,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   bool glds_path = wgrad_glds_ok(Cin, stride, KW, pad);
   bool write_dym = relu && (need_dx || glds_path);
   at::Tensor dym = write_dym ? at::empty_like(dy) : dy;
-  auto db32 = need_db ? at::zeros({256, Cout}, dy.options().dtype(at::kFloat))
-                      : at::Tensor();
+  auto db32 = need_db ? scratch_c32("dbs", 256, Cout, dy.options())
+                      : at::Tensor();  // bias_slot_reduce re-zeros it
   const u16* yp = relu ? bfp(y) : nullptr;
   float* dbp = need_db ? db32.data_ptr<float>() : nullptr;
   u16* dymp = write_dym ? bfp_mut(dym) : nullptr;
@@ -1578,9 +1600,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     int tiles = ceil_div(Mx, BM) * ceil_div(Cin, BN);
     int sk = pick_splitk(tiles, K);
     if (sk > 1) {
-      auto c32 = at::zeros({Mx, Cin}, dy.options().dtype(at::kFloat));
+      auto c32 = scratch_c32("d11", Mx, Cin, dy.options());
       launch_gemm(ga, gb, nullptr, &c32, nullptr, 0, Mx, Cin, K, sk);
-      run_finalize(c32, dx, nullptr, 0, Mx, Cin);
+      run_finalize(c32, dx, nullptr, 0, Mx, Cin, 1, true);
     } else {
       launch_gemm(ga, gb, &dx, nullptr, nullptr, 0, Mx, Cin, K, 1);
     }
@@ -1623,7 +1645,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   WgradDyGather gb{bfp(dyT), Cout, N, OH, OWp, Kred};
   int tiles = ceil_div(Kfull, BM) * ceil_div(Cout, BN);
   int sk = std::max(2, pick_splitk(tiles, Kred, 1024));
-  auto c32 = at::zeros({Kfull, Cout}, x.options().dtype(at::kFloat));
+  auto c32 = scratch_c32("wgt", Kfull, Cout, x.options());
   bool phase_ok = Cin % 8 != 0 && (stride == 2 || stride == 4) &&
                   ((KW - 1) / stride) < 8 && W * Cin <= 4096 &&
                   env_int("DVGG_WGRAD_PHASE", 1);
@@ -1653,7 +1675,7 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
                     MDiv(u32(OWp)), 0};
     launch_gemm_wgrad(ga, gb, &c32, Kfull, Cout, Kred, sk);
   }
-  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout);
+  run_finalize(c32, dw2d, nullptr, 0, Kfull, Cout, 1, true);
 
   at::Tensor db;
   if (need_db) {

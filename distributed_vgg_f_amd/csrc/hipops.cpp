@@ -36,10 +36,9 @@ std::vector<at::Tensor> dropout_fwd(const at::Tensor&, double, int64_t,
 at::Tensor dropout_bwd(const at::Tensor&, const at::Tensor&, double);
 
 // lrnpool.hip
-std::vector<at::Tensor> lrn_fwd(const at::Tensor&, int64_t, double, double,
-                                double);
-at::Tensor lrn_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&,
-                   const at::Tensor&, int64_t, double, double);
+at::Tensor lrn_fwd(const at::Tensor&, int64_t, double, double, double);
+at::Tensor lrn_bwd(const at::Tensor&, const at::Tensor&, int64_t, double,
+                   double, double);
 std::vector<at::Tensor> maxpool_fwd(const at::Tensor&, int64_t, int64_t,
                                     int64_t, int64_t);
 at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, int64_t, int64_t,

@@ -56,11 +56,14 @@ DEV float fast_exp2(float y) {
 }
 
 // ------------------------------------------------------------- LRN fwd
-// one thread per (n,h,w); C % 8 == 0. Saves sb, snb (bf16).
+// one thread per (n,h,w); C % 8 == 0. Saves NOTHING: backward recomputes
+// s (5-channel window sum of x^2, cheap FMAs over x it reloads anyway)
+// and sb/snb via the same fast_log2/fast_exp2 polynomials, so fwd is a
+// pure 2-stream (x in, y out) kernel and bwd a 3-stream one — vs the
+// round-1 design that stored+reloaded sb and snb (4- and 5-stream).
 __global__ void lrn_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
-                               u16* __restrict__ sb_out,
-                               u16* __restrict__ snb_out, i64 npos, int C,
-                               float an, float k, float beta) {
+                               i64 npos, int C, float an, float k,
+                               float beta) {
   // one thread per (position, 8-channel chunk): fully parallel, neighbor
   // chunks re-read through L1/L2
   i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
@@ -70,8 +73,8 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
   i64 pos = id / nch;
   const u16* xr = x + pos * C + ch * 8;
   float q[12];  // squares of channels ch*8-2 .. ch*8+9
+  s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
   {
-    s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = bf2f(u16(cur[j]));
@@ -88,52 +91,41 @@ __global__ void lrn_fwd_kernel(const u16* __restrict__ x, u16* __restrict__ y,
       q[10] = a * a; q[11] = b * b;
     }
   }
-  s16x8 cur = *reinterpret_cast<const s16x8*>(xr);
-  s16x8 yv, sbv, snbv;
+  s16x8 yv;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float sum = q[j] + q[j + 1] + q[j + 2] + q[j + 3] + q[j + 4];
     float sv = k + an * sum;
-    float ls = fast_log2(sv);
-    float sb = fast_exp2(-beta * ls);
-    // s^-(1+beta) = s^-beta / s: v_rcp_f32 (~1 ulp, fine for bf16 out)
-    // replaces the second exp2 polynomial
-    float snb = sb * __builtin_amdgcn_rcpf(sv);
+    float sb = fast_exp2(-beta * fast_log2(sv));
     yv[j] = short(f2bf(bf2f(u16(cur[j])) * sb));
-    sbv[j] = short(f2bf(sb));
-    snbv[j] = short(f2bf(snb));
   }
-  i64 o = pos * C + ch * 8;
-  *reinterpret_cast<s16x8*>(y + o) = yv;
-  *reinterpret_cast<s16x8*>(sb_out + o) = sbv;
-  *reinterpret_cast<s16x8*>(snb_out + o) = snbv;
+  *reinterpret_cast<s16x8*>(y + pos * C + ch * 8) = yv;
 }
 
-std::vector<at::Tensor> lrn_fwd(const at::Tensor& x, int64_t size,
-                                double alpha, double beta, double k) {
+at::Tensor lrn_fwd(const at::Tensor& x, int64_t size, double alpha,
+                   double beta, double k) {
   check_bf16(x, "x");
   TORCH_CHECK(size == 5, "LRN kernel is specialized for size=5");
   int C = x.size(-1);
   TORCH_CHECK(C % 8 == 0, "LRN needs C % 8 == 0");
   i64 npos = x.numel() / C;
   auto y = at::empty_like(x);
-  auto sb = at::empty_like(x);
-  auto snb = at::empty_like(x);
   i64 nthreads = npos * (C / 8);
   hipLaunchKernelGGL(lrn_fwd_kernel, dim3(int((nthreads + 255) / 256)),
-                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(y),
-                     bfp_mut(sb), bfp_mut(snb), npos, C, float(alpha / size),
-                     float(k), float(beta));
-  return {y, sb, snb};
+                     dim3(256), 0, cur_stream(), bfp(x), bfp_mut(y), npos, C,
+                     float(alpha / size), float(k), float(beta));
+  return y;
 }
 
 // ------------------------------------------------------------- LRN bwd
+// dx_c = dy_c * sb_c - coef * x_c * sum_{m in win(c)} dy_m * x_m * snb_m.
+// sb/snb recomputed from x: needs s for channels ch*8-2..ch*8+9, hence
+// x^2 for channels ch*8-4..ch*8+11 (16 values, +-4 neighbor reads served
+// by L1/L2 since adjacent chunks load those lines anyway).
 __global__ void lrn_bwd_kernel(const u16* __restrict__ dy,
                                const u16* __restrict__ x,
-                               const u16* __restrict__ sb,
-                               const u16* __restrict__ snb,
                                u16* __restrict__ dx, i64 npos, int C,
-                               float coef) {  // coef = 2*alpha*beta/size
+                               float an, float k, float beta, float coef) {
   i64 id = (i64)blockIdx.x * blockDim.x + threadIdx.x;
   int nch = C / 8;
   if (id >= npos * nch) return;
@@ -142,51 +134,69 @@ __global__ void lrn_bwd_kernel(const u16* __restrict__ dy,
   i64 o = pos * C + ch * 8;
   const u16* dyr = dy + o;
   const u16* xr = x + o;
-  const u16* snbr = snb + o;
-  float t[12];  // t = dy * x * snb for channels ch*8-2 .. ch*8+9
-  {
-    s16x8 dv = *reinterpret_cast<const s16x8*>(dyr);
-    s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
-    s16x8 sv = *reinterpret_cast<const s16x8*>(snbr);
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      t[2 + j] = bf2f(u16(dv[j])) * bf2f(u16(xv[j])) * bf2f(u16(sv[j]));
-    t[0] = t[1] = 0.f;
-    if (ch > 0) {
-      t[0] = bf2f(dyr[-2]) * bf2f(xr[-2]) * bf2f(snbr[-2]);
-      t[1] = bf2f(dyr[-1]) * bf2f(xr[-1]) * bf2f(snbr[-1]);
-    }
-    t[10] = t[11] = 0.f;
-    if (ch + 1 < nch) {
-      t[10] = bf2f(dyr[8]) * bf2f(xr[8]) * bf2f(snbr[8]);
-      t[11] = bf2f(dyr[9]) * bf2f(xr[9]) * bf2f(snbr[9]);
-    }
-  }
   s16x8 dv = *reinterpret_cast<const s16x8*>(dyr);
   s16x8 xv = *reinterpret_cast<const s16x8*>(xr);
-  s16x8 sbv = *reinterpret_cast<const s16x8*>(sb + o);
+  float q[16];    // x^2 for channels ch*8-4 .. ch*8+11
+  float dyx[12];  // dy*x for channels ch*8-2 .. ch*8+9 (0 out of range)
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float f = bf2f(u16(xv[j]));
+    q[4 + j] = f * f;
+    dyx[2 + j] = bf2f(u16(dv[j])) * f;
+  }
+  q[0] = q[1] = q[2] = q[3] = 0.f;
+  dyx[0] = dyx[1] = 0.f;
+  if (ch > 0) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float f = bf2f(xr[j - 4]);
+      q[j] = f * f;
+    }
+    dyx[0] = bf2f(dyr[-2]) * bf2f(xr[-2]);
+    dyx[1] = bf2f(dyr[-1]) * bf2f(xr[-1]);
+  }
+  q[12] = q[13] = q[14] = q[15] = 0.f;
+  dyx[10] = dyx[11] = 0.f;
+  if (ch + 1 < nch) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float f = bf2f(xr[8 + j]);
+      q[12 + j] = f * f;
+    }
+    dyx[10] = bf2f(dyr[8]) * bf2f(xr[8]);
+    dyx[11] = bf2f(dyr[9]) * bf2f(xr[9]);
+  }
+  float t[12];   // dy*x*snb for channels ch*8-2 .. ch*8+9
+  float sbj[8];  // sb for the own 8 channels
+#pragma unroll
+  for (int cc = 0; cc < 12; ++cc) {
+    // channel ch*8-2+cc: q window q[cc..cc+4] (q index m = channel+4-ch*8)
+    float sv = k + an * (q[cc] + q[cc + 1] + q[cc + 2] + q[cc + 3] +
+                         q[cc + 4]);
+    float sb = fast_exp2(-beta * fast_log2(sv));
+    t[cc] = dyx[cc] * sb * __builtin_amdgcn_rcpf(sv);
+    if (cc >= 2 && cc < 10) sbj[cc - 2] = sb;
+  }
   s16x8 out;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float wsum = t[j] + t[j + 1] + t[j + 2] + t[j + 3] + t[j + 4];
-    float v = bf2f(u16(dv[j])) * bf2f(u16(sbv[j])) -
-              coef * bf2f(u16(xv[j])) * wsum;
+    float v = bf2f(u16(dv[j])) * sbj[j] - coef * bf2f(u16(xv[j])) * wsum;
     out[j] = short(f2bf(v));
   }
   *reinterpret_cast<s16x8*>(dx + o) = out;
 }
 
-at::Tensor lrn_bwd(const at::Tensor& dy, const at::Tensor& x,
-                   const at::Tensor& sb, const at::Tensor& snb, int64_t size,
-                   double alpha, double beta) {
+at::Tensor lrn_bwd(const at::Tensor& dy, const at::Tensor& x, int64_t size,
+                   double alpha, double beta, double k) {
   check_bf16(dy, "dy");
   int C = dy.size(-1);
   i64 npos = dy.numel() / C;
   auto dx = at::empty_like(dy);
   i64 nthreads = npos * (C / 8);
   hipLaunchKernelGGL(lrn_bwd_kernel, dim3(int((nthreads + 255) / 256)),
-                     dim3(256), 0, cur_stream(), bfp(dy), bfp(x), bfp(sb),
-                     bfp(snb), bfp_mut(dx), npos, C,
+                     dim3(256), 0, cur_stream(), bfp(dy), bfp(x), bfp_mut(dx),
+                     npos, C, float(alpha / size), float(k), float(beta),
                      float(2.0 * alpha * beta / size));
   return dx;
 }

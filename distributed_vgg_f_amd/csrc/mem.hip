@@ -6,6 +6,8 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <unordered_map>
+
 #include "common.h"
 
 namespace dvgg {
@@ -119,10 +121,13 @@ __global__ void bias_grad_kernel(const u16* __restrict__ dy,
   atomicAdd(&db32[c], (a0 + a1) + (a2 + a3));
 }
 
-__global__ void cast_f32_bf16_kernel(const float* __restrict__ in,
+// cast + re-zero: db32 is a cached per-C scratch accumulator (atomics),
+// so zeroing it here (free: the line is already in L2) replaces a
+// launch-latency-bound Fill kernel before every fc bias grad.
+__global__ void cast_f32_bf16_kernel(float* __restrict__ in,
                                      u16* __restrict__ out, i64 n) {
   i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) out[i] = f2bf(in[i]);
+  if (i < n) { out[i] = f2bf(in[i]); in[i] = 0.f; }
 }
 
 at::Tensor bias_grad(const at::Tensor& dy,
@@ -130,7 +135,12 @@ at::Tensor bias_grad(const at::Tensor& dy,
   check_bf16(dy, "dy");
   int C = dy.size(-1);
   i64 R = dy.numel() / C;
-  auto db32 = at::zeros({C}, dy.options().dtype(at::kFloat));
+  static std::unordered_map<int, at::Tensor> cache;  // zeroed between uses
+  auto it = cache.find(C);
+  if (it == cache.end())
+    it = cache.emplace(C, at::zeros({C}, dy.options().dtype(at::kFloat)))
+             .first;
+  auto db32 = it->second;
   int rpb = 256;  // small row chunks: fill 256 CUs even at C=64
   dim3 grid(ceil_div(C, 256), ceil_div(R, rpb));
   hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0, cur_stream(),
@@ -148,21 +158,32 @@ at::Tensor bias_grad(const at::Tensor& dy,
 __global__ void sgd_kernel(float* __restrict__ w32, u16* __restrict__ wc,
                            const u16* __restrict__ g, float* __restrict__ m32,
                            i64 n, float lr, float mu, float wd, float scale) {
-  CHUNK_LOOP(n / 4) {
-    f32x4 w = *reinterpret_cast<f32x4*>(w32 + c * 4);
-    f32x4 m = *reinterpret_cast<f32x4*>(m32 + c * 4);
-    s16x4 gv = *reinterpret_cast<const s16x4*>(g + c * 4);
-    s16x4 wo;
+  // 8 elems per iteration: every access is a b128, and the two w/m pairs
+  // issue back-to-back for memory-level parallelism (20 B/elem streamed;
+  // the kernel is a pure HBM roofline problem)
+  CHUNK_LOOP(n / 8) {
+    f32x4 w0 = *reinterpret_cast<f32x4*>(w32 + c * 8);
+    f32x4 w1 = *reinterpret_cast<f32x4*>(w32 + c * 8 + 4);
+    f32x4 m0 = *reinterpret_cast<f32x4*>(m32 + c * 8);
+    f32x4 m1 = *reinterpret_cast<f32x4*>(m32 + c * 8 + 4);
+    s16x8 gv = *reinterpret_cast<const s16x8*>(g + c * 8);
+    s16x8 wo;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      float gj = bf2f(u16(gv[j])) * scale + wd * w[j];
-      m[j] = mu * m[j] + gj;
-      w[j] -= lr * m[j];
-      wo[j] = short(f2bf(w[j]));
+      float gj = bf2f(u16(gv[j])) * scale + wd * w0[j];
+      m0[j] = mu * m0[j] + gj;
+      w0[j] -= lr * m0[j];
+      wo[j] = short(f2bf(w0[j]));
+      float gk = bf2f(u16(gv[4 + j])) * scale + wd * w1[j];
+      m1[j] = mu * m1[j] + gk;
+      w1[j] -= lr * m1[j];
+      wo[4 + j] = short(f2bf(w1[j]));
     }
-    *reinterpret_cast<f32x4*>(w32 + c * 4) = w;
-    *reinterpret_cast<f32x4*>(m32 + c * 4) = m;
-    *reinterpret_cast<s16x4*>(wc + c * 4) = wo;
+    *reinterpret_cast<f32x4*>(w32 + c * 8) = w0;
+    *reinterpret_cast<f32x4*>(w32 + c * 8 + 4) = w1;
+    *reinterpret_cast<f32x4*>(m32 + c * 8) = m0;
+    *reinterpret_cast<f32x4*>(m32 + c * 8 + 4) = m1;
+    *reinterpret_cast<s16x8*>(wc + c * 8) = wo;
   }
 }
 
@@ -173,8 +194,8 @@ void sgd_step(at::Tensor& w32, at::Tensor& wc, const at::Tensor& g,
   TORCH_CHECK(w32.is_cuda() && w32.scalar_type() == at::kFloat);
   check_bf16(wc, "wc"); check_bf16(g, "g");
   i64 n = w32.numel();
-  TORCH_CHECK(n % 4 == 0, "arena must be 4-aligned");  // _ALIGN=128 in arena
-  hipLaunchKernelGGL(sgd_kernel, dim3(grid_for(n / 4)), dim3(256), 0,
+  TORCH_CHECK(n % 8 == 0, "arena must be 8-aligned");  // _ALIGN=128 in arena
+  hipLaunchKernelGGL(sgd_kernel, dim3(grid_for(n / 8)), dim3(256), 0,
                      cur_stream(), w32.data_ptr<float>(), bfp_mut(wc), bfp(g),
                      m32.data_ptr<float>(), n, float(lr), float(mu),
                      float(wd), float(scale));

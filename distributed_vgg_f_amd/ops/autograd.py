@@ -90,21 +90,23 @@ class LinearFunction(torch.autograd.Function):
 
 
 class LRNFunction(torch.autograd.Function):
-    """K4/K5. Fwd saves sb = s^-beta and snb = s^-(1+beta) (bf16) so the
-    backward pass needs no transcendentals (see csrc/lrnpool.hip)."""
+    """K4/K5. Fwd saves only x: backward recomputes the window scale s
+    (cheap FMAs over the x it reloads anyway) and sb/snb via the same
+    fast_log2/fast_exp2 polynomials — 2-stream fwd, 3-stream bwd instead
+    of storing+reloading sb/snb (csrc/lrnpool.hip)."""
 
     @staticmethod
     def forward(ctx, x, size: int, alpha: float, beta: float, k: float):
-        y, sb, snb = _ext().lrn_fwd(x, size, alpha, beta, k)
-        ctx.save_for_backward(x, sb, snb)
-        ctx.params = (size, alpha, beta)
+        y = _ext().lrn_fwd(x, size, alpha, beta, k)
+        ctx.save_for_backward(x)
+        ctx.params = (size, alpha, beta, k)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, sb, snb = ctx.saved_tensors
-        size, alpha, beta = ctx.params
-        dx = _ext().lrn_bwd(dy.contiguous(), x, sb, snb, size, alpha, beta)
+        (x,) = ctx.saved_tensors
+        size, alpha, beta, k = ctx.params
+        dx = _ext().lrn_bwd(dy.contiguous(), x, size, alpha, beta, k)
         return dx, None, None, None, None
 
 

@@ -133,11 +133,11 @@ def test_conv_wgrad(N, H, W, Ci, Co, k, s, p, name):
 def test_lrn_fwd_bwd():
     torch.manual_seed(20)
     x = rnd(2, 9, 9, 64, scale=2.0)
-    y, sb, snb = ext.lrn_fwd(x.to(DEV), 5, 1e-4, 0.75, 2.0)
+    y = ext.lrn_fwd(x.to(DEV), 5, 1e-4, 0.75, 2.0)
     ref = cpu_ref.lrn(x.float(), 5, 1e-4, 0.75, 2.0)
     check(y, ref, 0.03, "lrn_fwd")
     dy = rnd(2, 9, 9, 64, seed=21)
-    dx = ext.lrn_bwd(dy.to(DEV), x.to(DEV), sb, snb, 5, 1e-4, 0.75)
+    dx = ext.lrn_bwd(dy.to(DEV), x.to(DEV), 5, 1e-4, 0.75, 2.0)
     x32 = x.float().requires_grad_(True)
     cpu_ref.lrn(x32, 5, 1e-4, 0.75, 2.0).backward(dy.float())
     check(dx, x32.grad, 0.05, "lrn_bwd")
@@ -147,7 +147,7 @@ def test_lrn_large_alpha():
     # make the normalizer actually bite (s >> k) to exercise the poly path
     torch.manual_seed(22)
     x = rnd(1, 5, 5, 128, scale=4.0)
-    y, _, _ = ext.lrn_fwd(x.to(DEV), 5, 0.5, 0.75, 1.0)
+    y = ext.lrn_fwd(x.to(DEV), 5, 0.5, 0.75, 1.0)
     check(y, cpu_ref.lrn(x.float(), 5, 0.5, 0.75, 1.0), 0.04, "lrn_bite")
 
 
