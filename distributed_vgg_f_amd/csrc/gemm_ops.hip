@@ -1354,13 +1354,16 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H_,
     return dx;
   }
   // Bt[ci][(kh,kw,co)] = w[kh][kw][ci][co]: permute HWIO -> [Cin][KhKw*Cout]
-  auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
-    auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
-    if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
-    auto t = p.contiguous();
-    if (t.data_ptr() == w.data_ptr()) t = t.clone();  // see conv2d_bwd note
-    return t.view({Cin, (i64)KH * KW * Cout});
-  });
+  // KH*KW==1: [1][Cin][Cout] permuted is w itself — a free view, and no
+  // cached copy to refresh every step (was a DtoD blit per layer per step)
+  auto wD = (KH * KW == 1)
+      ? w.view({Cin, (i64)Cout})
+      : cached_weight_form(1, cache_w, w,
+                           [&](c10::optional<at::Tensor> prev) {
+          auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
+          if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
+          return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+        });
   TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
   ConvDgradGather ga{bfp(dy), M, H, W, Cout, OH, OW, stride, pad,
                      MDiv(u32(H * W)), MDiv(u32(W)), MDiv(u32(Cout)),
@@ -1582,15 +1585,17 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
       goto dgrad_done;
     }
     {
-    auto wD = cached_weight_form(1, cache_w, w, [&](c10::optional<at::Tensor> prev) {
-      auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
-      if (prev) { prev->view({Cin, KH * KW, Cout}).copy_(p); return *prev; }
-      auto t = p.contiguous();
-      // KH*KW==1: contiguous() returns a VIEW of w — clone so the cache
-      // entry owns its storage (the refresh copy_ must not alias w)
-      if (t.data_ptr() == w.data_ptr()) t = t.clone();
-      return t.view({Cin, (i64)KH * KW * Cout});
-    });
+    auto wD = (KH * KW == 1)  // free view of w; nothing to cache/refresh
+        ? w.view({Cin, (i64)Cout})
+        : cached_weight_form(1, cache_w, w,
+                             [&](c10::optional<at::Tensor> prev) {
+            auto p = w.view({KH * KW, Cin, Cout}).permute({1, 0, 2});
+            if (prev) {
+              prev->view({Cin, KH * KW, Cout}).copy_(p);
+              return *prev;
+            }
+            return p.contiguous().view({Cin, (i64)KH * KW * Cout});
+          });
     TORCH_CHECK(stride == 1 || stride == 2, "dgrad supports stride 1/2");
     int Mx = N * H * W;
     ConvDgradGather ga{bfp(dym), Mx, H, W, Cout, OH, OW, stride, pad,

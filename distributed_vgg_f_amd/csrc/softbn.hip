@@ -4,6 +4,8 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <unordered_map>
+
 #include "common.h"
 
 namespace dvgg {
@@ -109,8 +111,8 @@ __global__ void bn_reduce_kernel(const u16* __restrict__ x,
   atomicAdd(&sumsq[(i64)slot * C + c], q0 + q1);
 }
 
-__global__ void bn_stats_kernel(const float* __restrict__ sum,
-                                const float* __restrict__ sumsq,
+__global__ void bn_stats_kernel(float* __restrict__ sum,
+                                float* __restrict__ sumsq,
                                 const u16* __restrict__ gamma,
                                 const u16* __restrict__ beta,
                                 float* __restrict__ mean,
@@ -127,6 +129,11 @@ __global__ void bn_stats_kernel(const float* __restrict__ sum,
   for (int t = 0; t < 256; ++t) {
     sa += sum[(i64)t * C + c];
     qa += sumsq[(i64)t * C + c];
+  }
+#pragma unroll 8
+  for (int t = 0; t < 256; ++t) {  // cached_part contract: leave zeroed
+    sum[(i64)t * C + c] = 0.f;
+    sumsq[(i64)t * C + c] = 0.f;
   }
   float m = sa / float(R);
   float v = fmaxf(qa / float(R) - m * m, 0.f);
@@ -172,6 +179,20 @@ __global__ void bn_apply_kernel(const u16* __restrict__ x,
   *reinterpret_cast<s16x8*>(y + o) = out;
 }
 
+
+// Cached {2,256,C} fp32 partial-slot buffers for the BN reduces: the
+// stats kernel re-zeros the slots after summing them, so the per-layer
+// at::zeros Fill launch (2/layer/step on ResNet-50 = ~106 launch-latency
+// fills per step) disappears. tag 0 = fwd sum/sumsq, 1 = bwd dy/dyx.
+static at::Tensor cached_part(int tag, int C, const at::TensorOptions& f32) {
+  static std::unordered_map<i64, at::Tensor> cache;
+  i64 key = (i64)tag << 32 | (u32)C;
+  auto it = cache.find(key);
+  if (it == cache.end())
+    it = cache.emplace(key, at::zeros({2, 256, C}, f32)).first;
+  return it->second;
+}
+
 std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
                                const at::Tensor& beta, double eps, bool relu,
                                const c10::optional<at::Tensor>& run_mean,
@@ -183,7 +204,7 @@ std::vector<at::Tensor> bn_fwd(const at::Tensor& x, const at::Tensor& gamma,
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = x.numel() / C;
   auto f32 = x.options().dtype(at::kFloat);
-  auto part = at::zeros({2, 256, C}, f32);   // one fill for both partials
+  auto part = cached_part(0, C, f32);        // bn_stats re-zeros it
   auto sum = part[0], sumsq = part[1];
   auto stats = at::empty({4, C}, f32);       // mean/rstd/scale/shift
   auto mean = stats[0], rstd = stats[1], scale = stats[2], shift = stats[3];
@@ -273,8 +294,8 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
 // per-channel dx coefficients (dx = A*dy_masked + B*x + D), replacing two
 // slot-reduce launches + a param-grad launch + 40 scalar loads per 8
 // elements in the old apply kernel.
-__global__ void bn_bwd_stats_kernel(const float* __restrict__ sdyp,
-                                    const float* __restrict__ sdyxp,
+__global__ void bn_bwd_stats_kernel(float* __restrict__ sdyp,
+                                    float* __restrict__ sdyxp,
                                     const u16* __restrict__ gamma,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ rstd,
@@ -291,6 +312,11 @@ __global__ void bn_bwd_stats_kernel(const float* __restrict__ sdyp,
   for (int t = 0; t < 256; ++t) {
     a += sdyp[(i64)t * C + c];
     b += sdyxp[(i64)t * C + c];
+  }
+#pragma unroll 8
+  for (int t = 0; t < 256; ++t) {  // cached_part contract: leave zeroed
+    sdyp[(i64)t * C + c] = 0.f;
+    sdyxp[(i64)t * C + c] = 0.f;
   }
   dgamma[c] = f2bf(b);
   dbeta[c] = f2bf(a);
@@ -346,7 +372,7 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& dy, const at::Tensor& x,
   TORCH_CHECK(C % 8 == 0, "bn needs C % 8 == 0");
   i64 R = dy.numel() / C;
   auto f32 = dy.options().dtype(at::kFloat);
-  auto bpart = at::zeros({2, 256, C}, f32);
+  auto bpart = cached_part(1, C, f32);       // bn_bwd_stats re-zeros it
   auto sdyp = bpart[0], sdyxp = bpart[1];
   auto coef = at::empty({3, C}, f32);
   auto cA = coef[0], cB = coef[1], cD = coef[2];

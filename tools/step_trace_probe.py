@@ -29,24 +29,23 @@ def main():
     torch.cuda.synchronize()
     acts = [torch.profiler.ProfilerActivity.CPU,
             torch.profiler.ProfilerActivity.CUDA]
-    with torch.profiler.profile(activities=acts, with_stack=True) as prof:
+    with torch.profiler.profile(activities=acts, with_stack=True,
+                                with_modules=True) as prof:
         for _ in range(args.steps):
             tr.train_step()
         torch.cuda.synchronize()
-    evs = prof.key_averages(group_by_stack_n=12)
+    evs = prof.key_averages(group_by_stack_n=20)
     rows = []
     for e in evs:
         name = e.key
         if not ("fill" in name.lower() or "copy" in name.lower() or
                 "Memcpy" in name or "zero" in name.lower()):
             continue
-        frames = [f for f in (e.stack or [])
-                  if "distributed_vgg_f_amd" in f or "bench" in f]
-        rows.append((e.count, name[:60], frames[:3]))
+        rows.append((e.count, name[:70], list(e.stack or [])[:6]))
     rows.sort(reverse=True)
     print(f"== fill/copy events over {args.steps} steps "
-          f"(count, kernel, repo frames)")
-    for cnt, name, frames in rows[:40]:
+          f"(count, kernel, stack)")
+    for cnt, name, frames in rows[:50]:
         print(f"{cnt:5d}x {name}")
         for f in frames:
             print(f"         {f}")
