@@ -26,6 +26,19 @@ def _ext():
     return native.ext()
 
 
+
+def _fresh(t):
+    """Return a fresh view (new TensorImpl, use_count 1) of an
+    arena-homed gradient. AccumulateGrad only STORES an incoming grad
+    whose TensorImpl is not referenced elsewhere; the tensors our
+    kernels write (and ctx holds) fail that check, so torch silently
+    CLONED every param grad — one DtoD blit per param per step (146 on
+    ResNet-50). The alias passes the use-count check and .grad becomes
+    the arena view itself (verified by tests/test_gpu_engine
+    direct-grad assertion)."""
+    return t.view_as(t) if t is not None else None
+
+
 class ConvFunction(torch.autograd.Function):
     """K1/K2/K3 + fused bias/ReLU. x [N,H,W,C] bf16, w [kh,kw,Ci,Co] bf16."""
 
@@ -55,8 +68,8 @@ class ConvFunction(torch.autograd.Function):
         dx, dw, db = _ext().conv2d_bwd(dy.contiguous(), y, x, w, ctx.stride,
                                        ctx.pad, ctx.relu, need_dx, need_db,
                                        ctx.gw, ctx.gb, ctx.cw)
-        return (dx if need_dx else None, dw, db if need_db else None,
-                None, None, None)
+        return (dx if need_dx else None, _fresh(dw),
+                _fresh(db) if need_db else None, None, None, None)
 
 
 class LinearFunction(torch.autograd.Function):
@@ -86,7 +99,7 @@ class LinearFunction(torch.autograd.Function):
             dw = _ext().linear_wgrad(x, dy, ctx.gw)  # x^T @ dy, into arena
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _ext().bias_grad(dy, ctx.gb)
-        return dx, dw, db, None
+        return dx, _fresh(dw), _fresh(db), None
 
 
 class LRNFunction(torch.autograd.Function):
@@ -184,7 +197,7 @@ class BatchNormFunction(torch.autograd.Function):
         # ReLU mask fused into the BN backward kernels (no separate pass)
         dx, dgamma, dbeta = _ext().bn_bwd(dy.contiguous(), x, gamma, mean,
                                           rstd, y, ctx.relu, ctx.gg, ctx.gb)
-        return dx, dgamma, dbeta, None, None, None, None, None
+        return dx, _fresh(dgamma), _fresh(dbeta), None, None, None, None, None
 
 
 class BNAddReluFunction(torch.autograd.Function):
@@ -209,7 +222,7 @@ class BNAddReluFunction(torch.autograd.Function):
         dx, dgamma, dbeta = _ext().bn_bwd(da, x, gamma, mean, rstd,
                                           torch.empty(0, device=x.device),
                                           False, ctx.gg, ctx.gb)
-        return dx, dgamma, dbeta, da, None, None, None, None
+        return dx, _fresh(dgamma), _fresh(dbeta), da, None, None, None, None
 
 
 class AddReluFunction(torch.autograd.Function):

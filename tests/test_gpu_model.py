@@ -69,6 +69,28 @@ def test_trainer_gpu_loss_decreases():
     assert all(l == l for l in losses), losses  # no NaN
 
 
+def test_direct_grad_aliases_arena():
+    """After backward every param.grad must BE its arena view (storage
+    alias), not an AccumulateGrad clone: the all-reduce and sgd read the
+    arena, and a silent clone costs one DtoD blit per param per step
+    (ops/autograd.py _fresh)."""
+    for name in ("vggf", "resnet50"):
+        cfg = Config(model=name, batch_size=4, train_steps=1, log_every=0)
+        t = Trainer(cfg, device=DEV)
+        t.train_step()
+        t.arena.detach_grads()
+        images, labels = t.pipeline.next()
+        from distributed_vgg_f_amd import ops as _ops
+        loss = _ops.softmax_xent(t.model(images), labels).sum()
+        loss.backward()
+        g_base = t.arena.g.data_ptr()
+        g_end = g_base + t.arena.g.numel() * t.arena.g.element_size()
+        missing = [s.name for s in t.arena.slots
+                   if s.param.grad is None or
+                   not (g_base <= s.param.grad.data_ptr() < g_end)]
+        assert not missing, f"{name}: grads cloned off-arena: {missing}"
+
+
 def test_trainer_gpu_vgg16_resnet50_step():
     for name in ("vgg16", "resnet50"):
         cfg = Config(model=name, batch_size=4, train_steps=1, log_every=0)
