@@ -91,24 +91,42 @@ std::vector<at::Tensor> softmax_xent_fwd(const at::Tensor& logits,
 // Pass 2: normalize (+affine, optional fused ReLU); saves mean/rstd fp32.
 // partials land in 256 slots per channel (slot = blockIdx.y & 255):
 // direct atomics onto C floats from thousands of blocks serialize.
+// C >= 256: thread per channel (whole block live). C < 256 (power of
+// 2): threads fold into W = 256/C row sub-stripes per channel — without
+// this 3 of 4 waves exited instantly on C=64/128 layers and the reduce
+// ran ~4x short of the memory-level parallelism the HBM latency needs.
+// Sub-stripe partials combine through LDS, one atomic per channel.
 __global__ void bn_reduce_kernel(const u16* __restrict__ x,
                                  float* __restrict__ sum,
                                  float* __restrict__ sumsq, i64 R, int C,
                                  int rows_per_block) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float ls[256], lq[256];
+  int t = threadIdx.x;
+  int W = 1, c = blockIdx.x * blockDim.x + t, w = 0;
+  if (C < 256 && (C & (C - 1)) == 0) {
+    W = 256 / C; c = t & (C - 1); w = t / C;
+  } else if (c >= C) {
+    return;
+  }
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
   float s0 = 0.f, s1 = 0.f, q0 = 0.f, q1 = 0.f;
-  i64 r = r0;
-  for (; r + 2 <= r1; r += 2) {
+  i64 r = r0 + 2 * w;                     // stripes of 2 rows, stride 2W
+  for (; r + 2 <= r1; r += 2 * W) {
     float v0 = bf2f(x[r * C + c]), v1 = bf2f(x[(r + 1) * C + c]);
     s0 += v0; q0 += v0 * v0; s1 += v1; q1 += v1 * v1;
   }
-  for (; r < r1; ++r) { float v = bf2f(x[r * C + c]); s0 += v; q0 += v * v; }
+  if (r < r1) { float v = bf2f(x[r * C + c]); s0 += v; q0 += v * v; }
+  float sa = s0 + s1, qa = q0 + q1;
+  if (W > 1) {
+    ls[t] = sa; lq[t] = qa;
+    __syncthreads();
+    if (t >= C) return;
+    for (int j = 1; j < W; ++j) { sa += ls[t + j * C]; qa += lq[t + j * C]; }
+  }
   int slot = blockIdx.y & 255;
-  atomicAdd(&sum[(i64)slot * C + c], s0 + s1);
-  atomicAdd(&sumsq[(i64)slot * C + c], q0 + q1);
+  atomicAdd(&sum[(i64)slot * C + c], sa);
+  atomicAdd(&sumsq[(i64)slot * C + c], qa);
 }
 
 __global__ void bn_stats_kernel(float* __restrict__ sum,
@@ -262,14 +280,20 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
                                      float* __restrict__ sdy,
                                      float* __restrict__ sdyx, i64 R, int C,
                                      int rows_per_block) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float la[256], lb[256];
+  int t = threadIdx.x;
+  int W = 1, c = blockIdx.x * blockDim.x + t, w = 0;
+  if (C < 256 && (C & (C - 1)) == 0) {      // see bn_reduce_kernel note
+    W = 256 / C; c = t & (C - 1); w = t / C;
+  } else if (c >= C) {
+    return;
+  }
   i64 r0 = (i64)blockIdx.y * rows_per_block;
   i64 r1 = min(R, r0 + rows_per_block);
   float a0 = 0.f, a1 = 0.f, b0 = 0.f, b1 = 0.f;
   float m = mean[c], rs = rstd[c];
-  i64 r = r0;
-  for (; r + 2 <= r1; r += 2) {
+  i64 r = r0 + 2 * w;
+  for (; r + 2 <= r1; r += 2 * W) {
     float d0 = bf2f(dy[r * C + c]), d1 = bf2f(dy[(r + 1) * C + c]);
     if (RELU) {
       d0 = bf2f(yr[r * C + c]) > 0.f ? d0 : 0.f;
@@ -279,15 +303,22 @@ __global__ void bn_bwd_reduce_kernel(const u16* __restrict__ dy,
     float h1 = (bf2f(x[(r + 1) * C + c]) - m) * rs;
     a0 += d0; b0 += d0 * h0; a1 += d1; b1 += d1 * h1;
   }
-  for (; r < r1; ++r) {
+  if (r < r1) {
     float d = bf2f(dy[r * C + c]);
     if (RELU) d = bf2f(yr[r * C + c]) > 0.f ? d : 0.f;
     float h = (bf2f(x[r * C + c]) - m) * rs;
     a0 += d; b0 += d * h;
   }
+  float aa = a0 + a1, bb = b0 + b1;
+  if (W > 1) {
+    la[t] = aa; lb[t] = bb;
+    __syncthreads();
+    if (t >= C) return;
+    for (int j = 1; j < W; ++j) { aa += la[t + j * C]; bb += lb[t + j * C]; }
+  }
   int slot = blockIdx.y & 255;
-  atomicAdd(&sdy[(i64)slot * C + c], a0 + a1);
-  atomicAdd(&sdyx[(i64)slot * C + c], b0 + b1);
+  atomicAdd(&sdy[(i64)slot * C + c], aa);
+  atomicAdd(&sdyx[(i64)slot * C + c], bb);
 }
 
 // One pass over the 256-way partials: emits dgamma/dbeta and the three
