@@ -140,18 +140,30 @@ __global__ void bn_stats_kernel(float* __restrict__ sum,
                                 float* __restrict__ run_mean,
                                 float* __restrict__ run_var, i64 R, int C,
                                 float eps, float momentum) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float ls[256], lq[256];
+  int tid = threadIdx.x;
+  int W = 1, c = blockIdx.x * blockDim.x + tid, w = 0;
+  if (C < 256 && (C & (C - 1)) == 0) {    // see bn_reduce_kernel note
+    W = 256 / C; c = tid & (C - 1); w = tid / C;
+  } else if (c >= C) {
+    return;
+  }
   float sa = 0.f, qa = 0.f;
 #pragma unroll 8
-  for (int t = 0; t < 256; ++t) {
+  for (int t = w; t < 256; t += W) {
     sa += sum[(i64)t * C + c];
     qa += sumsq[(i64)t * C + c];
   }
 #pragma unroll 8
-  for (int t = 0; t < 256; ++t) {  // cached_part contract: leave zeroed
+  for (int t = w; t < 256; t += W) {  // cached_part contract: leave zeroed
     sum[(i64)t * C + c] = 0.f;
     sumsq[(i64)t * C + c] = 0.f;
+  }
+  if (W > 1) {
+    ls[tid] = sa; lq[tid] = qa;
+    __syncthreads();
+    if (tid >= C) return;
+    for (int j = 1; j < W; ++j) { sa += ls[tid + j * C]; qa += lq[tid + j * C]; }
   }
   float m = sa / float(R);
   float v = fmaxf(qa / float(R) - m * m, 0.f);
@@ -336,18 +348,30 @@ __global__ void bn_bwd_stats_kernel(float* __restrict__ sdyp,
                                     u16* __restrict__ dgamma,
                                     u16* __restrict__ dbeta, int C,
                                     float invR) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float la[256], lb[256];
+  int tid = threadIdx.x;
+  int W = 1, c = blockIdx.x * blockDim.x + tid, w = 0;
+  if (C < 256 && (C & (C - 1)) == 0) {    // see bn_reduce_kernel note
+    W = 256 / C; c = tid & (C - 1); w = tid / C;
+  } else if (c >= C) {
+    return;
+  }
   float a = 0.f, b = 0.f;
 #pragma unroll 8
-  for (int t = 0; t < 256; ++t) {
+  for (int t = w; t < 256; t += W) {
     a += sdyp[(i64)t * C + c];
     b += sdyxp[(i64)t * C + c];
   }
 #pragma unroll 8
-  for (int t = 0; t < 256; ++t) {  // cached_part contract: leave zeroed
+  for (int t = w; t < 256; t += W) {  // cached_part contract: leave zeroed
     sdyp[(i64)t * C + c] = 0.f;
     sdyxp[(i64)t * C + c] = 0.f;
+  }
+  if (W > 1) {
+    la[tid] = a; lb[tid] = b;
+    __syncthreads();
+    if (tid >= C) return;
+    for (int j = 1; j < W; ++j) { a += la[tid + j * C]; b += lb[tid + j * C]; }
   }
   dgamma[c] = f2bf(b);
   dbeta[c] = f2bf(a);
