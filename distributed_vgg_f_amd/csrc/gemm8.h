@@ -20,7 +20,17 @@ namespace dvgg {
 
 constexpr int G8_BM = 256, G8_BN = 256, G8_BK = 64, G8_THREADS = 512;
 
-DEV int sw16x32(int byte) { return byte ^ (((byte >> 9) & 1) << 5); }
+// LDS elem offset for (row, kf8) in a [rows][32]-elem half with the 16 B
+// segment XOR-swizzled by (row>>2)&3: a 16-lane b128 fragment read covers
+// rows r..r+15 at one kf8 — unswizzled, rows r, r+4, r+8, r+12 land on the
+// same bank quad (64 B rows: bank = row*16 + col dw mod 64), a structural
+// 4-way conflict on EVERY fragment read (measured 37% MFMA-busy). The
+// swizzle sends the 4 row-quads to the 4 distinct 16 B segments, so the 16
+// lanes cover all 64 banks exactly once. glds writes stay 16 B-aligned
+// (the XOR permutes whole segments) and remain conflict-free.
+DEV int g8sw(int row, int kf8) {
+  return row * 32 + (kf8 ^ (((row >> 2) & 3) << 3));
+}
 
 template <class GA, class GB, int EPI, bool SPLITK>
 __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
@@ -67,7 +77,11 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
 #pragma unroll
   for (int g = 0; g < 2; ++g) {
     int off = (wave * 2 + g) * 1024 + lane * 16;   // byte in 16 KB half
-    int row = off / 64, k8 = (off % 64) / 2;
+    int row = off / 64;
+    // glds LDS dests are hardware-fixed at (wave-uniform base + lane*16):
+    // to realize the swizzled layout each lane FETCHES the source segment
+    // that belongs at its fixed dest (seg_src = seg_dst ^ ((row>>2)&3))
+    int k8 = ((off % 64) ^ (((row >> 2) & 3) << 4)) / 2;
     rsa[g] = ga.prep(m0 + row);
     ak8[g] = k8;
     aoff[g] = off / 2;                             // elem offset in A half
@@ -100,7 +114,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8_kernel(
 // exactly covers "everything but the pair just issued".
 #define G8_RD(dst, base, idx16) \
   dst = as_bf16x8(*reinterpret_cast<const s16x8*>((base) + (idx16)))
-#define G8_PHASE(pbuf, h, WAITN, STAGE_STMT)                                   do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8_RD(bf[j], Bs_, (wn + j * 16 + (lane & 15)) * 32 + kfrag);               G8_RD(af[0], As_, (wm + (lane & 15)) * 32 + kfrag);                          G8_RD(af[1], As_, (wm + 16 + (lane & 15)) * 32 + kfrag);                     asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8_RD(af[2 * g + 2], As_,                                                        (wm + (2 * g + 2) * 16 + (lane & 15)) * 32 + kfrag);                   G8_RD(af[2 * g + 3], As_,                                                        (wm + (2 * g + 3) * 16 + (lane & 15)) * 32 + kfrag);                   asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 2 * g; i < 2 * g + 2; ++i)                                        _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
+#define G8_PHASE(pbuf, h, WAITN, STAGE_STMT)                                   do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8_HALF(pbuf, h);                                           const u16* Bs_ = As_ + HE;                                                   bf16x8 af[8], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8_RD(bf[j], Bs_, g8sw(wn + j * 16 + (lane & 15), kfrag));               G8_RD(af[0], As_, g8sw(wm + (lane & 15), kfrag));                          G8_RD(af[1], As_, g8sw(wm + 16 + (lane & 15), kfrag));                     asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8_RD(af[2 * g + 2], As_,                                                        g8sw(wm + (2 * g + 2) * 16 + (lane & 15), kfrag));                   G8_RD(af[2 * g + 3], As_,                                                        g8sw(wm + (2 * g + 3) * 16 + (lane & 15), kfrag));                   asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int i = 2 * g; i < 2 * g + 2; ++i)                                        _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[i], bf[j], acc[i][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
 
   // prologue: T0 both halves in flight
   G8_STAGE(0, 0, k_lo);
@@ -186,7 +200,8 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8n_kernel(
   for (int g = 0; g < 2; ++g) {
     int off = (wave * 2 + g) * 1024 + lane * 16;   // byte in 16 KB A half
     rsa[g] = ga.prep(m0 + off / 64);
-    ak8[g] = (off % 64) / 2;
+    // source-side swizzle; dest stays uniform+lane*16 (see gemm8_kernel)
+    ak8[g] = ((off % 64) ^ (((off / 64 >> 2) & 3) << 4)) / 2;
     aoff[g] = off / 2;
   }
   typename GB::Row rsb;
@@ -194,7 +209,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8n_kernel(
   {
     int off = wave * 1024 + lane * 16;             // byte in 8 KB B half
     rsb = gb.prep(n0 + off / 64);
-    bk8 = (off % 64) / 2;
+    bk8 = ((off % 64) ^ (((off / 64 >> 2) & 3) << 4)) / 2;
     boff = off / 2;
   }
   int tap = 0;
@@ -214,7 +229,7 @@ __global__ __launch_bounds__(G8_THREADS) void gemm8n_kernel(
   const int kfrag = (lane >> 4) * 8;
 
 #define G8N_RD(dst, base, idx16)                                               dst = as_bf16x8(*reinterpret_cast<const s16x8*>((base) + (idx16)))
-#define G8N_PHASE(pbuf, h, WAITN, STAGE_STMT)                                  do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8N_HALF(pbuf, h);                                          const u16* Bs_ = As_ + HEA;                                                  bf16x8 af[4], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8N_RD(bf[j], Bs_, (wn + j * 16 + (lane & 15)) * 32 + kfrag);              G8N_RD(af[0], As_, (wm + (lane & 15)) * 32 + kfrag);                         asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8N_RD(af[g + 1], As_,                                                           (wm + (g + 1) * 16 + (lane & 15)) * 32 + kfrag);                       asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[g][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[g], bf[j], acc[g][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
+#define G8N_PHASE(pbuf, h, WAITN, STAGE_STMT)                                  do {                                                                           asm volatile("s_waitcnt vmcnt(" #WAITN ")" ::: "memory");                    __builtin_amdgcn_s_barrier();                                                const u16* As_ = G8N_HALF(pbuf, h);                                          const u16* Bs_ = As_ + HEA;                                                  bf16x8 af[4], bf[4];                                                         _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    G8N_RD(bf[j], Bs_, g8sw(wn + j * 16 + (lane & 15), kfrag));              G8N_RD(af[0], As_, g8sw(wm + (lane & 15), kfrag));                         asm volatile("" ::: "memory");                                               STAGE_STMT;                                                                  _Pragma("unroll")                                                            for (int g = 0; g < 4; ++g) {                                                  if (g < 3) {                                                                   G8N_RD(af[g + 1], As_,                                                           g8sw(wm + (g + 1) * 16 + (lane & 15), kfrag));                       asm volatile("s_waitcnt lgkmcnt(1)" ::: "memory");                         } else {                                                                     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         }                                                                            __builtin_amdgcn_s_setprio(1);                                               _Pragma("unroll")                                                            for (int j = 0; j < 4; ++j)                                                    acc[g][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                             af[g], bf[j], acc[g][j], 0, 0, 0);                                   __builtin_amdgcn_s_setprio(0);                                             }                                                                          } while (0)
 
   G8N_STAGE(0, 0, 0);
   G8N_STAGE(0, 1, 0);
