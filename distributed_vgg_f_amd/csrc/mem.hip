@@ -187,6 +187,43 @@ __global__ void sgd_kernel(float* __restrict__ w32, u16* __restrict__ wc,
   }
 }
 
+// Nontemporal variant (DVGG_SGD_NT=1): every stream is touch-once, so
+// the hints bypass L2 allocation for the 1.2 GB pass.
+__global__ void sgd_nt_kernel(float* __restrict__ w32, u16* __restrict__ wc,
+                              const u16* __restrict__ g,
+                              float* __restrict__ m32, i64 n, float lr,
+                              float mu, float wd, float scale) {
+  CHUNK_LOOP(n / 8) {
+    f32x4 w0 = __builtin_nontemporal_load(
+        reinterpret_cast<const f32x4*>(w32 + c * 8));
+    f32x4 w1 = __builtin_nontemporal_load(
+        reinterpret_cast<const f32x4*>(w32 + c * 8 + 4));
+    f32x4 m0 = __builtin_nontemporal_load(
+        reinterpret_cast<const f32x4*>(m32 + c * 8));
+    f32x4 m1 = __builtin_nontemporal_load(
+        reinterpret_cast<const f32x4*>(m32 + c * 8 + 4));
+    s16x8 gv = __builtin_nontemporal_load(
+        reinterpret_cast<const s16x8*>(g + c * 8));
+    s16x8 wo;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gj = bf2f(u16(gv[j])) * scale + wd * w0[j];
+      m0[j] = mu * m0[j] + gj;
+      w0[j] -= lr * m0[j];
+      wo[j] = short(f2bf(w0[j]));
+      float gk = bf2f(u16(gv[4 + j])) * scale + wd * w1[j];
+      m1[j] = mu * m1[j] + gk;
+      w1[j] -= lr * m1[j];
+      wo[4 + j] = short(f2bf(w1[j]));
+    }
+    __builtin_nontemporal_store(w0, reinterpret_cast<f32x4*>(w32 + c * 8));
+    __builtin_nontemporal_store(w1, reinterpret_cast<f32x4*>(w32 + c * 8 + 4));
+    __builtin_nontemporal_store(m0, reinterpret_cast<f32x4*>(m32 + c * 8));
+    __builtin_nontemporal_store(m1, reinterpret_cast<f32x4*>(m32 + c * 8 + 4));
+    __builtin_nontemporal_store(wo, reinterpret_cast<s16x8*>(wc + c * 8));
+  }
+}
+
 void bump_weight_gen();  // gemm_ops.hip: weight-transform cache generation
 
 void sgd_step(at::Tensor& w32, at::Tensor& wc, const at::Tensor& g,
@@ -195,10 +232,20 @@ void sgd_step(at::Tensor& w32, at::Tensor& wc, const at::Tensor& g,
   check_bf16(wc, "wc"); check_bf16(g, "g");
   i64 n = w32.numel();
   TORCH_CHECK(n % 8 == 0, "arena must be 8-aligned");  // _ALIGN=128 in arena
-  hipLaunchKernelGGL(sgd_kernel, dim3(grid_for(n / 8)), dim3(256), 0,
-                     cur_stream(), w32.data_ptr<float>(), bfp_mut(wc), bfp(g),
-                     m32.data_ptr<float>(), n, float(lr), float(mu),
-                     float(wd), float(scale));
+  static const bool nt = [] {
+    const char* e = getenv("DVGG_SGD_NT");
+    return e && e[0] == '1';
+  }();
+  if (nt)
+    hipLaunchKernelGGL(sgd_nt_kernel, dim3(grid_for(n / 8)), dim3(256), 0,
+                       cur_stream(), w32.data_ptr<float>(), bfp_mut(wc),
+                       bfp(g), m32.data_ptr<float>(), n, float(lr), float(mu),
+                       float(wd), float(scale));
+  else
+    hipLaunchKernelGGL(sgd_kernel, dim3(grid_for(n / 8)), dim3(256), 0,
+                       cur_stream(), w32.data_ptr<float>(), bfp_mut(wc),
+                       bfp(g), m32.data_ptr<float>(), n, float(lr), float(mu),
+                       float(wd), float(scale));
   bump_weight_gen();  // wc changed: cached wT/wD/w8 refresh on next forward
 }
 
