@@ -42,25 +42,14 @@ static inline int grid_for(i64 chunks) {
 __global__ void relu_bwd_kernel(const u16* __restrict__ dy,
                                 const u16* __restrict__ y,
                                 u16* __restrict__ out, i64 n) {
-  // two 8-chunks per iteration (array halves): 4 b128 loads in flight
-  // per thread — the single-chunk form plateaued ~3 TB/s, latency-bound
-  i64 total8 = n / 8, H = (total8 + 1) / 2;
-  CHUNK_LOOP(H) {
-    i64 c1 = c + H;
-    bool two = c1 < total8;
-    s16x8 d0 = *reinterpret_cast<const s16x8*>(dy + c * 8);
-    s16x8 v0 = *reinterpret_cast<const s16x8*>(y + c * 8);
-    i64 o1 = two ? c1 * 8 : 0;
-    s16x8 d1 = *reinterpret_cast<const s16x8*>(dy + o1);
-    s16x8 v1 = *reinterpret_cast<const s16x8*>(y + o1);
-    s16x8 a, b;
+  CHUNK_LOOP(n / 8) {
+    s16x8 d = *reinterpret_cast<const s16x8*>(dy + c * 8);
+    s16x8 v = *reinterpret_cast<const s16x8*>(y + c * 8);
+    s16x8 o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      a[j] = (bf2f(u16(v0[j])) > 0.f) ? d0[j] : short(0);
-      b[j] = (bf2f(u16(v1[j])) > 0.f) ? d1[j] : short(0);
-    }
-    *reinterpret_cast<s16x8*>(out + c * 8) = a;
-    if (two) *reinterpret_cast<s16x8*>(out + o1) = b;
+    for (int j = 0; j < 8; ++j)
+      o[j] = (bf2f(u16(v[j])) > 0.f) ? d[j] : short(0);
+    *reinterpret_cast<s16x8*>(out + c * 8) = o;
   }
   i64 tail = n & 7;
   if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
@@ -73,7 +62,7 @@ at::Tensor relu_bwd(const at::Tensor& dy, const at::Tensor& y) {
   check_bf16(dy, "dy"); check_bf16(y, "y");
   auto out = at::empty_like(dy);
   i64 n = dy.numel();
-  hipLaunchKernelGGL(relu_bwd_kernel, dim3(grid_for(n / 16)), dim3(256), 0,
+  hipLaunchKernelGGL(relu_bwd_kernel, dim3(grid_for(n / 8)), dim3(256), 0,
                      cur_stream(), bfp(dy), bfp(y), bfp_mut(out), n);
   return out;
 }
