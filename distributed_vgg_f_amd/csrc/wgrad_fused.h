@@ -286,7 +286,14 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
   constexpr int R = 32 / CT;               // rows per tile / per set
   constexpr int RH = R + KH - 1;           // live window rows
   constexpr int S0 = (RH + R - 1) / R;     // window sets
-  constexpr int NSS = 2 * S0 + 2;          // ring slots (seam batch safe)
+  // Pipeline lookahead in TILES. LK=3 (deeper A/B rings, steady
+  // vmcnt(4), seam at LT-LK staging LK next-run tiles) was MEASURED
+  // WORSE: +6% on the 3x3 ct32 variant (906 vs 855 us — more in-flight
+  // sets pressure L2 without covering more useful latency) and +15% on
+  // ct16 (the 80->112 KB LDS growth halved its 2-workgroups/CU
+  // occupancy). The counts below stay generic in LK for re-evaluation.
+  constexpr int LK = 2;
+  constexpr int NSS = 2 * S0 + 2 * (LK - 1);  // A ring slots (seam safe)
   constexpr int CTW = CT + 8;
   constexpr int CPR = CTW / 8;
   constexpr int SETC = R * BMC * CPR;      // real chunks per set (<=512)
@@ -294,7 +301,7 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
   constexpr int SETE = 512 * 8;            // padded set elems (8 KiB)
   constexpr int BTC = 4 * BNC;             // real B chunks per tile
   constexpr int BTE = 512 * 8;             // padded B tile elems (8 KiB)
-  constexpr int NB = 4;                    // B ring slots
+  constexpr int NB = 2 * LK;               // B ring slots
   __shared__ u16 smem[NSS * SETE + NB * BTE];
 
   const int tiles_m = Cin / BMC, tiles_n = Cout / BNC;
@@ -376,10 +383,13 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
 #pragma unroll
   for (int s = 0; s < S0; ++s) WGR_STAGE_A(s % NSS, n_r, c0_r, s);
   WGR_STAGE_B(0, n_r, c0_r, 0);
-  WGR_STAGE_A(S0 % NSS, n_r, c0_r, S0);
-  WGR_STAGE_B(1, n_r, c0_r, 1);
-  int sfree = (S0 + 1) % NSS;  // next free A slot
-  int bfree = 2;
+#pragma unroll
+  for (int lk = 1; lk < LK; ++lk) {        // tiles 1..LK-1 of the run
+    WGR_STAGE_A((S0 + lk - 1) % NSS, n_r, c0_r, S0 + lk - 1);
+    WGR_STAGE_B(lk, n_r, c0_r, lk);
+  }
+  int sfree = (S0 + LK - 1) % NSS;  // next free A slot
+  int bfree = LK;
 
   for (; run < r_hi; ++run) {
     if (run > r_lo) { n_r = run / SEG; c0_r = (run % SEG) * 32; }
@@ -388,13 +398,26 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
     const int nc0 = (run + 1 < r_hi) ? ((run + 1) % SEG) * 32 : 0;
     for (int tile = 0; tile < LT; ++tile) {
       // ---- wait + barrier: current tile's window and B tile are in LDS
-      const bool seam = (tile == LT - 1) && !last_run;
-      if (tile == LT - 1 && last_run)
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      else if (seam)
-        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(S0 + 3) : "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      // younger-than-this-tile's-pair load counts (in-order retirement):
+      // steady = the LK-1 staged-ahead pairs; behind==1/0 tails shrink by
+      // 2 each; a pending seam batch adds its S0 + 2*LK - 1 loads.
+      const int behind = LT - 1 - tile;
+      if (last_run) {
+        if (behind == 0)
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        else if (LK == 3 && behind == 1)
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * (LK - 1)) : "memory");
+      } else {
+        if (behind == 0)
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(S0 + 2 * LK - 1)
+                       : "memory");
+        else if (LK == 3 && behind == 1)
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(S0 + 7) : "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * (LK - 1)) : "memory");
+      }
       __builtin_amdgcn_s_barrier();
 
       const u16* Bs_ = &smem[NSS * SETE + (i64)bbase * BTE];
@@ -423,12 +446,12 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
         if (kh == 0) {
           asm volatile("" ::: "memory");
           // ---- issue next stages under the MFMAs
-          if (tile + 2 < LT) {                  // regular: tile+2's new set
-            WGR_STAGE_A(sfree, n_r, c0_r, tile + 2 + S0 - 1);
+          if (tile + LK < LT) {                 // regular: tile+LK's set
+            WGR_STAGE_A(sfree, n_r, c0_r, tile + LK + S0 - 1);
             sfree = sfree + 1 == NSS ? 0 : sfree + 1;
-            WGR_STAGE_B(bfree, n_r, c0_r, tile + 2);
+            WGR_STAGE_B(bfree, n_r, c0_r, tile + LK);
             bfree = bfree + 1 == NB ? 0 : bfree + 1;
-          } else if (tile == LT - 2 && !last_run) {
+          } else if (tile == LT - LK && !last_run) {
             // seam batch: next run's prologue in the prologue's issue
             // order (tile0' window + B0', then tile1' top + B1')
 #pragma unroll
@@ -439,15 +462,16 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
             }
             WGR_STAGE_B(bfree, nn, nc0, 0);
             bfree = bfree + 1 == NB ? 0 : bfree + 1;
-            {
-              int sl = sfree + S0;
+#pragma unroll
+            for (int lk = 1; lk < LK; ++lk) {
+              int sl = sfree + S0 + lk - 1;
               while (sl >= NSS) sl -= NSS;
-              WGR_STAGE_A(sl, nn, nc0, S0);
+              WGR_STAGE_A(sl, nn, nc0, S0 + lk - 1);
+              WGR_STAGE_B(bfree, nn, nc0, lk);
+              bfree = bfree + 1 == NB ? 0 : bfree + 1;
             }
-            sfree = sfree + S0 + 1;
+            sfree = sfree + S0 + LK - 1;
             while (sfree >= NSS) sfree -= NSS;
-            WGR_STAGE_B(bfree, nn, nc0, 1);
-            bfree = bfree + 1 == NB ? 0 : bfree + 1;
           }
         }
 #pragma unroll
@@ -485,7 +509,7 @@ __global__ __launch_bounds__((WAVES_M * WAVES_N) * 64) void wgrad_ring_kernel(
     // in-run advance moved sbase by LT-1... recompute from sfree:
     // next run's sets occupy [sfree - (S0+1), sfree) -> first = that
     if (!last_run) {
-      sbase = sfree - (S0 + 1);
+      sbase = sfree - (S0 + LK - 1);
       while (sbase < 0) sbase += NSS;
     }
   }
