@@ -388,16 +388,29 @@ __global__ void cast_f32_bf16_2(const float* __restrict__ in,
 }
 
 // db[c] = bf16( sum over 256 slots of db32p[s][c] )
+// 1024 threads; for C < 1024 (power of 2) the spare waves fold into
+// slot sub-stripes + LDS combine — the single-wave form was a one-block
+// latency chain over the 256*C partials (~17 us for C=256).
 __global__ void bias_slot_reduce_kernel(float* __restrict__ db32p,
                                         u16* __restrict__ db, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float ls[1024];
+  int t = threadIdx.x;
+  int W = 1, c = blockIdx.x * blockDim.x + t, w = 0;
+  if (C < 1024 && (C & (C - 1)) == 0) {
+    W = 1024 / C; c = t & (C - 1); w = t / C;
+  } else if (c >= C) {
+    return;
+  }
   float a = 0.f;
-#pragma unroll 8
-  for (int s = 0; s < 256; ++s) a += db32p[(i64)s * C + c];
+  for (int s = w; s < 256; s += W) a += db32p[(i64)s * C + c];
+  for (int s = w; s < 256; s += W) db32p[(i64)s * C + c] = 0.f;
+  if (W > 1) {
+    ls[t] = a;
+    __syncthreads();
+    if (t >= C) return;
+    for (int j = 1; j < W; ++j) a += ls[t + j * C];
+  }
   db[c] = f2bf(a);
-#pragma unroll 8
-  for (int s = 0; s < 256; ++s) db32p[(i64)s * C + c] = 0.f;  // scratch_c32
 }
 
 // Vectorized preamble (see bwd_preamble_kernel below for semantics):
@@ -1219,16 +1232,28 @@ __global__ void nhwc2phase_nhwc_kernel(const u16* __restrict__ x,
   int PC = s * s * C;                      // real phase-channels
   int PCp = (PC + 7) & ~7;                 // stored (zero-padded to 8)
   int chunks = PCp / 8;
-  for (int job = t; job < Wq * chunks; job += blockDim.x) {
-    int wq = job / chunks, k8 = (job % chunks) * 8;
+  // fixed chunk per thread: the 8 (ph,pw,c) decodes (v_rcp div chains
+  // for C=3) hoist out of the position loop
+  int wstride = blockDim.x / chunks;
+  if (t >= chunks * wstride) return;       // (after the barrier above)
+  int k8 = (t % chunks) * 8;
+  int ofs[8], pw8[8];
+  bool inr[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int pcc = k8 + j;
+    int c = pcc % C, pw = (pcc / C) % s, ph = pcc / (C * s);
+    ofs[j] = ph * W * C + c;
+    pw8[j] = pw - pad;
+    inr[j] = pcc < PC;
+  }
+  for (int wq = t / chunks; wq < Wq; wq += wstride) {
     union { s16x8 v; u16 e[8]; } pk;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int pcc = k8 + j;
-      int c = pcc % C, pw = (pcc / C) % s, ph = pcc / (C * s);
-      int iw = wq * s + pw - pad;
-      pk.e[j] = (pcc < PC && (unsigned)iw < (unsigned)W)
-          ? rows[ph * W * C + iw * C + c] : u16(0);
+      int iw = wq * s + pw8[j];
+      pk.e[j] = (inr[j] && (unsigned)iw < (unsigned)W)
+          ? rows[ofs[j] + iw * C] : u16(0);
     }
     *reinterpret_cast<s16x8*>(
         out + (((i64)n * Hq + hq) * Wqp + wq) * PCp + k8) = pk.v;
@@ -1626,9 +1651,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor db;
     if (need_db) {
       db = db_out ? *db_out : at::empty({Cout}, dy.options());
-      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
-                         dim3(256), 0, s, db32.data_ptr<float>(),
-                         bfp_mut(db), Cout);
+      hipLaunchKernelGGL(bias_slot_reduce_kernel,
+                         dim3(ceil_div(Cout, 1024)), dim3(1024), 0, s,
+                         db32.data_ptr<float>(), bfp_mut(db), Cout);
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
@@ -1640,9 +1665,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor db;
     if (need_db) {
       db = db_out ? *db_out : at::empty({Cout}, dy.options());
-      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
-                         dim3(256), 0, s, db32.data_ptr<float>(),
-                         bfp_mut(db), Cout);
+      hipLaunchKernelGGL(bias_slot_reduce_kernel,
+                         dim3(ceil_div(Cout, 1024)), dim3(1024), 0, s,
+                         db32.data_ptr<float>(), bfp_mut(db), Cout);
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
@@ -1685,8 +1710,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   at::Tensor db;
   if (need_db) {
     db = db_out ? *db_out : at::empty({Cout}, dy.options());
-    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
-                       dim3(256), 0, s, db32.data_ptr<float>(), bfp_mut(db),
+    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 1024)),
+                       dim3(1024), 0, s, db32.data_ptr<float>(), bfp_mut(db),
                        Cout);
   }
   return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
