@@ -388,29 +388,18 @@ __global__ void cast_f32_bf16_2(const float* __restrict__ in,
 }
 
 // db[c] = bf16( sum over 256 slots of db32p[s][c] )
-// 1024 threads; for C < 1024 (power of 2) the spare waves fold into
-// slot sub-stripes + LDS combine — the single-wave form was a one-block
-// latency chain over the 256*C partials (~17 us for C=256).
+// (a 1024-thread sub-stripe fold was tried and measured WORSE,
+// 17.2 -> 20.9 us avg — the one-block form is already L2-served)
 __global__ void bias_slot_reduce_kernel(float* __restrict__ db32p,
                                         u16* __restrict__ db, int C) {
-  __shared__ float ls[1024];
-  int t = threadIdx.x;
-  int W = 1, c = blockIdx.x * blockDim.x + t, w = 0;
-  if (C < 1024 && (C & (C - 1)) == 0) {
-    W = 1024 / C; c = t & (C - 1); w = t / C;
-  } else if (c >= C) {
-    return;
-  }
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   float a = 0.f;
-  for (int s = w; s < 256; s += W) a += db32p[(i64)s * C + c];
-  for (int s = w; s < 256; s += W) db32p[(i64)s * C + c] = 0.f;
-  if (W > 1) {
-    ls[t] = a;
-    __syncthreads();
-    if (t >= C) return;
-    for (int j = 1; j < W; ++j) a += ls[t + j * C];
-  }
+#pragma unroll 8
+  for (int s = 0; s < 256; ++s) a += db32p[(i64)s * C + c];
   db[c] = f2bf(a);
+#pragma unroll 8
+  for (int s = 0; s < 256; ++s) db32p[(i64)s * C + c] = 0.f;  // scratch_c32
 }
 
 // Vectorized preamble (see bwd_preamble_kernel below for semantics):
@@ -1651,9 +1640,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor db;
     if (need_db) {
       db = db_out ? *db_out : at::empty({Cout}, dy.options());
-      hipLaunchKernelGGL(bias_slot_reduce_kernel,
-                         dim3(ceil_div(Cout, 1024)), dim3(1024), 0, s,
-                         db32.data_ptr<float>(), bfp_mut(db), Cout);
+      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                         dim3(256), 0, s, db32.data_ptr<float>(),
+                         bfp_mut(db), Cout);
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
@@ -1665,9 +1654,9 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
     at::Tensor db;
     if (need_db) {
       db = db_out ? *db_out : at::empty({Cout}, dy.options());
-      hipLaunchKernelGGL(bias_slot_reduce_kernel,
-                         dim3(ceil_div(Cout, 1024)), dim3(1024), 0, s,
-                         db32.data_ptr<float>(), bfp_mut(db), Cout);
+      hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                         dim3(256), 0, s, db32.data_ptr<float>(),
+                         bfp_mut(db), Cout);
     }
     return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
   }
@@ -1710,8 +1699,8 @@ std::vector<at::Tensor> conv2d_bwd(const at::Tensor& dy, const at::Tensor& y,
   at::Tensor db;
   if (need_db) {
     db = db_out ? *db_out : at::empty({Cout}, dy.options());
-    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 1024)),
-                       dim3(1024), 0, s, db32.data_ptr<float>(), bfp_mut(db),
+    hipLaunchKernelGGL(bias_slot_reduce_kernel, dim3(ceil_div(Cout, 256)),
+                       dim3(256), 0, s, db32.data_ptr<float>(), bfp_mut(db),
                        Cout);
   }
   return {need_dx ? dx : at::Tensor(), dw, need_db ? db : at::Tensor()};
