@@ -53,6 +53,22 @@ def test_linear_fwd_fc6_shape_splitk():
     check(y, ref, 0.05, "fc6")
 
 
+def test_splitk_scratch_rezero_contract():
+    """The split-K atomic accumulators are CACHED per (site, shape) and
+    re-zeroed by their finalize kernel (gemm_ops.hip scratch_c32), so a
+    broken re-zero shows up as the SECOND call of the same shape
+    accumulating on top of the first. Run the split-K fc6 shape twice
+    with different inputs and check the second result independently."""
+    torch.manual_seed(7)
+    w, b = rnd(9216, 4096, scale=0.02), rnd(4096)
+    wd, bd = w.to(DEV), b.to(DEV)
+    for trial in range(2):
+        x = rnd(256, 9216, scale=0.1, seed=100 + trial)
+        y = ext.linear_fwd(x.to(DEV), wd, bd, False)
+        check(y, cpu_ref.linear(x.float(), w.float(), b.float()), 0.05,
+              f"scratch-rezero trial {trial}")
+
+
 def test_linear_fwd_fc8_nonmult128():
     torch.manual_seed(2)
     x, w, b = rnd(12, 4096, scale=0.1), rnd(4096, 1000, scale=0.02), rnd(1000)
